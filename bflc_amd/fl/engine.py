@@ -1,0 +1,295 @@
+"""Barrier-driven committee-consensus FL round engine.
+
+One process per GPU; each rank owns a contiguous slice of the FL
+clients and a full replica of the deterministic ledger. A round replays
+the reference protocol without any polling or sleeps:
+
+  phase T  designated trainers run the local HIP train step
+  phase U  one rank-ordered all-gather publishes (meta, delta) updates;
+           every rank feeds ALL submissions to its ledger replica in the
+           same order => identical admission decisions
+           (reference UploadLocalUpdate guards, .cpp:215-257)
+  phase S  committee clients score every accepted candidate on their own
+           shard (reference local_scoring, main.py:196-217)
+  phase V  one all-gather publishes score maps; the comm_count-th feed
+           triggers the aggregation decision (.cpp:259-297)
+  phase A  every rank applies the identical weighted-FedAvg kernel with
+           a fixed accumulation order and commits (.cpp:349-455)
+
+The reference resolves the 16-trainers-race-for-10-slots with
+first-come PBFT ordering; here admission order is the deterministic
+(rank, client) order, so ranks can PREDICT which updates would be
+admitted and skip training the surplus trainers entirely — same
+semantics (rejected updates never influence state), less wasted compute.
+
+Data plane (flat fp32 delta tensors, the global model) stays resident on
+the GPU; the ledger's blobs hold b"" markers in engine mode (the
+ABI-compatible JSON blob path lives in bflc_amd.chain.client).
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from bflc_amd._ledger import Admit, CommitteeLedger
+from bflc_amd.comm import Transport
+from bflc_amd.config import FLConfig
+from bflc_amd.data.synthetic import Shard
+from bflc_amd.models import build_model
+from bflc_amd.ops import functional as O
+
+
+def client_rank(client_idx: int, n_clients: int, world: int) -> int:
+    """Contiguous array_split-style client->rank assignment."""
+    base, rem = divmod(n_clients, world)
+    # ranks 0..rem-1 own (base+1) clients
+    cut = rem * (base + 1)
+    if client_idx < cut:
+        return client_idx // (base + 1)
+    return rem + (client_idx - cut) // max(base, 1)
+
+
+@dataclass
+class RoundStats:
+    epoch: int
+    wall_s: float
+    train_s: float
+    gather_s: float
+    score_s: float
+    aggregate_s: float
+    global_loss: float
+    n_updates: int
+    n_selected: int
+    samples_trained: int
+    test_acc: Optional[float] = None
+
+
+class FLEngine:
+    def __init__(self, cfg: FLConfig, transport: Transport,
+                 shards: List[Shard], test_shard: Optional[Shard] = None,
+                 ) -> None:
+        self.cfg = cfg
+        self.t = transport
+        self.device = transport.device
+        self.rank = transport.rank
+        self.world = transport.world_size
+
+        # client ownership
+        self.origins = [f"node_{i}" for i in range(cfg.client_num)]
+        self.local_clients = [i for i in range(cfg.client_num)
+                              if client_rank(i, cfg.client_num, self.world)
+                              == self.rank]
+        self.shards = {i: shards[i].to(self.device) for i in self.local_clients}
+        self.test_shard = test_shard.to(self.device) if test_shard else None
+
+        # one shared model instance per rank (set_flat per client)
+        self.model = build_model(cfg, self.device)
+        self.global_flat = self.model.get_flat()  # device-resident fp32
+
+        # replicated ledger: registration = deterministic replay of the
+        # rank-ordered client lists (reference RegisterNode, .cpp:168-190)
+        self.ledger = CommitteeLedger(cfg.ledger_config())
+        gathered = self.t.all_gather_objects(self.local_clients)
+        for rank_clients in gathered:
+            for i in rank_clients:
+                self.ledger.register_node(self.origins[i])
+        assert self.ledger.epoch == 0, "registration did not start FL"
+        self.ledger.set_global_model(b"")
+
+        self._round = 0
+
+    # ------------------------------------------------------------------
+    def _planned_submitters(self) -> List[int]:
+        """The deterministic prefix of trainers whose updates will be
+        admitted (first needed_update_count in (rank, client) order)."""
+        roles = self.ledger.roles()
+        order: List[int] = []
+        gathered_order = range(self.cfg.client_num)  # origins are id-sorted
+        # global (rank, client-index) order == client index order because
+        # assignment is contiguous ascending
+        for i in gathered_order:
+            role = roles.get(self.origins[i], "trainer")
+            if role == "trainer" or self.cfg.self_scoring:
+                order.append(i)
+        return order[: self.cfg.needed_update_count]
+
+    # ------------------------------------------------------------------
+    def _local_train(self, client: int) -> Tuple[torch.Tensor, int, float]:
+        """Local train step (reference local_training, main.py:103-158):
+        start from the global model, run local_epochs passes of
+        minibatch SGD/Adam, return pseudo-gradient delta=(W0-W)/lr."""
+        cfg = self.cfg
+        shard = self.shards[client]
+        self.model.set_flat(self.global_flat)
+        n = shard.n
+        bs = min(cfg.batch_size, n)
+        total_batches = max(n // bs, 1)
+        avg_cost = 0.0
+        if cfg.optimizer == "adam":
+            m = torch.zeros_like(self.global_flat)
+            v = torch.zeros_like(self.global_flat)
+            step = 0
+        for _ in range(cfg.local_epochs):
+            for bi in range(total_batches):
+                xb = shard.x[bi * bs:(bi + 1) * bs]
+                yb = shard.y[bi * bs:(bi + 1) * bs]
+                self.model.zero_grad()
+                loss = self.model.loss(xb, yb)
+                loss.backward()
+                if cfg.optimizer == "adam":
+                    step += 1
+                    O.adam_step_(self.model.flat.data, self.model.flat.grad,
+                                 m, v, step, cfg.learning_rate)
+                else:
+                    O.sgd_step_(self.model.flat.data, self.model.flat.grad,
+                                cfg.learning_rate)
+                avg_cost += float(loss.detach()) / (total_batches
+                                                    * cfg.local_epochs)
+        # delta = (W0 - W)/lr  (reference main.py:153-154)
+        delta = self.global_flat.clone()
+        O.axpy_(delta, -1.0, self.model.flat.data)
+        delta.div_(cfg.learning_rate)
+        return delta, n, avg_cost
+
+    # ------------------------------------------------------------------
+    def _score_candidates(self, scorer: int,
+                          updates: List[Tuple[str, torch.Tensor]]
+                          ) -> Dict[str, float]:
+        """Committee scoring (reference local_scoring, main.py:196-217):
+        candidate = W0 - lr*delta, scored by accuracy on the scorer's own
+        local shard."""
+        shard = self.shards[scorer]
+        out: Dict[str, float] = {}
+        for origin, delta in updates:
+            cand = self.global_flat.clone()
+            O.axpy_(cand, -self.cfg.learning_rate, delta)
+            self.model.set_flat(cand)
+            out[origin] = self.model.accuracy(shard.x, shard.y)
+        return out
+
+    # ------------------------------------------------------------------
+    def run_round(self, eval_global: bool = False) -> RoundStats:
+        cfg, led = self.cfg, self.ledger
+        epoch = led.epoch
+        t0 = time.perf_counter()
+
+        # ---- phase T: local training on designated submitters ----------
+        submitters = self._planned_submitters()
+        local_subs = [i for i in submitters if i in self.shards]
+        local_updates: List[Tuple[str, torch.Tensor, int, float]] = []
+        samples_trained = 0
+        for i in local_subs:
+            delta, n, cost = self._local_train(i)
+            local_updates.append((self.origins[i], delta, n, cost))
+            samples_trained += n * cfg.local_epochs
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        t1 = time.perf_counter()
+
+        # ---- phase U: publish updates (one RCCL all-gather) -------------
+        metas = [(o, epoch, n, c) for (o, delta, n, c) in local_updates]
+        all_metas = self.t.all_gather_objects(metas)
+        max_subs = max((len(m) for m in all_metas), default=0)
+        P = self.global_flat.numel()
+        stack = torch.zeros(max(max_subs, 1), P, dtype=torch.float32,
+                            device=self.global_flat.device)
+        for j, (_, delta, _, _) in enumerate(local_updates):
+            stack[j] = delta
+        gathered = self.t.all_gather_tensor(stack)
+        updates: List[Tuple[str, torch.Tensor]] = []
+        for r, rank_metas in enumerate(all_metas):
+            for j, (origin, ep, n, c) in enumerate(rank_metas):
+                code = led.upload_local_update(origin, b"", ep, n, c)
+                if code == Admit.ACCEPTED:
+                    updates.append((origin, gathered[r][j].to(self.device)))
+        t2 = time.perf_counter()
+
+        # ---- phase S: committee scoring ---------------------------------
+        roles = led.roles()
+        if cfg.self_scoring:
+            local_scorers = list(self.shards.keys())
+        else:
+            local_scorers = [i for i in self.local_clients
+                             if roles.get(self.origins[i]) == "comm"]
+        my_scores = [(self.origins[i], self._score_candidates(i, updates))
+                     for i in local_scorers]
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        t3 = time.perf_counter()
+
+        # ---- phase V: publish scores, decide ----------------------------
+        all_scores = self.t.all_gather_objects(my_scores)
+        decision = None
+        for rank_scores in all_scores:
+            for origin, smap in rank_scores:
+                d = led.upload_scores(origin, epoch, smap)
+                if d is not None:
+                    decision = d
+        if decision is None:
+            raise RuntimeError(
+                f"round {epoch}: no aggregation decision "
+                f"(scores={led.score_count}/{cfg.comm_count})")
+
+        # ---- phase A: weighted FedAvg + commit --------------------------
+        by_origin = dict(updates)
+        sel = decision.selected
+        K = len(sel)
+        deltas = torch.empty(K, P, dtype=torch.float32,
+                             device=self.global_flat.device)
+        weights = torch.empty(K, dtype=torch.float32,
+                              device=self.global_flat.device)
+        for k, (origin, w) in enumerate(sel):
+            deltas[k] = by_origin[origin]
+            weights[k] = float(w)
+        self.last_decision = decision
+        avg = O.weighted_fedavg(deltas, weights)
+        # global -= lr * avg  (reference .cpp:403-414)
+        O.axpy_(self.global_flat, -cfg.learning_rate, avg)
+        led.commit_aggregate(b"")
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        t4 = time.perf_counter()
+
+        acc = None
+        if eval_global and self.test_shard is not None:
+            acc = self.evaluate_global()
+        self._round += 1
+        return RoundStats(
+            epoch=epoch, wall_s=t4 - t0, train_s=t1 - t0, gather_s=t2 - t1,
+            score_s=t3 - t2, aggregate_s=t4 - t3,
+            global_loss=led.global_loss, n_updates=len(updates),
+            n_selected=K, samples_trained=samples_trained, test_acc=acc)
+
+    # ------------------------------------------------------------------
+    def evaluate_global(self) -> float:
+        """Sponsor evaluation (reference run_sponsor, main.py:280-340)."""
+        assert self.test_shard is not None
+        self.model.set_flat(self.global_flat)
+        return self.model.accuracy(self.test_shard.x, self.test_shard.y)
+
+    def run(self, rounds: int, eval_every: int = 0) -> List[RoundStats]:
+        out = []
+        for r in range(rounds):
+            ev = eval_every > 0 and (r + 1) % eval_every == 0
+            out.append(self.run_round(eval_global=ev))
+        return out
+
+    # ------------------------------------------------------------------
+    # checkpoint / resume (chain persistence -> snapshot)
+    def save(self, path: str) -> None:
+        torch.save({
+            "config": self.cfg.to_dict(),
+            "ledger": self.ledger.snapshot(),
+            "global_flat": self.global_flat.cpu(),
+            "round": self._round,
+        }, path)
+
+    def load(self, path: str) -> None:
+        ck = torch.load(path, map_location="cpu", weights_only=False)
+        assert ck["config"]["model"] == self.cfg.model
+        self.ledger.restore(ck["ledger"])
+        self.global_flat = ck["global_flat"].to(self.device)
+        self._round = ck["round"]

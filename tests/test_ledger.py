@@ -1,0 +1,226 @@
+"""Property tests for the deterministic committee ledger.
+
+Mirrors the reference contract's state machine semantics
+(CommitteePrecompiled.cpp:132-311, Aggregate .cpp:349-455): epoch
+guards, duplicate rejection, update quota, committee-only scoring,
+median -> top-k -> rotation, and checkpoint/restore.
+"""
+import pytest
+
+from bflc_amd._ledger import Admit, CommitteeLedger, LedgerConfig, median_ref
+
+
+def make_ledger(n=20, comm=4, needed=10, agg=6):
+    lc = LedgerConfig()
+    lc.client_num, lc.comm_count = n, comm
+    lc.needed_update_count, lc.aggregate_count = needed, agg
+    return CommitteeLedger(lc)
+
+
+def register_all(led, n=20):
+    for i in range(n):
+        led.register_node(f"node_{i}")
+
+
+class TestRegistration:
+    def test_epoch_starts_uninitialized(self):
+        led = make_ledger()
+        assert led.epoch == -999  # reference .cpp:322
+
+    def test_epoch_zero_after_full_registration(self):
+        led = make_ledger()
+        register_all(led)
+        assert led.epoch == 0
+
+    def test_first_registrants_become_committee(self):
+        led = make_ledger()
+        register_all(led)
+        assert led.committee() == [f"node_{i}" for i in range(4)]
+        assert len(led.trainers()) == 16
+
+    def test_duplicate_registration_ignored(self):
+        led = make_ledger()
+        register_all(led)
+        assert led.register_node("node_0") is False
+        assert led.epoch == 0
+
+    def test_unregistered_query_state_defaults_trainer(self):
+        led = make_ledger()
+        role, ep = led.query_state("ghost")
+        assert role == "trainer" and ep == -999  # reference .cpp:196-199
+
+
+class TestUpdateAdmission:
+    def test_reject_before_start(self):
+        led = make_ledger()
+        assert led.upload_local_update("node_0", b"", -999, 1, 0.0) == \
+            Admit.NOT_STARTED
+
+    def test_stale_epoch_rejected(self):
+        led = make_ledger()
+        register_all(led)
+        assert led.upload_local_update("node_5", b"", 1, 10, 0.0) == \
+            Admit.STALE_EPOCH  # reference .cpp:225-226
+
+    def test_duplicate_rejected(self):
+        led = make_ledger()
+        register_all(led)
+        assert led.upload_local_update("node_5", b"a", 0, 10, 0.0) == \
+            Admit.ACCEPTED
+        assert led.upload_local_update("node_5", b"b", 0, 10, 0.0) == \
+            Admit.DUPLICATE  # reference .cpp:232-233
+
+    def test_quota(self):
+        led = make_ledger()
+        register_all(led)
+        for i in range(10):
+            assert led.upload_local_update(f"node_{4+i}", b"", 0, 10, 0.0) \
+                == Admit.ACCEPTED
+        assert led.upload_local_update("node_15", b"", 0, 10, 0.0) == \
+            Admit.QUOTA_FULL  # reference .cpp:239-244
+
+    def test_query_all_updates_empty_until_quota(self):
+        led = make_ledger()
+        register_all(led)
+        led.upload_local_update("node_5", b"x", 0, 10, 0.0)
+        assert led.query_all_updates() == []  # reference .cpp:304-307
+        for i in range(9):
+            led.upload_local_update(f"node_{6+i}", b"", 0, 10, 0.0)
+        ups = led.query_all_updates()
+        assert len(ups) == 10
+        assert ups[0] == ("node_5", b"x")  # acceptance order
+
+
+class TestScoring:
+    def _filled(self):
+        led = make_ledger()
+        register_all(led)
+        for i in range(10):
+            led.upload_local_update(f"node_{4+i}", b"", 0, 100 + i, 0.5)
+        return led
+
+    def test_trainer_scores_rejected(self):
+        led = self._filled()
+        assert led.upload_scores("node_10", 0, {"node_4": 1.0}) is None
+        assert led.score_count == 0  # reference .cpp:272-275
+
+    def test_stale_epoch_scores_rejected(self):
+        led = self._filled()
+        assert led.upload_scores("node_0", 1, {"node_4": 1.0}) is None
+        assert led.score_count == 0  # reference .cpp:266-269
+
+    def test_overwrite_does_not_double_count(self):
+        led = self._filled()
+        led.upload_scores("node_0", 0, {"node_4": 1.0})
+        led.upload_scores("node_0", 0, {"node_4": 0.9})
+        assert led.score_count == 1  # divergence: reference .cpp:279-289
+
+    def test_aggregation_on_last_committee_score(self):
+        led = self._filled()
+        trainers = [f"node_{4+i}" for i in range(10)]
+        for c in range(3):
+            scores = {t: 0.5 + 0.01 * i for i, t in enumerate(trainers)}
+            assert led.upload_scores(f"node_{c}", 0, scores) is None
+        scores = {t: 0.5 + 0.01 * i for i, t in enumerate(trainers)}
+        dec = led.upload_scores("node_3", 0, scores)
+        assert dec is not None  # reference .cpp:296-297
+        # top 6 by median score: node_13..node_8 in descending order
+        assert [s for s, _ in dec.selected] == \
+            [f"node_{13 - i}" for i in range(6)]
+        assert dec.total_weight == sum(100 + i for i in range(9, 3, -1))
+        # next committee = top 4
+        assert dec.next_committee == [f"node_{13 - i}" for i in range(4)]
+
+    def test_commit_rotates_roles_and_clears(self):
+        led = self._filled()
+        trainers = [f"node_{4+i}" for i in range(10)]
+        for c in range(4):
+            scores = {t: float(i) for i, t in enumerate(trainers)}
+            dec = led.upload_scores(f"node_{c}", 0, scores)
+        led.commit_aggregate(b"G1")
+        assert led.epoch == 1
+        assert led.update_count == 0 and led.score_count == 0
+        assert led.query_all_updates() == []
+        comm = led.committee()
+        assert comm == sorted([f"node_{13 - i}" for i in range(4)])
+        # old committee all back to trainer (reference .cpp:443-455)
+        for c in range(4):
+            assert led.query_state(f"node_{c}")[0] == "trainer"
+
+    def test_median_tiebreak_deterministic(self):
+        led = self._filled()
+        trainers = [f"node_{4+i}" for i in range(10)]
+        for c in range(4):
+            dec = led.upload_scores(f"node_{c}", 0,
+                                    {t: 0.5 for t in trainers})
+        # all tied -> lexicographic id order
+        assert [s for s, _ in dec.selected] == sorted(trainers)[:6]
+
+
+class TestMedian:
+    def test_even(self):
+        assert median_ref([1.0, 2.0, 3.0, 4.0]) == 2.5
+
+    def test_odd(self):
+        assert median_ref([3.0, 1.0, 2.0]) == 2.0
+
+    def test_reference_getmid_semantics(self):
+        # GetMid (.cpp:81-115): even n averages the two middles
+        assert median_ref([0.9, 0.1, 0.5, 0.7]) == pytest.approx(0.6)
+
+    def test_single(self):
+        assert median_ref([0.42]) == pytest.approx(0.42)
+
+
+class TestDeterminism:
+    def test_same_feed_same_state(self):
+        def run():
+            led = make_ledger()
+            register_all(led)
+            for i in range(10):
+                led.upload_local_update(f"node_{4+i}", b"", 0, 50 + i, 0.1)
+            trainers = [f"node_{4+i}" for i in range(10)]
+            for c in range(4):
+                dec = led.upload_scores(
+                    f"node_{c}", 0,
+                    {t: (hash(t + str(c)) % 100) / 100 for t in trainers})
+            led.commit_aggregate(b"G")
+            return led.snapshot()
+
+        a, b = run(), run()
+        assert a["roles"] == b["roles"]
+        assert a["epoch"] == b["epoch"]
+
+
+class TestSnapshot:
+    def test_roundtrip(self):
+        led = make_ledger()
+        register_all(led)
+        led.set_global_model(b"MODEL")
+        led.upload_local_update("node_7", b"U", 0, 33, 0.25)
+        snap = led.snapshot()
+
+        led2 = make_ledger()
+        led2.restore(snap)
+        assert led2.epoch == 0
+        assert led2.update_count == 1
+        assert led2.query_global_model() == (b"MODEL", 0)
+        assert led2.roles() == led.roles()
+        assert led2.update_meta("node_7") == (33, 0.25)
+        # duplicate still rejected after restore
+        assert led2.upload_local_update("node_7", b"U", 0, 33, 0.25) == \
+            Admit.DUPLICATE
+
+
+class TestSelfScoringWorld1:
+    def test_single_node_world(self):
+        led = make_ledger(n=1, comm=1, needed=1, agg=1)
+        led.register_node("node_0")
+        assert led.epoch == 0
+        assert led.query_state("node_0")[0] == "comm"
+        assert led.upload_local_update("node_0", b"", 0, 10, 0.5) == \
+            Admit.ACCEPTED
+        dec = led.upload_scores("node_0", 0, {"node_0": 0.8})
+        assert dec is not None and dec.selected == [("node_0", 10)]
+        led.commit_aggregate(b"")
+        assert led.epoch == 1
