@@ -63,7 +63,9 @@ def _torch_paths() -> tuple[list[str], list[str], list[str]]:
     from torch.utils import cpp_extension as ce
 
     incs = [f"-I{p}" for p in ce.include_paths(device_type="cuda")]
-    libdirs = [f"-L{p}" for p in ce.library_paths(device_type="cuda")]
+    libdirs = []
+    for p in ce.library_paths(device_type="cuda"):
+        libdirs += [f"-L{p}", f"-Wl,-rpath,{p}"]
     libs = ["-ltorch", "-ltorch_cpu", "-ltorch_python", "-lc10",
             "-ltorch_hip", "-lc10_hip", "-lamdhip64"]
     abi = int(torch._C._GLIBCXX_USE_CXX11_ABI)

@@ -197,7 +197,8 @@ class _MaxPool2dFn(torch.autograd.Function):
         (idx,) = ctx.saved_tensors
         dy = dy.contiguous()
         if dy.is_cuda:
-            dx = hip_ops().maxpool2d_bwd(dy, idx, list(ctx.in_shape))
+            dx = hip_ops().maxpool2d_bwd(dy, idx, list(ctx.in_shape),
+                                         ctx.kernel, ctx.stride)
         else:
             dx = F.max_unpool2d(dy, idx, ctx.kernel, ctx.stride,
                                 output_size=ctx.in_shape[-2:])

@@ -1,0 +1,67 @@
+// pybind11 bindings for the bflc_amd gfx950 HIP kernels.
+#include <torch/extension.h>
+
+#include <tuple>
+#include <vector>
+
+namespace bflc {
+
+// elementwise.hip
+void axpy_(torch::Tensor y, torch::Tensor x, double alpha);
+void sgd_step_(torch::Tensor p, torch::Tensor g, double lr);
+void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, long step, double lr, double beta1,
+                double beta2, double eps);
+torch::Tensor weighted_fedavg(torch::Tensor deltas, torch::Tensor w);
+torch::Tensor relu_fwd(torch::Tensor x);
+torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy);
+
+// softmax_ce.hip
+std::tuple<torch::Tensor, torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
+                                                        torch::Tensor target);
+torch::Tensor softmax_ce_bwd(torch::Tensor probs, torch::Tensor target,
+                             torch::Tensor gloss);
+
+// reduce.hip
+double accuracy(torch::Tensor logits, torch::Tensor target);
+
+// gemm_bf16.hip
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor dy);
+
+// conv_im2col.hip
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         long stride, long pad);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
+    long pad);
+std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
+                                                       long kernel,
+                                                       long stride);
+torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx,
+                            std::vector<long> in_shape, long kernel,
+                            long stride);
+
+}  // namespace bflc
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "bflc_amd hand-written gfx950 (CDNA4) HIP kernels";
+  m.def("axpy_", &bflc::axpy_, "y += alpha*x (fp32, in place)");
+  m.def("sgd_step_", &bflc::sgd_step_, "fused flat SGD");
+  m.def("adam_step_", &bflc::adam_step_, "fused flat Adam");
+  m.def("weighted_fedavg", &bflc::weighted_fedavg,
+        "fixed-order weighted FedAvg reduce");
+  m.def("relu_fwd", &bflc::relu_fwd);
+  m.def("relu_bwd", &bflc::relu_bwd);
+  m.def("softmax_ce_fwd", &bflc::softmax_ce_fwd,
+        "fused softmax cross-entropy fwd (loss, probs)");
+  m.def("softmax_ce_bwd", &bflc::softmax_ce_bwd);
+  m.def("accuracy", &bflc::accuracy, "fused argmax-compare-reduce");
+  m.def("linear_fwd", &bflc::linear_fwd, "MFMA bf16 GEMM + bias");
+  m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");
+  m.def("conv2d_fwd", &bflc::conv2d_fwd, "im2col + MFMA GEMM, NCHW");
+  m.def("conv2d_bwd", &bflc::conv2d_bwd, "(dx, dw, db)");
+  m.def("maxpool2d_fwd", &bflc::maxpool2d_fwd);
+  m.def("maxpool2d_bwd", &bflc::maxpool2d_bwd);
+}

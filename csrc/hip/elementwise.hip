@@ -1,0 +1,224 @@
+// Flat elementwise FL kernels (gfx950): fused SGD/Adam update, AXPY
+// delta/candidate math, weighted FedAvg reduce, ReLU.
+//
+// These own the reference ops: SGD apply_gradients (main.py:127-130),
+// Adam (main.py:126), delta extraction (W0-W)/lr and candidate
+// reconstruction W0-lr*dW (main.py:153-154, 215-216), and the on-chain
+// weighted FedAvg accumulate/apply (CommitteePrecompiled.cpp:373-414).
+//
+// All are HBM-bandwidth-bound: grid-stride float4 (16 B/lane) accesses,
+// grids sized >> 256 workgroups to fill all 8 XCDs.
+
+#include "common.h"
+
+namespace bflc {
+
+namespace {
+
+constexpr int kBlock = 256;
+
+__global__ void axpy_kernel(float* __restrict__ y,
+                            const float* __restrict__ x, float alpha,
+                            long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = n / 4;
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  float4* y4 = reinterpret_cast<float4*>(y);
+  for (long k = i; k < n4; k += stride) {
+    float4 a = y4[k], b = x4[k];
+    a.x = fmaf(alpha, b.x, a.x);
+    a.y = fmaf(alpha, b.y, a.y);
+    a.z = fmaf(alpha, b.z, a.z);
+    a.w = fmaf(alpha, b.w, a.w);
+    y4[k] = a;
+  }
+  for (long k = n4 * 4 + i; k < n; k += stride)
+    y[k] = fmaf(alpha, x[k], y[k]);
+}
+
+__global__ void sgd_kernel(float* __restrict__ p,
+                           const float* __restrict__ g, float lr, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = n / 4;
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  float4* p4 = reinterpret_cast<float4*>(p);
+  for (long k = i; k < n4; k += stride) {
+    float4 a = p4[k], b = g4[k];
+    a.x = fmaf(-lr, b.x, a.x);
+    a.y = fmaf(-lr, b.y, a.y);
+    a.z = fmaf(-lr, b.z, a.z);
+    a.w = fmaf(-lr, b.w, a.w);
+    p4[k] = a;
+  }
+  for (long k = n4 * 4 + i; k < n; k += stride)
+    p[k] = fmaf(-lr, g[k], p[k]);
+}
+
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            float lr, float beta1, float beta2, float eps,
+                            float bc1, float bc2, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride) {
+    float gk = g[k];
+    float mk = beta1 * m[k] + (1.f - beta1) * gk;
+    float vk = beta2 * v[k] + (1.f - beta2) * gk * gk;
+    m[k] = mk;
+    v[k] = vk;
+    float mhat = mk / bc1;
+    float vhat = vk / bc2;
+    p[k] -= lr * mhat / (sqrtf(vhat) + eps);
+  }
+}
+
+// avg[i] = sum_k w[k] * deltas[k][i] / wsum — FIXED ascending-k order so
+// every rank computes a bitwise-identical aggregate (the determinism the
+// reference got from single-sequential C++ loops, .cpp:374-399).
+__global__ void fedavg_kernel(const float* __restrict__ deltas,
+                              const float* __restrict__ w, int K, long P,
+                              float wsum, float* __restrict__ out) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long p4 = P / 4;
+  const float4* d4 = reinterpret_cast<const float4*>(deltas);
+  float4* o4 = reinterpret_cast<float4*>(out);
+  for (long j = i; j < p4; j += stride) {
+    float4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k = 0; k < K; ++k) {  // fixed order — do not reorder
+      float wk = w[k];
+      float4 d = d4[(long)k * p4 + j];
+      acc.x = fmaf(wk, d.x, acc.x);
+      acc.y = fmaf(wk, d.y, acc.y);
+      acc.z = fmaf(wk, d.z, acc.z);
+      acc.w = fmaf(wk, d.w, acc.w);
+    }
+    float inv = 1.f / wsum;
+    acc.x *= inv; acc.y *= inv; acc.z *= inv; acc.w *= inv;
+    o4[j] = acc;
+  }
+  for (long j = p4 * 4 + i; j < P; j += stride) {
+    float acc = 0.f;
+    for (int k = 0; k < K; ++k) acc = fmaf(w[k], deltas[(long)k * P + j], acc);
+    out[j] = acc / wsum;
+  }
+}
+
+template <typename T>
+__global__ void relu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride) {
+    float v = (float)x[k];
+    y[k] = (T)(v > 0.f ? v : 0.f);
+  }
+}
+
+template <typename T>
+__global__ void relu_bwd_kernel(const T* __restrict__ y,
+                                const T* __restrict__ dy, T* __restrict__ dx,
+                                long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride)
+    dx[k] = ((float)y[k] > 0.f) ? dy[k] : (T)0.f;
+}
+
+inline int grid_for(long n, int per_thread = 4) {
+  long blocks = (n + (long)kBlock * per_thread - 1) / ((long)kBlock * per_thread);
+  // >> 256 workgroups fills the 8 XCDs; cap to keep launch sane
+  return (int)std::min<long>(std::max<long>(blocks, 1), 8192);
+}
+
+}  // namespace
+
+void axpy_(torch::Tensor y, torch::Tensor x, double alpha) {
+  CHECK_GPU(y); CHECK_GPU(x); CHECK_CONTIG(y); CHECK_CONTIG(x);
+  TORCH_CHECK(y.scalar_type() == at::kFloat && x.scalar_type() == at::kFloat);
+  long n = y.numel();
+  TORCH_CHECK(x.numel() == n);
+  hipLaunchKernelGGL(axpy_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                     cur_stream(), y.data_ptr<float>(), x.data_ptr<float>(),
+                     (float)alpha, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void sgd_step_(torch::Tensor p, torch::Tensor g, double lr) {
+  CHECK_GPU(p); CHECK_GPU(g); CHECK_CONTIG(p); CHECK_CONTIG(g);
+  TORCH_CHECK(p.scalar_type() == at::kFloat && g.scalar_type() == at::kFloat);
+  long n = p.numel();
+  TORCH_CHECK(g.numel() == n);
+  hipLaunchKernelGGL(sgd_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                     cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
+                     (float)lr, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, long step, double lr, double beta1,
+                double beta2, double eps) {
+  CHECK_GPU(p); CHECK_CONTIG(p);
+  long n = p.numel();
+  float bc1 = 1.f - powf((float)beta1, (float)step);
+  float bc2 = 1.f - powf((float)beta2, (float)step);
+  hipLaunchKernelGGL(adam_kernel, dim3(grid_for(n, 2)), dim3(kBlock), 0,
+                     cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(), (float)lr,
+                     (float)beta1, (float)beta2, (float)eps, bc1, bc2, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+torch::Tensor weighted_fedavg(torch::Tensor deltas, torch::Tensor w) {
+  CHECK_GPU(deltas); CHECK_CONTIG(deltas); CHECK_GPU(w); CHECK_CONTIG(w);
+  TORCH_CHECK(deltas.dim() == 2 && w.dim() == 1);
+  TORCH_CHECK(deltas.size(0) == w.size(0));
+  TORCH_CHECK(deltas.scalar_type() == at::kFloat);
+  int K = (int)deltas.size(0);
+  long P = deltas.size(1);
+  float wsum = w.sum().item<float>();
+  auto out = torch::empty({P}, deltas.options());
+  hipLaunchKernelGGL(fedavg_kernel, dim3(grid_for(P, 2)), dim3(kBlock), 0,
+                     cur_stream(), deltas.data_ptr<float>(),
+                     w.data_ptr<float>(), K, P, wsum, out.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
+torch::Tensor relu_fwd(torch::Tensor x) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  auto y = torch::empty_like(x);
+  long n = x.numel();
+  if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(relu_fwd_kernel<bf16>, dim3(grid_for(n)), dim3(kBlock),
+                       0, cur_stream(), (const bf16*)x.data_ptr(),
+                       (bf16*)y.data_ptr(), n);
+  } else {
+    hipLaunchKernelGGL(relu_fwd_kernel<float>, dim3(grid_for(n)), dim3(kBlock),
+                       0, cur_stream(), x.data_ptr<float>(),
+                       y.data_ptr<float>(), n);
+  }
+  HIP_CHECK(hipGetLastError());
+  return y;
+}
+
+torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy) {
+  CHECK_GPU(y); CHECK_CONTIG(y); CHECK_CONTIG(dy);
+  auto dx = torch::empty_like(dy);
+  long n = y.numel();
+  if (y.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(relu_bwd_kernel<bf16>, dim3(grid_for(n)), dim3(kBlock),
+                       0, cur_stream(), (const bf16*)y.data_ptr(),
+                       (const bf16*)dy.data_ptr(), (bf16*)dx.data_ptr(), n);
+  } else {
+    hipLaunchKernelGGL(relu_bwd_kernel<float>, dim3(grid_for(n)), dim3(kBlock),
+                       0, cur_stream(), y.data_ptr<float>(),
+                       dy.data_ptr<float>(), dx.data_ptr<float>(), n);
+  }
+  HIP_CHECK(hipGetLastError());
+  return dx;
+}
+
+}  // namespace bflc

@@ -1,0 +1,28 @@
+// Internal API between the GEMM translation unit and its users (linear,
+// conv). All tensors bf16 on GPU; accumulation fp32 in MFMA AGPRs.
+#pragma once
+
+#include <torch/extension.h>
+
+namespace bflc {
+
+// Epilogue store modes.
+enum class EpStore : int {
+  kPlain = 0,   // C[m * N + n]
+  kConvNCHW = 1,  // m=(img,ohw) rows: C[((m/OHW)*N + n)*OHW + m%OHW]
+};
+
+// C[M,N] = A op B (+bias), fp32 accumulate, bf16 in/out.
+//   ta: A accessed as A[k*M + m] (stored [K,M]), else A[m*K + k]
+//   tb: B accessed as B[n*K + k] (stored [N,K]), else B[k*N + n]
+//   bias: optional [N] bf16 added per column; relu: fused max(0,.)
+//   ohw: only for EpStore::kConvNCHW
+void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
+                   torch::Tensor& C, long M, long N, long K, bool ta, bool tb,
+                   const torch::Tensor* bias, bool relu, EpStore store,
+                   long ohw);
+
+// colsum: out[n] = sum_m X[m,n]  (bias gradient)
+torch::Tensor colsum_bf16(const torch::Tensor& X);
+
+}  // namespace bflc

@@ -1,0 +1,208 @@
+"""GPU numerics tests: every gfx950 HIP kernel vs the plain fp32 torch
+oracle (computed on the same bf16-rounded inputs).
+
+Asymmetric random operands throughout — transpose-detecting
+(cdna_hip_programming.md §5.4 rule 16).
+"""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def hip():
+    from bflc_amd.ops import functional as fn
+    return fn.hip_ops()
+
+
+def bf(x):
+    return x.to(DEV, torch.bfloat16).contiguous()
+
+
+def assert_close(y, ref, rel=0.02):
+    y = y.float().cpu()
+    ref = ref.float().cpu()
+    scale = ref.abs().max().clamp_min(1.0)
+    torch.testing.assert_close(y, ref, rtol=rel, atol=float(scale) * rel)
+
+
+class TestLinear:
+    @pytest.mark.parametrize("m,k,n", [
+        (100, 5, 2),        # reference logreg shape (main.py:120)
+        (100, 3136, 128),   # FEMNIST fc1
+        (100, 128, 62),     # FEMNIST head
+        (37, 100, 62),      # odd sizes / predication
+        (256, 512, 512),    # square-ish
+        (2036, 5, 2),       # sponsor global test shape
+    ])
+    def test_fwd(self, m, k, n):
+        torch.manual_seed(0)
+        x = torch.randn(m, k)
+        w = torch.randn(k, n)
+        b = torch.randn(n)
+        y = hip().linear_fwd(bf(x), bf(w), bf(b))
+        ref = bf(x).float().cpu() @ bf(w).float().cpu() + bf(b).float().cpu()
+        assert_close(y, ref)
+
+    @pytest.mark.parametrize("m,k,n", [(100, 5, 2), (64, 3136, 128),
+                                       (37, 100, 62), (128, 256, 192)])
+    def test_bwd(self, m, k, n):
+        torch.manual_seed(1)
+        x, w, dy = torch.randn(m, k), torch.randn(k, n), torch.randn(m, n)
+        dx, dw, db = hip().linear_bwd(bf(x), bf(w), bf(dy))
+        xf, wf, dyf = (bf(t).float().cpu() for t in (x, w, dy))
+        assert_close(dx, dyf @ wf.t())
+        assert_close(dw, xf.t() @ dyf)
+        assert_close(db, dyf.sum(0))
+
+
+class TestSoftmaxCE:
+    @pytest.mark.parametrize("m,c", [(100, 2), (256, 62), (64, 1000), (7, 10)])
+    def test_fwd_bwd(self, m, c):
+        torch.manual_seed(2)
+        logits = torch.randn(m, c) * 3
+        y = torch.randint(0, c, (m,))
+        loss, probs = hip().softmax_ce_fwd(bf(logits), y.to(DEV))
+        lf = bf(logits).float().cpu()
+        ref_loss = torch.nn.functional.cross_entropy(lf, y)
+        assert abs(float(loss) - float(ref_loss)) < 2e-3 * max(1, ref_loss.abs())
+        assert_close(probs, torch.softmax(lf, dim=1), rel=0.01)
+
+        g = torch.tensor(2.0)
+        dl = hip().softmax_ce_bwd(probs, y.to(DEV), g.to(DEV))
+        l2 = lf.clone().requires_grad_(True)
+        (torch.nn.functional.cross_entropy(l2, y) * 2.0).backward()
+        assert_close(dl, l2.grad, rel=0.03)
+
+
+class TestConv:
+    @pytest.mark.parametrize("n,c,h,k,r,stride,pad", [
+        (4, 1, 28, 32, 3, 1, 1),   # FEMNIST conv1
+        (4, 32, 14, 64, 3, 1, 1),  # FEMNIST conv2
+        (2, 3, 32, 16, 3, 1, 1),   # resnet20 stem
+        (2, 16, 32, 32, 3, 2, 1),  # strided downsample
+        (2, 8, 9, 8, 3, 1, 0),     # odd, no pad
+        (2, 8, 8, 16, 1, 1, 0),    # 1x1 conv
+    ])
+    def test_fwd_bwd(self, n, c, h, k, r, stride, pad):
+        torch.manual_seed(3)
+        x = torch.randn(n, c, h, h)
+        w = torch.randn(k, c, r, r) * (1.0 / math.sqrt(c * r * r))
+        b = torch.randn(k)
+        y = hip().conv2d_fwd(bf(x), bf(w), bf(b), stride, pad)
+        xf, wf, bfl = (bf(t).float().cpu() for t in (x, w, b))
+        ref = torch.nn.functional.conv2d(xf, wf, bfl, stride=stride,
+                                         padding=pad)
+        assert y.shape == ref.shape
+        assert_close(y, ref)
+
+        dy = torch.randn_like(ref)
+        dx, dw, db = hip().conv2d_bwd(bf(x), bf(w), bf(dy), stride, pad)
+        x2 = xf.clone().requires_grad_(True)
+        w2 = wf.clone().requires_grad_(True)
+        b2 = bfl.clone().requires_grad_(True)
+        out = torch.nn.functional.conv2d(x2, w2, b2, stride=stride,
+                                         padding=pad)
+        (out * bf(dy).float().cpu()).sum().backward()
+        assert_close(dx, x2.grad, rel=0.03)
+        assert_close(dw, w2.grad, rel=0.03)
+        assert_close(db, b2.grad, rel=0.03)
+
+
+class TestPoolReluAcc:
+    def test_maxpool(self):
+        torch.manual_seed(4)
+        x = torch.randn(3, 8, 14, 14)
+        y, idx = hip().maxpool2d_fwd(bf(x), 2, 2)
+        ref, ridx = torch.nn.functional.max_pool2d(
+            bf(x).float().cpu(), 2, 2, return_indices=True)
+        assert_close(y, ref, rel=0.01)
+        dy = torch.randn_like(ref)
+        dx = hip().maxpool2d_bwd(bf(dy), idx, [3, 8, 14, 14], 2, 2)
+        ref_dx = torch.nn.functional.max_unpool2d(
+            bf(dy).float().cpu(), ridx, 2, 2, output_size=(14, 14))
+        assert_close(dx, ref_dx, rel=0.01)
+
+    def test_relu(self):
+        x = torch.randn(1000)
+        y = hip().relu_fwd(bf(x))
+        assert torch.equal(y.cpu().float(), torch.relu(bf(x).cpu().float()))
+        dy = torch.randn(1000)
+        dx = hip().relu_bwd(y, bf(dy))
+        ref = bf(dy).cpu().float() * (bf(x).cpu().float() > 0)
+        assert_close(dx, ref, rel=0.01)
+
+    def test_accuracy(self):
+        torch.manual_seed(5)
+        logits = torch.randn(1000, 10)
+        y = torch.randint(0, 10, (1000,))
+        a = hip().accuracy(bf(logits), y.to(DEV))
+        ref = (bf(logits).float().cpu().argmax(1) == y).float().mean()
+        assert a == pytest.approx(float(ref), abs=1e-6)
+
+
+class TestFlatOps:
+    def test_axpy_sgd_exact(self):
+        y = torch.randn(100001, device=DEV)
+        x = torch.randn(100001, device=DEV)
+        y2 = y.clone()
+        hip().axpy_(y, x, 2.5)
+        assert torch.equal(y, y2 + 2.5 * x)
+
+        p = torch.randn(12345, device=DEV)
+        g = torch.randn(12345, device=DEV)
+        p2 = p.clone()
+        hip().sgd_step_(p, g, 0.1)
+        ref = torch.tensor(0.1, device=DEV)  # fmaf(-lr, g, p)
+        assert torch.allclose(p, p2 - 0.1 * g, atol=1e-7)
+
+    def test_adam_matches_cpu(self):
+        from bflc_amd.ops import functional as O
+        torch.manual_seed(6)
+        p = torch.randn(5000)
+        g = torch.randn(5000)
+        m = torch.zeros(5000)
+        v = torch.zeros(5000)
+        pg, gg, mg, vg = (t.to(DEV) for t in (p, g, m, v))
+        for step in (1, 2, 3):
+            O.adam_step_(p, g, m, v, step, 0.01)
+            hip().adam_step_(pg, gg, mg, vg, step, 0.01, 0.9, 0.999, 1e-8)
+        assert torch.allclose(pg.cpu(), p, atol=1e-5)
+
+    def test_fedavg_matches_and_deterministic(self):
+        torch.manual_seed(7)
+        deltas = torch.randn(6, 200000, device=DEV)
+        w = torch.tensor([305., 300., 310., 290., 305., 295.], device=DEV)
+        a1 = hip().weighted_fedavg(deltas, w)
+        a2 = hip().weighted_fedavg(deltas, w)
+        assert torch.equal(a1, a2)  # bitwise determinism
+        ref = torch.zeros(200000)
+        dc = deltas.cpu()
+        for k in range(6):
+            ref += dc[k] * w[k].cpu()
+        ref /= w.sum().cpu()
+        assert torch.allclose(a1.cpu(), ref, atol=1e-4)
+
+
+class TestEndToEndGPU:
+    def test_femnist_round_and_determinism(self):
+        import __graft_entry__ as ge
+        ge.smoke()
+
+    def test_logreg_fl_learns_on_gpu(self):
+        from bflc_amd.config import FLConfig
+        from bflc_amd.comm import Transport
+        from bflc_amd.data import make_federated
+        from bflc_amd.fl import FLEngine
+        cfg = FLConfig()  # 20 clients, reference constants
+        shards, test = make_federated(cfg)
+        t = Transport(device=torch.device(DEV))
+        eng = FLEngine(cfg, t, shards, test)
+        stats = eng.run(8)
+        acc = eng.evaluate_global()
+        assert acc > 0.85
+        assert stats[-1].global_loss < stats[0].global_loss
