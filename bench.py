@@ -1,0 +1,120 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: committee-consensus FL rounds, FEMNIST 4-layer CNN
+(BASELINE.json metric: global-model test acc + wall-clock/FL-round at
+1/2/4/8 FL nodes = GPUs).
+
+One process per GPU (torchrun), one FL node per GPU, non-IID Dirichlet
+shards of synthetic FEMNIST-shaped data (28x28x1, 62 classes), random
+init, bf16 compute. A step = one full FL round (local train -> RCCL
+all-gather of updates -> committee scoring -> all-gather of scores ->
+weighted FedAvg commit). Weak scaling: per-GPU work fixed as N grows.
+
+Prints ONE JSON line on rank 0 (driver contract).
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", default="femnist_cnn")
+    ap.add_argument("--samples-per-client", type=int, default=3072)
+    ap.add_argument("--batch-size", type=int, default=1024)
+    ap.add_argument("--eval", action="store_true", help="eval every round")
+    args = ap.parse_args()
+
+    from bflc_amd.config import FLConfig
+    from bflc_amd.comm import Transport
+    from bflc_amd.data import make_federated
+    from bflc_amd.fl import FLEngine
+
+    n = int(os.environ.get("WORLD_SIZE", args.gpus))
+    cfg = FLConfig.for_world(
+        n, model=args.model, n_class=62,
+        samples_per_client=args.samples_per_client,
+        batch_size=args.batch_size, partition="dirichlet",
+        dirichlet_alpha=0.3, eval_samples=4096, learning_rate=0.01)
+
+    t = Transport()
+    shards, test = make_federated(cfg)
+    eng = FLEngine(cfg, t, shards, test)
+
+    use_cuda = t.device.type == "cuda"
+
+    # ---- warmup (untimed) ----
+    for _ in range(args.warmup):
+        eng.run_round()
+    t.barrier()
+    if use_cuda:
+        torch.cuda.synchronize(t.device)
+
+    # ---- timed: exactly --steps FL rounds ----
+    t0 = time.perf_counter()
+    samples = 0
+    for _ in range(args.steps):
+        st = eng.run_round()
+    t.barrier()
+    if use_cuda:
+        torch.cuda.synchronize(t.device)
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if t.is_distributed:
+        e = torch.tensor([elapsed])
+        if t.backend == "nccl":
+            e = e.to(t.device)
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(e.cpu().item())
+
+    # whole-job samples trained per round (deterministic across ranks)
+    per_round = cfg.needed_update_count * cfg.samples_per_client \
+        * cfg.local_epochs
+    total_samples = per_round * args.steps
+    acc = eng.evaluate_global()
+
+    if t.rank == 0:
+        out = {
+            "metric": "fl_train_samples_per_s",
+            "value": total_samples / elapsed,
+            "unit": "samples/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no perf numbers
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic",
+            "test_acc": acc,
+            "global_loss": eng.ledger.global_loss,
+            "fl_epoch": eng.ledger.epoch,
+            "config": {
+                "model": cfg.model,
+                "global_batch": cfg.batch_size * cfg.needed_update_count,
+                "seq_len": None,
+                "img": "1x28x28x62cls",
+                "parallelism": f"fl_nodes{n}",
+                "clients": cfg.client_num,
+                "committee": cfg.comm_count,
+                "update_quota": cfg.needed_update_count,
+                "aggregate_top_k": cfg.aggregate_count,
+                "samples_per_client": cfg.samples_per_client,
+                "batch_size": cfg.batch_size,
+                "partition": cfg.partition,
+            },
+        }
+        print(json.dumps(out), flush=True)
+    t.barrier()
+    t.close()
+
+
+if __name__ == "__main__":
+    main()
