@@ -28,6 +28,8 @@ def main() -> None:
     ap.add_argument("--samples-per-client", type=int, default=3072)
     ap.add_argument("--batch-size", type=int, default=1024)
     ap.add_argument("--eval", action="store_true", help="eval every round")
+    ap.add_argument("--phases", action="store_true",
+                    help="print per-phase timings of each timed round")
     args = ap.parse_args()
 
     from bflc_amd.config import FLConfig
@@ -57,9 +59,13 @@ def main() -> None:
 
     # ---- timed: exactly --steps FL rounds ----
     t0 = time.perf_counter()
-    samples = 0
     for _ in range(args.steps):
         st = eng.run_round()
+        if args.phases and t.rank == 0:
+            print(f"# round {st.epoch}: wall {st.wall_s*1e3:.1f}ms "
+                  f"train {st.train_s*1e3:.1f} gather {st.gather_s*1e3:.1f} "
+                  f"score {st.score_s*1e3:.1f} agg {st.aggregate_s*1e3:.1f}",
+                  flush=True)
     t.barrier()
     if use_cuda:
         torch.cuda.synchronize(t.device)

@@ -127,7 +127,8 @@ class FLEngine:
         n = shard.n
         bs = min(cfg.batch_size, n)
         total_batches = max(n // bs, 1)
-        avg_cost = 0.0
+        # accumulate the loss on-device; ONE host sync per client per round
+        cost_accum = torch.zeros((), device=self.device)
         if cfg.optimizer == "adam":
             m = torch.zeros_like(self.global_flat)
             v = torch.zeros_like(self.global_flat)
@@ -146,8 +147,8 @@ class FLEngine:
                 else:
                     O.sgd_step_(self.model.flat.data, self.model.flat.grad,
                                 cfg.learning_rate)
-                avg_cost += float(loss.detach()) / (total_batches
-                                                    * cfg.local_epochs)
+                cost_accum += loss.detach()
+        avg_cost = float(cost_accum) / (total_batches * cfg.local_epochs)
         # delta = (W0 - W)/lr  (reference main.py:153-154)
         delta = self.global_flat.clone()
         O.axpy_(delta, -1.0, self.model.flat.data)
