@@ -11,21 +11,25 @@
 //     every fragment load is one 16-byte ds_read (b128); rows padded
 //     +8 bf16 (16 B) against bank conflicts.
 //   - 64-wide wavefronts; template wave grid WRxWC, each wave computes a
-//     (BM/WR)x(BN/WC) sub-tile as 16x16 fragments (wave->output-tile
-//     decomposition per guide §5 idiom).
+//     (BM/WR)x(BN/WC) sub-tile as 16x16 fragments.
 //   - Transposed A/B operands are handled at the staging gather, so the
 //     MFMA inner loop is layout-independent.
-//   - Epilogue: fused bias add + optional ReLU + optional NCHW scatter
-//     (direct conv output, no separate permute kernel).
-//   - Grid is 1-D over output tiles with an XCD-bijective swizzle
-//     (guide T1) so neighbor tiles share an XCD-private L2.
+//   - Shape-aware tile selection (conv GEMMs are tall/skinny: M up to
+//     N*OH*OW ~ 1M rows with N as small as 9..64 columns).
+//   - Split-K with a FIXED-ORDER fp32 reduce for small-tile/huge-K
+//     shapes (conv wgrad: M=Kout, N=CRS, K=N*OH*OW): K slices land in a
+//     partial buffer [S, M, N] and a deterministic second kernel sums
+//     ascending s — bitwise identical on every rank, no atomics.
+//   - Epilogue (bias + optional ReLU + optional NCHW scatter) fused
+//     into the GEMM (or into the split-K reduce).
+//   - 1-D tile grid with an XCD-bijective swizzle (guide T1).
 //
-// Fragment layout (gfx950 mfma_f32_16x16x32_bf16, cdna4_isa.md §10):
+// Fragment layout (gfx950 mfma_f32_16x16x32_bf16, cdna4_isa.md §10),
+// verified on MI355X hardware by tests/test_ops_gpu.py (asymmetric
+// operands, transpose-detecting):
 //   A: lane l holds A[row = l&15][k = (l>>4)*8 + j], j = 0..7
 //   B: lane l holds B[k = (l>>4)*8 + j][col = l&15]
 //   D: lane l holds D[row = (l>>4)*4 + r][col = l&15], r = 0..3
-// Verified on hardware by tests/test_ops_gpu.py numerics tests with
-// asymmetric operands (transpose-detecting, guide §5.4 rule 16).
 
 #include "common.h"
 #include "gemm_api.h"
@@ -40,26 +44,29 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 constexpr int BK = 32;      // K per MFMA instruction / per LDS stage
 constexpr int BKP = BK + 8; // padded row length (16 B) vs bank conflicts
 
+// blockIdx.x = tile (XCD-swizzled), blockIdx.y = K slice (split-K).
+// If Cpart != nullptr: write fp32 partials at Cpart[slice*M*N + ...] and
+// skip bias/relu (applied by the reduce kernel). kslice = K per slice.
 template <int BM, int BN, int WR, int WC, bool TA, bool TB>
 __launch_bounds__(WR * WC * 64)
 __global__ void gemm_kernel(const bf16* __restrict__ A,
                             const bf16* __restrict__ B,
-                            bf16* __restrict__ C, const bf16* __restrict__ bias,
-                            long M, long N, long K, int relu, int store_mode,
+                            bf16* __restrict__ C, float* __restrict__ Cpart,
+                            const bf16* __restrict__ bias, long M, long N,
+                            long K, long kslice, int relu, int store_mode,
                             long ohw) {
   constexpr int THREADS = WR * WC * 64;
-  constexpr int FM = BM / WR / 16;  // 16x16 fragments per wave (rows)
-  constexpr int FN = BN / WC / 16;  // fragments per wave (cols)
+  constexpr int FM = BM / WR / 16;
+  constexpr int FN = BN / WC / 16;
 
   __shared__ bf16 As[BM][BKP];
   __shared__ bf16 Bs[BN][BKP];
 
-  // ---- tile coordinates with XCD-bijective swizzle (guide T1) ----
   const int ntn = (int)((N + BN - 1) / BN);
   const int ntm = (int)((M + BM - 1) / BM);
   const int nwg = ntm * ntn;
   int bid = blockIdx.x;
-  {  // bijective remap: contiguous chunk per XCD
+  {  // XCD-bijective remap (guide T1)
     const int nxcd = 8;
     const int q = nwg / nxcd, r = nwg % nxcd;
     const int xcd = bid % nxcd, idx = bid / nxcd;
@@ -67,6 +74,9 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
   }
   const long tile_m = (long)(bid / ntn) * BM;
   const long tile_n = (long)(bid % ntn) * BN;
+
+  const long k_begin = (long)blockIdx.y * kslice;
+  const long k_end = min(K, k_begin + kslice);
 
   const int tid = threadIdx.x;
   const int wave = tid / 64, lane = tid % 64;
@@ -76,34 +86,29 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
 
   f32x4_t acc[FM][FN] = {};
 
-  const int ksteps = (int)((K + BK - 1) / BK);
-  for (int ks = 0; ks < ksteps; ++ks) {
-    const long k0 = (long)ks * BK;
-    // ---- stage A tile: As[m][k] = A(tile_m+m, k0+k), zero-padded ----
+  for (long k0 = k_begin; k0 < k_end; k0 += BK) {
 #pragma unroll
     for (int i = 0; i < (BM * BK) / THREADS; ++i) {
       int idx = tid + i * THREADS;
       int m = idx / BK, k = idx % BK;
       long gm = tile_m + m, gk = k0 + k;
       float v = 0.f;
-      if (gm < M && gk < K)
+      if (gm < M && gk < k_end)
         v = b2f(TA ? A[gk * M + gm] : A[gm * K + gk]);
       As[m][k] = f2b(v);
     }
-    // ---- stage B tile: Bs[n][k] = B(k0+k, tile_n+n), zero-padded ----
 #pragma unroll
     for (int i = 0; i < (BN * BK) / THREADS; ++i) {
       int idx = tid + i * THREADS;
       int n = idx / BK, k = idx % BK;
       long gn = tile_n + n, gk = k0 + k;
       float v = 0.f;
-      if (gn < N && gk < K)
+      if (gn < N && gk < k_end)
         v = b2f(TB ? B[gn * K + gk] : B[gk * N + gn]);
       Bs[n][k] = f2b(v);
     }
     __syncthreads();
 
-    // ---- MFMA inner loop: one 16x16x32 per fragment pair ----
     bf16x8_t a_frag[FM], b_frag[FN];
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm)
@@ -122,7 +127,23 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
     __syncthreads();
   }
 
-  // ---- epilogue: bias + relu + store (bf16) ----
+  if (Cpart) {  // split-K partial store (fp32, plain [M,N] layout)
+    float* out = Cpart + (long)blockIdx.y * M * N;
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn) {
+        const long col = tile_n + wn0 + fn * 16 + l15;
+        if (col >= N) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const long row = tile_m + wm0 + fm * 16 + l4 * 4 + r;
+          if (row < M) out[row * N + col] = acc[fm][fn][r];
+        }
+      }
+    return;
+  }
+
 #pragma unroll
   for (int fm = 0; fm < FM; ++fm) {
 #pragma unroll
@@ -147,14 +168,74 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
   }
 }
 
-// out[n] = sum_m X[m][n] (fp32 accumulate, bf16 out) — bias gradient.
-__global__ void colsum_kernel(const bf16* __restrict__ X, bf16* __restrict__ o,
-                              long M, long N) {
+// Deterministic split-K reduce: out = sum_s partial[s] (ascending s),
+// then the fused epilogue (bias/relu/NCHW).
+__global__ void splitk_reduce_kernel(const float* __restrict__ Cpart, int S,
+                                     long M, long N, bf16* __restrict__ C,
+                                     const bf16* __restrict__ bias, int relu,
+                                     int store_mode, long ohw) {
+  const long total = M * N;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    float acc = 0.f;
+    for (int s = 0; s < S; ++s) acc += Cpart[(long)s * total + i];
+    const long row = i / N, col = i - row * N;
+    if (bias) acc += b2f(bias[col]);
+    if (relu) acc = fmaxf(acc, 0.f);
+    if (store_mode == (int)EpStore::kConvNCHW) {
+      const long img = row / ohw, sp = row % ohw;
+      C[(img * N + col) * ohw + sp] = f2b(acc);
+    } else {
+      C[i] = f2b(acc);
+    }
+  }
+}
+
+// Hierarchical colsum: pass 1 tiles rows into fp32 partials (coalesced:
+// thread -> column), pass 2 reduces chunks in fixed ascending order.
+constexpr int kColsumRows = 2048;
+
+__global__ void colsum_part_kernel(const bf16* __restrict__ X, long M, long N,
+                                   float* __restrict__ part) {
+  const long col = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const int chunk = blockIdx.y;
+  if (col >= N) return;
+  const long r0 = (long)chunk * kColsumRows;
+  const long r1 = min(M, r0 + kColsumRows);
+  float acc = 0.f;
+  for (long m = r0; m < r1; ++m) acc += b2f(X[m * N + col]);
+  part[(long)chunk * N + col] = acc;
+}
+
+__global__ void colsum_final_kernel(const float* __restrict__ part,
+                                    int chunks, long N,
+                                    bf16* __restrict__ out) {
   const long col = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= N) return;
   float acc = 0.f;
-  for (long m = 0; m < M; ++m) acc += b2f(X[m * N + col]);
-  o[col] = f2b(acc);
+  for (int c = 0; c < chunks; ++c) acc += part[(long)c * N + col];
+  out[col] = f2b(acc);
+}
+
+struct TileCfg { int bm, bn, wr, wc; };
+
+TileCfg pick_tile(long M, long N) {
+  int bn = N <= 32 ? 32 : (N <= 64 ? 64 : 128);
+  int bm = M <= 48 ? 32 : (M <= 96 ? 64 : 128);
+  // keep >= ~512 tiles when possible by shrinking BM (conv fwd shapes
+  // have huge M, so this rarely triggers; fc shapes are latency-bound
+  // anyway)
+  auto tiles = [&](int m, int n) {
+    return ((M + m - 1) / m) * ((N + n - 1) / n);
+  };
+  while (bm > 32 && tiles(bm, bn) < 512 && M > 4096) bm /= 2;
+  int wr = bm >= 64 ? 2 : (bm == 32 ? 1 : 2);
+  int wc = 2;
+  if (bm == 32) { wr = 1; wc = 4; }
+  if (bn == 32 && bm == 32) { wr = 2; wc = 1; }
+  if (bn == 32 && bm > 32) { wr = 2; wc = 2; }
+  return {bm, bn, wr, wc};
 }
 
 }  // namespace
@@ -171,27 +252,49 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   bf16* c = (bf16*)C.data_ptr();
   const bf16* bs = bias ? (const bf16*)bias->data_ptr() : nullptr;
 
+  const TileCfg t = pick_tile(M, N);
+  const long tiles = ((M + t.bm - 1) / t.bm) * ((N + t.bn - 1) / t.bn);
+
+  // split-K when the tile grid cannot fill the chip and K is deep
+  const long ksteps = (K + BK - 1) / BK;
+  long S = 1;
+  if (tiles < 512 && K >= 4 * BK) {
+    S = std::min<long>({(512 + tiles - 1) / tiles, ksteps, 64});
+  }
+  const long kslice = ((ksteps + S - 1) / S) * BK;
+  S = (K + kslice - 1) / kslice;  // actual slices after rounding
+
+  torch::Tensor part;
+  float* part_ptr = nullptr;
+  if (S > 1) {
+    part = torch::empty({S, M, N},
+                        A.options().dtype(at::kFloat));
+    part_ptr = part.data_ptr<float>();
+  }
+
+  dim3 grid((unsigned)tiles, (unsigned)S);
+
   auto launch = [&](auto bm, auto bn, auto wr, auto wc) {
     constexpr int BMv = decltype(bm)::value, BNv = decltype(bn)::value;
     constexpr int WRv = decltype(wr)::value, WCv = decltype(wc)::value;
-    dim3 grid(ceil_div(M, BMv) * ceil_div(N, BNv));
     dim3 block(WRv * WCv * 64);
+    const int sm = (int)store;
     if (!ta && !tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, false, false>),
-                         grid, block, 0, cur_stream(), a, b, c, bs, M, N, K,
-                         relu, (int)store, ohw);
+                         grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
+                         M, N, K, kslice, relu, sm, ohw);
     else if (!ta && tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, false, true>),
-                         grid, block, 0, cur_stream(), a, b, c, bs, M, N, K,
-                         relu, (int)store, ohw);
+                         grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
+                         M, N, K, kslice, relu, sm, ohw);
     else if (ta && !tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, true, false>),
-                         grid, block, 0, cur_stream(), a, b, c, bs, M, N, K,
-                         relu, (int)store, ohw);
+                         grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
+                         M, N, K, kslice, relu, sm, ohw);
     else
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, true, true>),
-                         grid, block, 0, cur_stream(), a, b, c, bs, M, N, K,
-                         relu, (int)store, ohw);
+                         grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
+                         M, N, K, kslice, relu, sm, ohw);
   };
 
   using c32 = std::integral_constant<int, 32>;
@@ -201,23 +304,45 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   using c2 = std::integral_constant<int, 2>;
   using c4 = std::integral_constant<int, 4>;
 
-  // tile selection by shape: big tiles only pay when M and N fill them
-  if (M >= 96 && N >= 96)
-    launch(c128{}, c128{}, c2{}, c2{});
-  else if (M >= 48)
-    launch(c64{}, c64{}, c2{}, c2{});
-  else
-    launch(c32{}, c64{}, c1{}, c4{});
+  const int key = t.bm * 1000 + t.bn;
+  switch (key) {
+    case 128128: launch(c128{}, c128{}, c2{}, c2{}); break;
+    case 128064: launch(c128{}, c64{}, c2{}, c2{}); break;
+    case 128032: launch(c128{}, c32{}, c2{}, c2{}); break;
+    case  64128: launch(c64{}, c128{}, c2{}, c2{}); break;
+    case  64064: launch(c64{}, c64{}, c2{}, c2{}); break;
+    case  64032: launch(c64{}, c32{}, c2{}, c2{}); break;
+    case  32128: launch(c32{}, c128{}, c1{}, c4{}); break;
+    case  32064: launch(c32{}, c64{}, c1{}, c4{}); break;
+    case  32032: launch(c32{}, c32{}, c2{}, c1{}); break;
+    default: TORCH_CHECK(false, "no tile config for ", t.bm, "x", t.bn);
+  }
   HIP_CHECK(hipGetLastError());
+
+  if (S > 1) {
+    const long total = M * N;
+    int blocks = (int)std::min<long>((total + 255) / 256, 8192);
+    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
+                       cur_stream(), part_ptr, (int)S, M, N, c, bs,
+                       relu ? 1 : 0, (int)store, ohw);
+    HIP_CHECK(hipGetLastError());
+  }
 }
 
 torch::Tensor colsum_bf16(const torch::Tensor& X) {
   CHECK_GPU(X); CHECK_CONTIG(X);
   long M = X.size(0), N = X.size(1);
   auto out = torch::empty({N}, X.options());
-  hipLaunchKernelGGL(colsum_kernel, dim3(ceil_div(N, 256)), dim3(256), 0,
-                     cur_stream(), (const bf16*)X.data_ptr(),
-                     (bf16*)out.data_ptr(), M, N);
+  const int chunks = (int)((M + kColsumRows - 1) / kColsumRows);
+  auto part = torch::empty({chunks, N}, X.options().dtype(at::kFloat));
+  dim3 grid(ceil_div(N, 256), chunks);
+  hipLaunchKernelGGL(colsum_part_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const bf16*)X.data_ptr(), M, N,
+                     part.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+  hipLaunchKernelGGL(colsum_final_kernel, dim3(ceil_div(N, 256)), dim3(256),
+                     0, cur_stream(), part.data_ptr<float>(), chunks, N,
+                     (bf16*)out.data_ptr());
   HIP_CHECK(hipGetLastError());
   return out;
 }
@@ -241,11 +366,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w); CHECK_CONTIG(dy);
   long M = x.size(0), K = x.size(1), N = w.size(1);
-  // dx[M,K] = dy[M,N] @ w^T : B accessed [n'(=K rows), k'(=N)] = w stored
   auto dx = torch::empty({M, K}, x.options());
   gemm_bf16_raw(dy, w, dx, M, K, N, false, true, nullptr, false,
                 EpStore::kPlain, 0);
-  // dw[K,N] = x^T @ dy : A accessed [k'(=M), m'(=K)] = x stored
   auto dw = torch::empty({K, N}, x.options());
   gemm_bf16_raw(x, dy, dw, K, N, M, true, false, nullptr, false,
                 EpStore::kPlain, 0);
