@@ -74,11 +74,11 @@ def _torch_paths() -> tuple[list[str], list[str], list[str]]:
 
 
 HIP_SOURCES = [
-    "ops_common.hip",
     "elementwise.hip",
     "softmax_ce.hip",
     "gemm_bf16.hip",
     "conv_im2col.hip",
+    "batchnorm.hip",
     "reduce.hip",
     "bindings.cpp",
 ]

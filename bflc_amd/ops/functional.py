@@ -209,6 +209,105 @@ def maxpool2d(x, kernel: int = 2, stride: Optional[int] = None) -> torch.Tensor:
     return _MaxPool2dFn.apply(x, kernel, stride or kernel)
 
 
+class _BatchNormFn(torch.autograd.Function):
+    """BatchNorm2d in batch-stats mode (train AND eval — no running
+    buffers; the FedBN-style simplification so the flat parameter vector
+    is exactly {gamma, beta}). GPU: hierarchical fixed-order reduction
+    kernels (csrc/hip/batchnorm.hip); CPU: fp32 oracle."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps: float):
+        if x.is_cuda:
+            y, mean, invstd = hip_ops().batchnorm_fwd(x, gamma, beta, eps,
+                                                      False)
+        else:
+            xf = x.float()
+            mean = xf.mean(dim=(0, 2, 3))
+            var = xf.var(dim=(0, 2, 3), unbiased=False)
+            invstd = (var + eps).rsqrt()
+            y = ((xf - mean[None, :, None, None])
+                 * invstd[None, :, None, None]
+                 * gamma.float()[None, :, None, None]
+                 + beta.float()[None, :, None, None]).to(x.dtype)
+        ctx.save_for_backward(x, gamma, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, invstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            dx, dgamma, dbeta = hip_ops().batchnorm_bwd(x, dy, mean, invstd,
+                                                        gamma)
+        else:
+            xf, dyf = x.float(), dy.float()
+            n = x.numel() / x.shape[1]
+            xhat = (xf - mean[None, :, None, None]) * invstd[None, :, None,
+                                                            None]
+            sdy = dyf.sum(dim=(0, 2, 3))
+            sdyx = (dyf * xhat).sum(dim=(0, 2, 3))
+            dx = (gamma.float() * invstd)[None, :, None, None] * (
+                dyf - sdy[None, :, None, None] / n
+                - xhat * sdyx[None, :, None, None] / n)
+            dx = dx.to(x.dtype)
+            dgamma = sdyx.to(x.dtype)
+            dbeta = sdy.to(x.dtype)
+        return dx, dgamma, dbeta, None
+
+
+def batchnorm2d(x, gamma, beta, eps: float = 1e-5) -> torch.Tensor:
+    return _BatchNormFn.apply(x, gamma, beta, eps)
+
+
+class _GlobalAvgPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.hw = (x.shape[2], x.shape[3])
+        if x.is_cuda:
+            return hip_ops().global_avgpool_fwd(x)
+        return x.mean(dim=(2, 3))
+
+    @staticmethod
+    def backward(ctx, dy):
+        h, w = ctx.hw
+        dy = dy.contiguous()
+        if dy.is_cuda:
+            return hip_ops().global_avgpool_bwd(dy, h, w)
+        return (dy / (h * w))[:, :, None, None].expand(-1, -1, h, w) \
+            .contiguous()
+
+
+def global_avgpool(x) -> torch.Tensor:
+    return _GlobalAvgPoolFn.apply(x)
+
+
+class _AddReluFn(torch.autograd.Function):
+    """Fused residual add + relu (one kernel instead of two)."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        if a.is_cuda:
+            y = hip_ops().add_relu_fwd(a, b.contiguous())
+        else:
+            y = torch.relu(a + b)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if y.is_cuda:
+            da = hip_ops().add_relu_bwd(y, dy)
+        else:
+            da = dy * (y > 0).to(dy.dtype)
+        return da, da
+
+
+def add_relu(a, b) -> torch.Tensor:
+    return _AddReluFn.apply(a, b)
+
+
 # ---------------------------------------------------------------------------
 # non-autograd FL math (flat-tensor ops)
 # ---------------------------------------------------------------------------

@@ -43,6 +43,18 @@ torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx,
                             std::vector<long> in_shape, long kernel,
                             long stride);
 
+// batchnorm.hip
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
+    torch::Tensor x, torch::Tensor gamma, torch::Tensor beta, double eps,
+    bool relu);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
+    torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
+    torch::Tensor invstd, torch::Tensor gamma);
+torch::Tensor global_avgpool_fwd(torch::Tensor x);
+torch::Tensor global_avgpool_bwd(torch::Tensor dy, long H, long W);
+torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b);
+torch::Tensor add_relu_bwd(torch::Tensor y, torch::Tensor dy);
+
 }  // namespace bflc
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -64,4 +76,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd", &bflc::conv2d_bwd, "(dx, dw, db)");
   m.def("maxpool2d_fwd", &bflc::maxpool2d_fwd);
   m.def("maxpool2d_bwd", &bflc::maxpool2d_bwd);
+  m.def("batchnorm_fwd", &bflc::batchnorm_fwd,
+        "(y, mean, invstd) — batch-stats BN, optional fused relu");
+  m.def("batchnorm_bwd", &bflc::batchnorm_bwd, "(dx, dgamma, dbeta)");
+  m.def("global_avgpool_fwd", &bflc::global_avgpool_fwd);
+  m.def("global_avgpool_bwd", &bflc::global_avgpool_bwd);
+  m.def("add_relu_fwd", &bflc::add_relu_fwd, "fused residual add + relu");
+  m.def("add_relu_bwd", &bflc::add_relu_bwd);
 }

@@ -121,7 +121,10 @@ __global__ void maxpool_fwd_kernel(const bf16* __restrict__ x,
   }
 }
 
-// stride >= kernel (non-overlapping): each input cell has one window.
+// Gather maxpool backward: for each input cell, visit every window that
+// covers it (up to ceil(k/s)^2 with overlapping windows, e.g. the
+// ResNet-50 stem's k=3 s=2) and take dy where the recorded argmax
+// matches. Deterministic fixed iteration order, no atomics.
 __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
                                    const long* __restrict__ idx,
                                    bf16* __restrict__ dx, int NC, int H,
@@ -132,13 +135,15 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
   for (; i < total; i += stride) {
     const int iw = (int)(i % W), ih = (int)((i / W) % H);
     const long nc = i / ((long)W * H);
-    const int oh = ih / st, ow = iw / st;
-    bf16 v = f2b(0.f);
-    if (oh < OH && ow < OW && ih - oh * st < kk && iw - ow * st < kk) {
-      const long o = (nc * OH + oh) * OW + ow;
-      if (idx[o] == (long)ih * W + iw) v = dy[o];
-    }
-    dx[i] = v;
+    const int oh_lo = max(0, (ih - kk + st) / st), oh_hi = min(OH - 1, ih / st);
+    const int ow_lo = max(0, (iw - kk + st) / st), ow_hi = min(OW - 1, iw / st);
+    float acc = 0.f;
+    for (int oh = oh_lo; oh <= oh_hi; ++oh)
+      for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+        const long o = (nc * OH + oh) * OW + ow;
+        if (idx[o] == (long)ih * W + iw) acc += b2f(dy[o]);
+      }
+    dx[i] = f2b(acc);
   }
 }
 
@@ -231,7 +236,6 @@ std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
                                                        long kernel,
                                                        long stride) {
   CHECK_GPU(x); CHECK_CONTIG(x);
-  TORCH_CHECK(stride >= kernel, "maxpool: non-overlapping windows only");
   int N = (int)x.size(0), C = (int)x.size(1);
   int H = (int)x.size(2), W = (int)x.size(3);
   int OH = (H - (int)kernel) / (int)stride + 1;

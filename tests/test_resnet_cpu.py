@@ -1,0 +1,97 @@
+"""ResNet model family tests (CPU oracle path)."""
+import pytest
+import torch
+import torch.nn.functional as F
+
+from bflc_amd.config import FLConfig
+from bflc_amd.models import build_model
+from bflc_amd.ops import functional as O
+
+
+class TestBatchNormOracle:
+    def test_fwd_bwd_matches_torch(self):
+        torch.manual_seed(0)
+        x = torch.randn(4, 8, 7, 7, requires_grad=True)
+        g = torch.randn(8, requires_grad=True)
+        b = torch.randn(8, requires_grad=True)
+        y = O.batchnorm2d(x, g, b)
+        ref = F.batch_norm(x, None, None, g, b, training=True, eps=1e-5)
+        assert torch.allclose(y, ref, atol=1e-5)
+        dy = torch.randn_like(y)
+        (y * dy).sum().backward()
+        x2 = x.detach().clone().requires_grad_(True)
+        g2 = g.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        (F.batch_norm(x2, None, None, g2, b2, training=True) * dy).sum() \
+            .backward()
+        assert torch.allclose(x.grad, x2.grad, atol=1e-4)
+        assert torch.allclose(g.grad, g2.grad, atol=1e-4)
+        assert torch.allclose(b.grad, b2.grad, atol=1e-4)
+
+    def test_gap_and_add_relu(self):
+        x = torch.randn(2, 4, 5, 5, requires_grad=True)
+        y = O.global_avgpool(x)
+        assert torch.allclose(y, x.mean(dim=(2, 3)))
+        y.sum().backward()
+        assert torch.allclose(x.grad, torch.full_like(x, 1 / 25.0))
+
+        a = torch.randn(100, requires_grad=True)
+        b = torch.randn(100, requires_grad=True)
+        z = O.add_relu(a, b)
+        assert torch.allclose(z, torch.relu(a + b))
+        z.sum().backward()
+        mask = (a + b > 0).float()
+        assert torch.allclose(a.grad, mask)
+        assert torch.allclose(b.grad, mask)
+
+
+class TestResNet20:
+    def test_forward_backward(self):
+        cfg = FLConfig(model="resnet20", n_class=10, client_num=1,
+                       comm_count=1, needed_update_count=1,
+                       aggregate_count=1)
+        m = build_model(cfg, torch.device("cpu"))
+        x = torch.randn(4, 3, 32, 32)
+        y = torch.randint(0, 10, (4,))
+        logits = m.forward(x)
+        assert logits.shape == (4, 10)
+        loss = m.loss(x, y)
+        loss.backward()
+        assert m.flat.grad is not None
+        assert float(m.flat.grad.abs().sum()) > 0
+        # param count sanity: resnet20 ~= 0.27M params
+        assert 0.25e6 < m.numel < 0.30e6
+
+    def test_loss_decreases_with_sgd(self):
+        cfg = FLConfig(model="resnet20", n_class=4, client_num=1,
+                       comm_count=1, needed_update_count=1,
+                       aggregate_count=1, learning_rate=0.05)
+        m = build_model(cfg, torch.device("cpu"))
+        torch.manual_seed(0)
+        x = torch.randn(16, 3, 32, 32)
+        y = torch.randint(0, 4, (16,))
+        losses = []
+        for _ in range(6):
+            m.zero_grad()
+            loss = m.loss(x, y)
+            loss.backward()
+            O.sgd_step_(m.flat.data, m.flat.grad, 0.05)
+            losses.append(float(loss))
+        assert losses[-1] < losses[0]
+
+
+class TestResNet50:
+    def test_forward_backward_small_input(self):
+        cfg = FLConfig(model="resnet50", n_class=10, client_num=1,
+                       comm_count=1, needed_update_count=1,
+                       aggregate_count=1)
+        m = build_model(cfg, torch.device("cpu"))
+        # 64x64 input keeps the CPU test fast; the conv/pool stack and
+        # global avgpool handle any spatial size
+        x = torch.randn(2, 3, 64, 64)
+        y = torch.randint(0, 10, (2,))
+        loss = m.loss(x, y)
+        loss.backward()
+        assert float(m.flat.grad.abs().sum()) > 0
+        # ~25.5M params at 1000 classes; ~23.5M at 10
+        assert 20e6 < m.numel < 27e6

@@ -1,0 +1,133 @@
+"""GPU numerics for BN/pool/add_relu kernels + ResNet end-to-end."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def hip():
+    from bflc_amd.ops import functional as fn
+    return fn.hip_ops()
+
+
+def bf(x):
+    return x.to(DEV, torch.bfloat16).contiguous()
+
+
+def assert_close(y, ref, rel=0.03):
+    y = y.float().cpu()
+    ref = ref.float().cpu()
+    scale = ref.abs().max().clamp_min(1.0)
+    torch.testing.assert_close(y, ref, rtol=rel, atol=float(scale) * rel)
+
+
+class TestBatchNormGPU:
+    @pytest.mark.parametrize("n,c,h", [(4, 8, 7), (16, 64, 8), (2, 3, 32)])
+    def test_fwd_bwd(self, n, c, h):
+        torch.manual_seed(0)
+        x = torch.randn(n, c, h, h)
+        g = torch.rand(c) + 0.5
+        b = torch.randn(c)
+        y, mean, invstd = hip().batchnorm_fwd(bf(x), bf(g), bf(b), 1e-5,
+                                              False)
+        xf = bf(x).float().cpu()
+        gf, bfl = bf(g).float().cpu(), bf(b).float().cpu()
+        ref = torch.nn.functional.batch_norm(xf, None, None, gf, bfl,
+                                             training=True, eps=1e-5)
+        assert_close(y, ref)
+
+        dy = torch.randn_like(ref)
+        dx, dgamma, dbeta = hip().batchnorm_bwd(bf(x), bf(dy), mean, invstd,
+                                                bf(g))
+        x2 = xf.clone().requires_grad_(True)
+        g2 = gf.clone().requires_grad_(True)
+        b2 = bfl.clone().requires_grad_(True)
+        out = torch.nn.functional.batch_norm(x2, None, None, g2, b2,
+                                             training=True, eps=1e-5)
+        (out * bf(dy).float().cpu()).sum().backward()
+        assert_close(dx, x2.grad, rel=0.05)
+        assert_close(dgamma, g2.grad, rel=0.05)
+        assert_close(dbeta, b2.grad, rel=0.05)
+
+    def test_fwd_deterministic(self):
+        x = bf(torch.randn(8, 16, 14, 14))
+        g, b = bf(torch.rand(16)), bf(torch.randn(16))
+        y1, m1, i1 = hip().batchnorm_fwd(x, g, b, 1e-5, False)
+        y2, m2, i2 = hip().batchnorm_fwd(x, g, b, 1e-5, False)
+        assert torch.equal(y1, y2) and torch.equal(m1, m2)
+
+
+class TestPoolAddRelu:
+    def test_gap(self):
+        x = torch.randn(3, 5, 9, 9)
+        y = hip().global_avgpool_fwd(bf(x))
+        assert_close(y, bf(x).float().cpu().mean(dim=(2, 3)), rel=0.01)
+        dy = torch.randn(3, 5)
+        dx = hip().global_avgpool_bwd(bf(dy), 9, 9)
+        ref = (bf(dy).float().cpu() / 81)[:, :, None, None] \
+            .expand(-1, -1, 9, 9)
+        assert_close(dx, ref, rel=0.01)
+
+    def test_add_relu(self):
+        a, b = torch.randn(5000), torch.randn(5000)
+        y = hip().add_relu_fwd(bf(a), bf(b))
+        ref = torch.relu(bf(a).float().cpu() + bf(b).float().cpu())
+        assert_close(y, ref, rel=0.01)
+        dy = torch.randn(5000)
+        da = hip().add_relu_bwd(y, bf(dy))
+        refd = bf(dy).float().cpu() * (ref > 0)
+        assert_close(da, refd, rel=0.01)
+
+    def test_overlapping_maxpool(self):
+        # ResNet-50 stem: k=3 s=2 overlapping windows
+        torch.manual_seed(1)
+        x = torch.randn(2, 4, 13, 13)
+        y, idx = hip().maxpool2d_fwd(bf(x), 3, 2)
+        ref = torch.nn.functional.max_pool2d(bf(x).float().cpu(), 3, 2)
+        assert_close(y, ref, rel=0.01)
+        dy = torch.randn_like(ref)
+        dx = hip().maxpool2d_bwd(bf(dy), idx, [2, 4, 13, 13], 3, 2)
+        x2 = bf(x).float().cpu().requires_grad_(True)
+        (torch.nn.functional.max_pool2d(x2, 3, 2) *
+         bf(dy).float().cpu()).sum().backward()
+        assert_close(dx, x2.grad, rel=0.01)
+
+
+class TestResNetGPU:
+    @pytest.mark.parametrize("model,hw,nclass", [("resnet20", 32, 10),
+                                                 ("resnet50", 64, 10)])
+    def test_train_step_runs_and_learns(self, model, hw, nclass):
+        from bflc_amd.config import FLConfig
+        from bflc_amd.models import build_model
+        from bflc_amd.ops import functional as O
+        cfg = FLConfig(model=model, n_class=nclass, client_num=1,
+                       comm_count=1, needed_update_count=1,
+                       aggregate_count=1, learning_rate=0.05)
+        m = build_model(cfg, torch.device(DEV))
+        torch.manual_seed(0)
+        x = torch.randn(16, 3, hw, hw)
+        y = torch.randint(0, nclass, (16,), device=DEV)
+        losses = []
+        for _ in range(5):
+            m.zero_grad()
+            loss = m.loss(x, y)
+            loss.backward()
+            O.sgd_step_(m.flat.data, m.flat.grad, cfg.learning_rate)
+            losses.append(float(loss))
+        assert losses[-1] < losses[0]
+
+    def test_resnet20_fl_round(self):
+        from bflc_amd.config import FLConfig
+        from bflc_amd.comm import Transport
+        from bflc_amd.data import make_federated
+        from bflc_amd.fl import FLEngine
+        cfg = FLConfig.for_world(1, model="resnet20", n_class=10,
+                                 samples_per_client=256, batch_size=128,
+                                 eval_samples=128)
+        shards, test = make_federated(cfg)
+        eng = FLEngine(cfg, Transport(device=torch.device(DEV)), shards,
+                       test)
+        st = eng.run_round(eval_global=True)
+        assert st.epoch == 0 and st.test_acc is not None
