@@ -70,7 +70,8 @@ class RoundStats:
 class FLEngine:
     def __init__(self, cfg: FLConfig, transport: Transport,
                  shards: List[Shard], test_shard: Optional[Shard] = None,
-                 ) -> None:
+                 metrics_path: Optional[str] = None) -> None:
+        from bflc_amd.utils import JsonlLogger
         self.cfg = cfg
         self.t = transport
         self.device = transport.device
@@ -100,6 +101,7 @@ class FLEngine:
         self.ledger.set_global_model(b"")
 
         self._round = 0
+        self.metrics = JsonlLogger(metrics_path, rank=self.rank)
 
     # ------------------------------------------------------------------
     def _planned_submitters(self) -> List[int]:
@@ -258,11 +260,14 @@ class FLEngine:
         if eval_global and self.test_shard is not None:
             acc = self.evaluate_global()
         self._round += 1
-        return RoundStats(
+        stats = RoundStats(
             epoch=epoch, wall_s=t4 - t0, train_s=t1 - t0, gather_s=t2 - t1,
             score_s=t3 - t2, aggregate_s=t4 - t3,
             global_loss=led.global_loss, n_updates=len(updates),
             n_selected=K, samples_trained=samples_trained, test_acc=acc)
+        self.metrics.log("round", stats,
+                         selected=[o for o, _ in decision.selected])
+        return stats
 
     # ------------------------------------------------------------------
     def evaluate_global(self) -> float:

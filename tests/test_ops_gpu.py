@@ -88,6 +88,7 @@ class TestConv:
         (2, 16, 32, 32, 3, 2, 1),  # strided downsample
         (2, 8, 9, 8, 3, 1, 0),     # odd, no pad
         (2, 8, 8, 16, 1, 1, 0),    # 1x1 conv
+        (2, 3, 32, 16, 7, 2, 3),   # resnet50 stem 7x7/2
     ])
     def test_fwd_bwd(self, n, c, h, k, r, stride, pad):
         torch.manual_seed(3)

@@ -96,15 +96,18 @@ class TestPoolAddRelu:
 
 
 class TestResNetGPU:
-    @pytest.mark.parametrize("model,hw,nclass", [("resnet20", 32, 10),
-                                                 ("resnet50", 64, 10)])
-    def test_train_step_runs_and_learns(self, model, hw, nclass):
+    @pytest.mark.parametrize("model,hw,nclass,lr", [
+        ("resnet20", 32, 10, 0.05),
+        # lr 0.05 diverges on the fp32 CPU oracle too — use 0.01
+        ("resnet50", 64, 10, 0.01),
+    ])
+    def test_train_step_runs_and_learns(self, model, hw, nclass, lr):
         from bflc_amd.config import FLConfig
         from bflc_amd.models import build_model
         from bflc_amd.ops import functional as O
         cfg = FLConfig(model=model, n_class=nclass, client_num=1,
                        comm_count=1, needed_update_count=1,
-                       aggregate_count=1, learning_rate=0.05)
+                       aggregate_count=1, learning_rate=lr)
         m = build_model(cfg, torch.device(DEV))
         torch.manual_seed(0)
         x = torch.randn(16, 3, hw, hw)
