@@ -6,9 +6,13 @@ GEMMs) plus square reference shapes. Within-process interleaved A/B
 (guide §5.4 rule 24): alternate ours/torch per round, report medians.
 """
 import argparse
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def time_fn(fn, iters=20, warmup=5):
