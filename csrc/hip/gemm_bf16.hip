@@ -39,6 +39,7 @@ namespace bflc {
 namespace {
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+typedef __attribute__((ext_vector_type(8))) unsigned short u16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 constexpr int BK = 32;      // K per MFMA instruction / per LDS stage
@@ -54,7 +55,7 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
                             bf16* __restrict__ C, float* __restrict__ Cpart,
                             const bf16* __restrict__ bias, long M, long N,
                             long K, long kslice, int relu, int store_mode,
-                            long ohw) {
+                            long ohw, int vecA, int vecB) {
   constexpr int THREADS = WR * WC * 64;
   constexpr int FM = BM / WR / 16;
   constexpr int FN = BN / WC / 16;
@@ -86,26 +87,99 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
 
   f32x4_t acc[FM][FN] = {};
 
+  const bf16 zero = f2b(0.f);
   for (long k0 = k_begin; k0 < k_end; k0 += BK) {
+    // ---- stage A tile (16-byte vector path when layout permits) ----
+    if (vecA) {
+      constexpr int GROUPS = (BM * BK) / 8;
 #pragma unroll
-    for (int i = 0; i < (BM * BK) / THREADS; ++i) {
-      int idx = tid + i * THREADS;
-      int m = idx / BK, k = idx % BK;
-      long gm = tile_m + m, gk = k0 + k;
-      float v = 0.f;
-      if (gm < M && gk < k_end)
-        v = b2f(TA ? A[gk * M + gm] : A[gm * K + gk]);
-      As[m][k] = f2b(v);
+      for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
+        const int g = tid + i * THREADS;
+        if (GROUPS % THREADS != 0 && g >= GROUPS) break;
+        if (!TA) {  // A[M][K]: 8 consecutive k per thread
+          const int m = g / (BK / 8), k8 = (g % (BK / 8)) * 8;
+          const long gm = tile_m + m, gk = k0 + k8;
+          if (gm < M && gk + 8 <= k_end) {
+            *reinterpret_cast<bf16x8_t*>(&As[m][k8]) =
+                *reinterpret_cast<const bf16x8_t*>(&A[gm * K + gk]);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              As[m][k8 + j] = (gm < M && gk + j < k_end)
+                                  ? A[gm * K + gk + j] : zero;
+          }
+        } else {    // A[K][M]: 8 consecutive m (same k) per thread
+          const int k = g / (BM / 8), m8 = (g % (BM / 8)) * 8;
+          const long gm = tile_m + m8, gk = k0 + k;
+          if (gk < k_end && gm + 8 <= M) {
+            const u16x8_t v =
+                *reinterpret_cast<const u16x8_t*>(&A[gk * M + gm]);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              *reinterpret_cast<unsigned short*>(&As[m8 + j][k]) = v[j];
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              As[m8 + j][k] = (gk < k_end && gm + j < M)
+                                  ? A[gk * M + gm + j] : zero;
+          }
+        }
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < (BM * BK) / THREADS; ++i) {
+        int idx = tid + i * THREADS;
+        int m = idx / BK, k = idx % BK;
+        long gm = tile_m + m, gk = k0 + k;
+        As[m][k] = (gm < M && gk < k_end)
+                       ? (TA ? A[gk * M + gm] : A[gm * K + gk]) : zero;
+      }
     }
+    // ---- stage B tile ----
+    if (vecB) {
+      constexpr int GROUPS = (BN * BK) / 8;
 #pragma unroll
-    for (int i = 0; i < (BN * BK) / THREADS; ++i) {
-      int idx = tid + i * THREADS;
-      int n = idx / BK, k = idx % BK;
-      long gn = tile_n + n, gk = k0 + k;
-      float v = 0.f;
-      if (gn < N && gk < k_end)
-        v = b2f(TB ? B[gn * K + gk] : B[gk * N + gn]);
-      Bs[n][k] = f2b(v);
+      for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
+        const int g = tid + i * THREADS;
+        if (GROUPS % THREADS != 0 && g >= GROUPS) break;
+        if (TB) {   // B[N][K]: 8 consecutive k per thread
+          const int n = g / (BK / 8), k8 = (g % (BK / 8)) * 8;
+          const long gn = tile_n + n, gk = k0 + k8;
+          if (gn < N && gk + 8 <= k_end) {
+            *reinterpret_cast<bf16x8_t*>(&Bs[n][k8]) =
+                *reinterpret_cast<const bf16x8_t*>(&B[gn * K + gk]);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              Bs[n][k8 + j] = (gn < N && gk + j < k_end)
+                                  ? B[gn * K + gk + j] : zero;
+          }
+        } else {    // B[K][N]: 8 consecutive n (same k) per thread
+          const int k = g / (BN / 8), n8 = (g % (BN / 8)) * 8;
+          const long gn = tile_n + n8, gk = k0 + k;
+          if (gk < k_end && gn + 8 <= N) {
+            const u16x8_t v =
+                *reinterpret_cast<const u16x8_t*>(&B[gk * N + gn]);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              *reinterpret_cast<unsigned short*>(&Bs[n8 + j][k]) = v[j];
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              Bs[n8 + j][k] = (gk < k_end && gn + j < N)
+                                  ? B[gk * N + gn + j] : zero;
+          }
+        }
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < (BN * BK) / THREADS; ++i) {
+        int idx = tid + i * THREADS;
+        int n = idx / BK, k = idx % BK;
+        long gn = tile_n + n, gk = k0 + k;
+        Bs[n][k] = (gn < N && gk < k_end)
+                       ? (TB ? B[gn * K + gk] : B[gk * N + gn]) : zero;
+      }
     }
     __syncthreads();
 
@@ -273,6 +347,8 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   }
 
   dim3 grid((unsigned)tiles, (unsigned)S);
+  const int vecA = ta ? (M % 8 == 0) : (K % 8 == 0);
+  const int vecB = tb ? (K % 8 == 0) : (N % 8 == 0);
 
   auto launch = [&](auto bm, auto bn, auto wr, auto wc) {
     constexpr int BMv = decltype(bm)::value, BNv = decltype(bn)::value;
@@ -282,19 +358,19 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
     if (!ta && !tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, false, false>),
                          grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
-                         M, N, K, kslice, relu, sm, ohw);
+                         M, N, K, kslice, relu, sm, ohw, vecA, vecB);
     else if (!ta && tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, false, true>),
                          grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
-                         M, N, K, kslice, relu, sm, ohw);
+                         M, N, K, kslice, relu, sm, ohw, vecA, vecB);
     else if (ta && !tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, true, false>),
                          grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
-                         M, N, K, kslice, relu, sm, ohw);
+                         M, N, K, kslice, relu, sm, ohw, vecA, vecB);
     else
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, true, true>),
                          grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
-                         M, N, K, kslice, relu, sm, ohw);
+                         M, N, K, kslice, relu, sm, ohw, vecA, vecB);
   };
 
   using c32 = std::integral_constant<int, 32>;
