@@ -24,39 +24,49 @@ struct ConvShape {
 };
 
 // col[(n*OH+oh)*OW+ow][(c*R+r)*S+s] = x[n][c][oh*st-pad+r][ow*st-pad+s]
+// One thread per (m, c, r) row-of-S: the S reads are contiguous in x and
+// the S writes contiguous in col; 32-bit index math (the 64-bit div/mod
+// chains of the per-element version were VALU-bound at ~190us/call).
 __global__ void im2col_kernel(const bf16* __restrict__ x,
-                              bf16* __restrict__ col, ConvShape sh) {
-  const long total = sh.M() * sh.CRS();
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
-  const long crs = sh.CRS();
-  for (; i < total; i += stride) {
-    const long m = i / crs;
-    const int k = (int)(i - m * crs);
-    const int s = k % sh.S, r = (k / sh.S) % sh.R, c = k / (sh.S * sh.R);
-    const int ow = (int)(m % sh.OW), oh = (int)((m / sh.OW) % sh.OH);
-    const int n = (int)(m / ((long)sh.OW * sh.OH));
+                              bf16* __restrict__ col, ConvShape sh,
+                              int total_mcr) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int stride = gridDim.x * blockDim.x;
+  const bf16 zero = f2b(0.f);
+  for (; i < total_mcr; i += stride) {
+    const int r = i % sh.R;
+    const int c = (i / sh.R) % sh.C;
+    const int m = i / (sh.R * sh.C);
+    const int ow = m % sh.OW, oh = (m / sh.OW) % sh.OH;
+    const int n = m / (sh.OW * sh.OH);
     const int ih = oh * sh.stride - sh.pad + r;
-    const int iw = ow * sh.stride - sh.pad + s;
-    bf16 v = f2b(0.f);
-    if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)
-      v = x[(((long)n * sh.C + c) * sh.H + ih) * sh.W + iw];
-    col[i] = v;
+    bf16* out = col + (size_t)m * (sh.C * sh.R * sh.S)
+                + (c * sh.R + r) * sh.S;
+    if (ih < 0 || ih >= sh.H) {
+      for (int s = 0; s < sh.S; ++s) out[s] = zero;
+      continue;
+    }
+    const int iw0 = ow * sh.stride - sh.pad;
+    const bf16* src = x + ((size_t)(n * sh.C + c) * sh.H + ih) * sh.W;
+    for (int s = 0; s < sh.S; ++s) {
+      const int iw = iw0 + s;
+      out[s] = (iw >= 0 && iw < sh.W) ? src[iw] : zero;
+    }
   }
 }
 
 // Gather col2im: dx[n][c][ih][iw] = sum over (r,s) with valid (oh,ow) of
 // dcol[(n*OH+oh)*OW+ow][(c*R+r)*S+s]. Deterministic (fixed r,s order).
 __global__ void col2im_kernel(const bf16* __restrict__ dcol,
-                              bf16* __restrict__ dx, ConvShape sh) {
-  const long total = (long)sh.N * sh.C * sh.H * sh.W;
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
-  const long crs = sh.CRS();
+                              bf16* __restrict__ dx, ConvShape sh,
+                              int total) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int stride = gridDim.x * blockDim.x;
+  const size_t crs = sh.CRS();
   for (; i < total; i += stride) {
-    const int iw = (int)(i % sh.W), ih = (int)((i / sh.W) % sh.H);
-    const int c = (int)((i / ((long)sh.W * sh.H)) % sh.C);
-    const int n = (int)(i / ((long)sh.W * sh.H * sh.C));
+    const int iw = i % sh.W, ih = (i / sh.W) % sh.H;
+    const int c = (i / (sh.W * sh.H)) % sh.C;
+    const int n = i / (sh.W * sh.H * sh.C);
     float acc = 0.f;
     for (int r = 0; r < sh.R; ++r) {
       const int oh_num = ih + sh.pad - r;
@@ -68,8 +78,8 @@ __global__ void col2im_kernel(const bf16* __restrict__ dcol,
         if (ow_num < 0 || ow_num % sh.stride) continue;
         const int ow = ow_num / sh.stride;
         if (ow >= sh.OW) continue;
-        const long m = ((long)n * sh.OH + oh) * sh.OW + ow;
-        const long k = ((long)c * sh.R + r) * sh.S + s;
+        const size_t m = ((size_t)n * sh.OH + oh) * sh.OW + ow;
+        const int k = (c * sh.R + r) * sh.S + s;
         acc += b2f(dcol[m * crs + k]);
       }
     }
@@ -80,17 +90,16 @@ __global__ void col2im_kernel(const bf16* __restrict__ dcol,
 // NCHW [N,K,OH,OW] -> [N*OH*OW, K] (dy2 for the backward GEMMs)
 __global__ void nchw_to_mk_kernel(const bf16* __restrict__ src,
                                   bf16* __restrict__ dst, int N, int K,
-                                  long ohw) {
-  const long total = (long)N * K * ohw;
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
+                                  int ohw, int total) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int stride = gridDim.x * blockDim.x;
   for (; i < total; i += stride) {
     // i indexes dst [(n*ohw+sp) * K + k]
-    const int k = (int)(i % K);
-    const long m = i / K;
-    const long sp = m % ohw;
-    const long n = m / ohw;
-    dst[i] = src[((long)n * K + k) * ohw + sp];
+    const int k = i % K;
+    const int m = i / K;
+    const int sp = m % ohw;
+    const int n = m / ohw;
+    dst[i] = src[((size_t)n * K + k) * ohw + sp];
   }
 }
 
@@ -148,7 +157,7 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
 }
 
 inline int ew_grid(long n) {
-  return (int)std::min<long>((n + 1023) / 1024, 8192);
+  return (int)std::min<long>((n + 255) / 256, 16384);
 }
 
 ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
@@ -166,10 +175,12 @@ ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
 
 torch::Tensor im2col(const torch::Tensor& x, const ConvShape& sh) {
   auto col = torch::empty({sh.M(), sh.CRS()}, x.options());
-  const long total = sh.M() * sh.CRS();
-  hipLaunchKernelGGL(im2col_kernel, dim3(ew_grid(total)), dim3(1024), 0,
-                     cur_stream(), (const bf16*)x.data_ptr(),
-                     (bf16*)col.data_ptr(), sh);
+  const long total_mcr = sh.M() * sh.C * sh.R;
+  TORCH_CHECK(total_mcr < INT32_MAX && sh.M() * sh.CRS() / 8 < INT32_MAX,
+              "conv im2col index overflow; reduce batch");
+  hipLaunchKernelGGL(im2col_kernel, dim3(ew_grid(total_mcr)), dim3(256),
+                     0, cur_stream(), (const bf16*)x.data_ptr(),
+                     (bf16*)col.data_ptr(), sh, (int)total_mcr);
   HIP_CHECK(hipGetLastError());
   return col;
 }
@@ -177,10 +188,11 @@ torch::Tensor im2col(const torch::Tensor& x, const ConvShape& sh) {
 torch::Tensor dy_to_mk(const torch::Tensor& dy, const ConvShape& sh) {
   auto dy2 = torch::empty({sh.M(), (long)sh.Kout}, dy.options());
   const long total = sh.M() * sh.Kout;
-  hipLaunchKernelGGL(nchw_to_mk_kernel, dim3(ew_grid(total)), dim3(1024), 0,
+  TORCH_CHECK(total < INT32_MAX, "dy2 index overflow");
+  hipLaunchKernelGGL(nchw_to_mk_kernel, dim3(ew_grid(total)), dim3(256), 0,
                      cur_stream(), (const bf16*)dy.data_ptr(),
                      (bf16*)dy2.data_ptr(), sh.N, sh.Kout,
-                     (long)sh.OH * sh.OW);
+                     (int)(sh.OH * sh.OW), (int)total);
   HIP_CHECK(hipGetLastError());
   return dy2;
 }
@@ -217,9 +229,10 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   gemm_bf16_raw(dy2, w2, dcol, sh.M(), sh.CRS(), sh.Kout, false, false,
                 nullptr, false, EpStore::kPlain, 0);
   auto dx = torch::empty_like(x);
-  hipLaunchKernelGGL(col2im_kernel, dim3(ew_grid(dx.numel())), dim3(1024), 0,
+  TORCH_CHECK(dx.numel() < INT32_MAX, "col2im index overflow");
+  hipLaunchKernelGGL(col2im_kernel, dim3(ew_grid(dx.numel())), dim3(256), 0,
                      cur_stream(), (const bf16*)dcol.data_ptr(),
-                     (bf16*)dx.data_ptr(), sh);
+                     (bf16*)dx.data_ptr(), sh, (int)dx.numel());
   HIP_CHECK(hipGetLastError());
 
   // wgrad: dW[Kout, CRS] = dy2^T @ col

@@ -266,20 +266,31 @@ __global__ void splitk_reduce_kernel(const float* __restrict__ Cpart, int S,
   }
 }
 
-// Hierarchical colsum: pass 1 tiles rows into fp32 partials (coalesced:
-// thread -> column), pass 2 reduces chunks in fixed ascending order.
-constexpr int kColsumRows = 2048;
+// Hierarchical colsum: pass 1 tiles rows into fp32 partials, pass 2
+// reduces chunks in fixed ascending order. Block = 8 row-lanes x 32
+// columns so each block has 8-way row parallelism (a single serial
+// row-walk per thread was latency-bound at ~300us/call).
+constexpr int kColsumRows = 1024;
 
 __global__ void colsum_part_kernel(const bf16* __restrict__ X, long M, long N,
                                    float* __restrict__ part) {
-  const long col = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long col = (long)blockIdx.x * 32 + (threadIdx.x & 31);
+  const int rlane = threadIdx.x >> 5;  // 0..7
   const int chunk = blockIdx.y;
-  if (col >= N) return;
   const long r0 = (long)chunk * kColsumRows;
   const long r1 = min(M, r0 + kColsumRows);
   float acc = 0.f;
-  for (long m = r0; m < r1; ++m) acc += b2f(X[m * N + col]);
-  part[(long)chunk * N + col] = acc;
+  if (col < N)
+    for (long m = r0 + rlane; m < r1; m += 8) acc += b2f(X[m * N + col]);
+  __shared__ float red[8][33];
+  red[rlane][threadIdx.x & 31] = acc;
+  __syncthreads();
+  if (rlane == 0 && col < N) {
+    float t = 0.f;
+#pragma unroll
+    for (int r = 0; r < 8; ++r) t += red[r][threadIdx.x & 31];
+    part[(long)chunk * N + col] = t;
+  }
 }
 
 __global__ void colsum_final_kernel(const float* __restrict__ part,
@@ -333,7 +344,10 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   const long ksteps = (K + BK - 1) / BK;
   long S = 1;
   if (tiles < 512 && K >= 4 * BK) {
-    S = std::min<long>({(512 + tiles - 1) / tiles, ksteps, 64});
+    // cap slices by the partial-buffer budget (fp32 [S, M, N] <= 256 MB)
+    const long budget = (256L << 20) / std::max<long>(M * N * 4, 1);
+    S = std::min<long>({(512 + tiles - 1) / tiles, ksteps, 512, budget});
+    S = std::max<long>(S, 1);
   }
   const long kslice = ((ksteps + S - 1) / S) * BK;
   S = (K + kslice - 1) / kslice;  // actual slices after rounding
@@ -411,7 +425,7 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
   auto out = torch::empty({N}, X.options());
   const int chunks = (int)((M + kColsumRows - 1) / kColsumRows);
   auto part = torch::empty({chunks, N}, X.options().dtype(at::kFloat));
-  dim3 grid(ceil_div(N, 256), chunks);
+  dim3 grid(ceil_div(N, 32), chunks);
   hipLaunchKernelGGL(colsum_part_kernel, grid, dim3(256), 0, cur_stream(),
                      (const bf16*)X.data_ptr(), M, N,
                      part.data_ptr<float>());
