@@ -144,11 +144,9 @@ class FLEngine:
                 loss.backward()
                 if cfg.optimizer == "adam":
                     step += 1
-                    O.adam_step_(self.model.flat.data, self.model.flat.grad,
-                                 m, v, step, cfg.learning_rate)
+                    self.model.adam_step(m, v, step, cfg.learning_rate)
                 else:
-                    O.sgd_step_(self.model.flat.data, self.model.flat.grad,
-                                cfg.learning_rate)
+                    self.model.sgd_step(cfg.learning_rate)
                 cost_accum += loss.detach()
         avg_cost = float(cost_accum) / (total_batches * cfg.local_epochs)
         # delta = (W0 - W)/lr  (reference main.py:153-154)

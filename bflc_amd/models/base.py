@@ -38,8 +38,18 @@ class FlatModel:
             self._offsets[name] = (off, shape)
             off += n
         self.numel = off
+        # fp32 master vector; when computing in bf16 a separate bf16
+        # shadow leaf carries the forward/backward (its .grad is the flat
+        # bf16 gradient) and the fused sgd_master_/adam_master_ kernels
+        # update master + shadow in one pass — no per-parameter casts.
         self.flat = torch.zeros(off, dtype=torch.float32, device=self.device)
-        self.flat.requires_grad_(True)
+        if self.compute_dtype == torch.float32:
+            self.flat.requires_grad_(True)
+            self.cflat = self.flat
+        else:
+            self.cflat = torch.zeros(off, dtype=self.compute_dtype,
+                                     device=self.device)
+            self.cflat.requires_grad_(True)
 
     # -- subclass interface -------------------------------------------------
     def specs(self) -> Sequence[Tuple[str, Tuple[int, ...], str]]:
@@ -75,16 +85,25 @@ class FlatModel:
                 raise ValueError(init)
         with torch.no_grad():
             self.flat.copy_(host.to(self.device))
+        self._sync_shadow()
+
+    def _sync_shadow(self) -> None:
+        if self.cflat is self.flat:
+            return
+        with torch.no_grad():
+            if self.device.type == "cuda":
+                from bflc_amd.ops.functional import hip_ops
+                hip_ops().refresh_shadow_(self.flat.detach(),
+                                          self.cflat.data)
+            else:
+                self.cflat.data.copy_(self.flat.detach())
 
     def p(self, name: str) -> torch.Tensor:
         """Autograd view of a parameter in compute dtype (fresh per call:
-        views must be re-derived after in-place flat updates)."""
+        views must be re-derived after in-place updates)."""
         off, shape = self._offsets[name]
         n = int(math.prod(shape))
-        v = self.flat[off:off + n].view(shape)
-        if self.compute_dtype != torch.float32:
-            v = v.to(self.compute_dtype)
-        return v
+        return self.cflat[off:off + n].view(shape)
 
     def get_flat(self) -> torch.Tensor:
         return self.flat.detach().clone()
@@ -92,11 +111,37 @@ class FlatModel:
     def set_flat(self, v: torch.Tensor) -> None:
         with torch.no_grad():
             self.flat.copy_(v.to(self.device))
+        self._sync_shadow()
 
     def zero_grad(self) -> None:
-        if self.flat.grad is not None:
-            self.flat.grad.detach_()
-            self.flat.grad.zero_()
+        if self.cflat.grad is not None:
+            self.cflat.grad.detach_()
+            self.cflat.grad.zero_()
+
+    # -- fused optimizer steps (fp32 master + shadow refresh) ----------
+    def sgd_step(self, lr: float) -> None:
+        g = self.cflat.grad
+        assert g is not None
+        if self.cflat is self.flat:
+            from bflc_amd.ops.functional import sgd_step_
+            sgd_step_(self.flat.data, g, lr)
+        else:
+            from bflc_amd.ops.functional import hip_ops
+            hip_ops().sgd_master_(self.flat.detach(), self.cflat.data, g,
+                                  float(lr))
+
+    def adam_step(self, m: torch.Tensor, v: torch.Tensor, step: int,
+                  lr: float) -> None:
+        g = self.cflat.grad
+        assert g is not None
+        if self.cflat is self.flat:
+            from bflc_amd.ops.functional import adam_step_
+            adam_step_(self.flat.data, g, m, v, step, lr)
+        else:
+            from bflc_amd.ops.functional import hip_ops
+            hip_ops().adam_master_(self.flat.detach(), self.cflat.data, g,
+                                   m, v, int(step), float(lr), 0.9, 0.999,
+                                   1e-8)
 
     # -- train/eval ---------------------------------------------------------
     def loss(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:

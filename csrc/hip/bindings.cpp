@@ -13,6 +13,12 @@ void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, long step, double lr, double beta1,
                 double beta2, double eps);
 torch::Tensor weighted_fedavg(torch::Tensor deltas, torch::Tensor w);
+void sgd_master_(torch::Tensor p, torch::Tensor shadow, torch::Tensor g,
+                 double lr);
+void adam_master_(torch::Tensor p, torch::Tensor shadow, torch::Tensor g,
+                  torch::Tensor m, torch::Tensor v, long step, double lr,
+                  double beta1, double beta2, double eps);
+void refresh_shadow_(torch::Tensor p, torch::Tensor shadow);
 torch::Tensor relu_fwd(torch::Tensor x);
 torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy);
 
@@ -64,6 +70,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step_", &bflc::adam_step_, "fused flat Adam");
   m.def("weighted_fedavg", &bflc::weighted_fedavg,
         "fixed-order weighted FedAvg reduce");
+  m.def("sgd_master_", &bflc::sgd_master_,
+        "fused fp32-master SGD + bf16 shadow refresh");
+  m.def("adam_master_", &bflc::adam_master_);
+  m.def("refresh_shadow_", &bflc::refresh_shadow_, "shadow = bf16(master)");
   m.def("relu_fwd", &bflc::relu_fwd);
   m.def("relu_bwd", &bflc::relu_bwd);
   m.def("softmax_ce_fwd", &bflc::softmax_ce_fwd,

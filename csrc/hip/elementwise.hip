@@ -106,6 +106,50 @@ __global__ void fedavg_kernel(const float* __restrict__ deltas,
   }
 }
 
+// Mixed-precision fused updates: fp32 master + bf16 compute shadow in
+// ONE pass (replaces per-parameter cast kernels each step).
+__global__ void sgd_master_kernel(float* __restrict__ p,
+                                  bf16* __restrict__ shadow,
+                                  const bf16* __restrict__ g, float lr,
+                                  long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride) {
+    float v = fmaf(-lr, b2f(g[k]), p[k]);
+    p[k] = v;
+    shadow[k] = f2b(v);
+  }
+}
+
+__global__ void adam_master_kernel(float* __restrict__ p,
+                                   bf16* __restrict__ shadow,
+                                   const bf16* __restrict__ g,
+                                   float* __restrict__ m,
+                                   float* __restrict__ v, float lr,
+                                   float beta1, float beta2, float eps,
+                                   float bc1, float bc2, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride) {
+    float gk = b2f(g[k]);
+    float mk = beta1 * m[k] + (1.f - beta1) * gk;
+    float vk = beta2 * v[k] + (1.f - beta2) * gk * gk;
+    m[k] = mk;
+    v[k] = vk;
+    float pv = p[k] - lr * (mk / bc1) / (sqrtf(vk / bc2) + eps);
+    p[k] = pv;
+    shadow[k] = f2b(pv);
+  }
+}
+
+// shadow refresh: shadow = bf16(master)
+__global__ void cast_f32_bf16_kernel(const float* __restrict__ p,
+                                     bf16* __restrict__ shadow, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride) shadow[k] = f2b(p[k]);
+}
+
 template <typename T>
 __global__ void relu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                 long n) {
@@ -185,6 +229,44 @@ torch::Tensor weighted_fedavg(torch::Tensor deltas, torch::Tensor w) {
                      w.data_ptr<float>(), K, P, wsum, out.data_ptr<float>());
   HIP_CHECK(hipGetLastError());
   return out;
+}
+
+void sgd_master_(torch::Tensor p, torch::Tensor shadow, torch::Tensor g,
+                 double lr) {
+  CHECK_GPU(p); CHECK_CONTIG(p); CHECK_CONTIG(shadow); CHECK_CONTIG(g);
+  TORCH_CHECK(p.scalar_type() == at::kFloat &&
+              shadow.scalar_type() == at::kBFloat16 &&
+              g.scalar_type() == at::kBFloat16);
+  long n = p.numel();
+  hipLaunchKernelGGL(sgd_master_kernel, dim3(grid_for(n, 2)), dim3(kBlock),
+                     0, cur_stream(), p.data_ptr<float>(),
+                     (bf16*)shadow.data_ptr(), (const bf16*)g.data_ptr(),
+                     (float)lr, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void adam_master_(torch::Tensor p, torch::Tensor shadow, torch::Tensor g,
+                  torch::Tensor m, torch::Tensor v, long step, double lr,
+                  double beta1, double beta2, double eps) {
+  CHECK_GPU(p); CHECK_CONTIG(p); CHECK_CONTIG(shadow); CHECK_CONTIG(g);
+  long n = p.numel();
+  float bc1 = 1.f - powf((float)beta1, (float)step);
+  float bc2 = 1.f - powf((float)beta2, (float)step);
+  hipLaunchKernelGGL(adam_master_kernel, dim3(grid_for(n, 2)), dim3(kBlock),
+                     0, cur_stream(), p.data_ptr<float>(),
+                     (bf16*)shadow.data_ptr(), (const bf16*)g.data_ptr(),
+                     m.data_ptr<float>(), v.data_ptr<float>(), (float)lr,
+                     (float)beta1, (float)beta2, (float)eps, bc1, bc2, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void refresh_shadow_(torch::Tensor p, torch::Tensor shadow) {
+  CHECK_GPU(p); CHECK_CONTIG(p); CHECK_CONTIG(shadow);
+  long n = p.numel();
+  hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3(grid_for(n)), dim3(kBlock),
+                     0, cur_stream(), p.data_ptr<float>(),
+                     (bf16*)shadow.data_ptr(), n);
+  HIP_CHECK(hipGetLastError());
 }
 
 torch::Tensor relu_fwd(torch::Tensor x) {

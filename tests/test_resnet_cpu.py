@@ -75,7 +75,7 @@ class TestResNet20:
             m.zero_grad()
             loss = m.loss(x, y)
             loss.backward()
-            O.sgd_step_(m.flat.data, m.flat.grad, 0.05)
+            m.sgd_step(0.05)
             losses.append(float(loss))
         assert losses[-1] < losses[0]
 

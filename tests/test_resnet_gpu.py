@@ -117,7 +117,7 @@ class TestResNetGPU:
             m.zero_grad()
             loss = m.loss(x, y)
             loss.backward()
-            O.sgd_step_(m.flat.data, m.flat.grad, cfg.learning_rate)
+            m.sgd_step(cfg.learning_rate)
             losses.append(float(loss))
         assert losses[-1] < losses[0]
 
