@@ -24,9 +24,11 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--model", default="femnist_cnn")
-    ap.add_argument("--samples-per-client", type=int, default=3072)
-    ap.add_argument("--batch-size", type=int, default=1024)
+    ap.add_argument("--model", default="femnist_cnn",
+                    choices=["femnist_cnn", "resnet20", "resnet50",
+                             "mlp", "logreg"])
+    ap.add_argument("--samples-per-client", type=int, default=None)
+    ap.add_argument("--batch-size", type=int, default=None)
     ap.add_argument("--eval", action="store_true", help="eval every round")
     ap.add_argument("--phases", action="store_true",
                     help="print per-phase timings of each timed round")
@@ -37,12 +39,28 @@ def main() -> None:
     from bflc_amd.data import make_federated
     from bflc_amd.fl import FLEngine
 
+    # Per-model defaults = the BASELINE.json configs:
+    #   femnist_cnn -> FEMNIST 1x28x28, 62 classes (headline, config 2)
+    #   resnet20    -> CIFAR-10 3x32x32 (config 3)
+    #   resnet50    -> synthetic ImageNet-shape 3x224x224 (config 5)
+    MODEL_DEFAULTS = {
+        "femnist_cnn": dict(n_class=62, spc=3072, bs=1024, img="1x28x28x62cls"),
+        "resnet20": dict(n_class=10, spc=2048, bs=512, img="3x32x32x10cls"),
+        "resnet50": dict(n_class=1000, spc=256, bs=64, img="3x224x224x1000cls"),
+        "mlp": dict(n_class=2, spc=3072, bs=1024, img=None),
+        "logreg": dict(n_class=2, spc=3072, bs=1024, img=None),
+    }
+    md = MODEL_DEFAULTS[args.model]
+    spc = args.samples_per_client or md["spc"]
+    bs = args.batch_size or md["bs"]
+
     n = int(os.environ.get("WORLD_SIZE", args.gpus))
     cfg = FLConfig.for_world(
-        n, model=args.model, n_class=62,
-        samples_per_client=args.samples_per_client,
-        batch_size=args.batch_size, partition="dirichlet",
-        dirichlet_alpha=0.3, eval_samples=4096, learning_rate=0.01)
+        n, model=args.model, n_class=md["n_class"],
+        samples_per_client=spc,
+        batch_size=bs, partition="dirichlet",
+        dirichlet_alpha=0.3, eval_samples=min(4096, 2 * spc),
+        learning_rate=0.01)
 
     t = Transport()
     shards, test = make_federated(cfg)
@@ -106,7 +124,7 @@ def main() -> None:
                 "model": cfg.model,
                 "global_batch": cfg.batch_size * cfg.needed_update_count,
                 "seq_len": None,
-                "img": "1x28x28x62cls",
+                "img": md["img"],
                 "parallelism": f"fl_nodes{n}",
                 "clients": cfg.client_num,
                 "committee": cfg.comm_count,
