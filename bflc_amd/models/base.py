@@ -125,10 +125,14 @@ class FlatModel:
         if self.cflat is self.flat:
             from bflc_amd.ops.functional import sgd_step_
             sgd_step_(self.flat.data, g, lr)
-        else:
+        elif self.device.type == "cuda":
             from bflc_amd.ops.functional import hip_ops
             hip_ops().sgd_master_(self.flat.detach(), self.cflat.data, g,
                                   float(lr))
+        else:  # CPU oracle of the fused master-SGD kernel
+            with torch.no_grad():
+                self.flat.add_(g.float(), alpha=-lr)
+                self.cflat.data.copy_(self.flat.to(self.compute_dtype))
 
     def adam_step(self, m: torch.Tensor, v: torch.Tensor, step: int,
                   lr: float) -> None:
@@ -137,11 +141,16 @@ class FlatModel:
         if self.cflat is self.flat:
             from bflc_amd.ops.functional import adam_step_
             adam_step_(self.flat.data, g, m, v, step, lr)
-        else:
+        elif self.device.type == "cuda":
             from bflc_amd.ops.functional import hip_ops
             hip_ops().adam_master_(self.flat.detach(), self.cflat.data, g,
                                    m, v, int(step), float(lr), 0.9, 0.999,
                                    1e-8)
+        else:  # CPU oracle of the fused master-Adam kernel
+            from bflc_amd.ops.functional import adam_step_
+            with torch.no_grad():
+                adam_step_(self.flat.data, g.float(), m, v, step, lr)
+                self.cflat.data.copy_(self.flat.to(self.compute_dtype))
 
     # -- train/eval ---------------------------------------------------------
     def loss(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:

@@ -95,3 +95,51 @@ class TestResNet50:
         assert float(m.flat.grad.abs().sum()) > 0
         # ~25.5M params at 1000 classes; ~23.5M at 10
         assert 20e6 < m.numel < 27e6
+
+
+class TestMixedPrecisionCPU:
+    """CPU oracle of the bf16-shadow mixed-precision path (the GPU path
+    uses the fused sgd_master_/adam_master_ kernels; on CPU base.py
+    implements the same math with torch ops)."""
+
+    def _model(self, optimizer="sgd"):
+        cfg = FLConfig(model="mlp", n_features=16, n_class=4, client_num=1,
+                       comm_count=1, needed_update_count=1,
+                       aggregate_count=1, optimizer=optimizer)
+        return cfg, build_model(cfg, torch.device("cpu"),
+                                compute_dtype=torch.bfloat16)
+
+    def test_master_stays_fp32_and_learns(self):
+        cfg, m = self._model()
+        assert m.flat.dtype == torch.float32
+        assert m.cflat.dtype == torch.bfloat16
+        torch.manual_seed(0)
+        x = torch.randn(64, 16)
+        y = torch.randint(0, 4, (64,))
+        losses = []
+        for _ in range(30):
+            m.zero_grad()
+            loss = m.loss(x, y)
+            loss.backward()
+            m.sgd_step(0.1)
+            losses.append(float(loss.detach()))
+        assert losses[-1] < 0.75 * losses[0]
+        # shadow tracks master (equal after bf16 rounding)
+        assert torch.equal(m.cflat.data, m.flat.to(torch.bfloat16))
+
+    def test_adam_master_step(self):
+        cfg, m = self._model("adam")
+        mom = torch.zeros(m.numel)
+        vel = torch.zeros(m.numel)
+        torch.manual_seed(0)
+        x = torch.randn(64, 16)
+        y = torch.randint(0, 4, (64,))
+        losses = []
+        for step in range(1, 21):
+            m.zero_grad()
+            loss = m.loss(x, y)
+            loss.backward()
+            m.adam_step(mom, vel, step, 0.01)
+            losses.append(float(loss.detach()))
+        assert losses[-1] < losses[0]
+        assert torch.equal(m.cflat.data, m.flat.to(torch.bfloat16))
