@@ -67,6 +67,21 @@ def main():
               f" {tt*1e3:>9.3f} {flops/tt/1e12:>8.1f} | "
               f"{tt/to:>5.2f}x")
 
+    # Operand-layout ablation (gemm_raw): staging differs per layout —
+    # K-contiguous operands use 16-B vector LDS writes, the others scatter.
+    print("\nlayout ablation (ours only):")
+    for M, K, N in [(4096, 4096, 4096), (12544, 2304, 512)]:
+        flops = 2.0 * M * K * N
+        for ta, tb in [(False, False), (False, True),
+                       (True, False), (True, True)]:
+            A = torch.randn(*((K, M) if ta else (M, K)), device=dev,
+                            dtype=torch.bfloat16)
+            B = torch.randn(*((N, K) if tb else (K, N)), device=dev,
+                            dtype=torch.bfloat16)
+            t = time_fn(lambda: hip.gemm_raw(A, B, ta, tb), args.iters, 3)
+            print(f"  {M}x{K}x{N} ta={int(ta)} tb={int(tb)}: "
+                  f"{t*1e3:8.3f} ms {flops/t/1e12:7.1f} TF")
+
 
 if __name__ == "__main__":
     main()

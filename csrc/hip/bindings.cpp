@@ -33,6 +33,7 @@ double accuracy(torch::Tensor logits, torch::Tensor target);
 
 // gemm_bf16.hip
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b);
+torch::Tensor gemm_raw(torch::Tensor A, torch::Tensor B, bool ta, bool tb);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy);
 
@@ -81,6 +82,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_bwd", &bflc::softmax_ce_bwd);
   m.def("accuracy", &bflc::accuracy, "fused argmax-compare-reduce");
   m.def("linear_fwd", &bflc::linear_fwd, "MFMA bf16 GEMM + bias");
+  m.def("gemm_raw", &bflc::gemm_raw, "raw GEMM (bench/ablation)");
   m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");
   m.def("conv2d_fwd", &bflc::conv2d_fwd, "im2col + MFMA GEMM, NCHW");
   m.def("conv2d_bwd", &bflc::conv2d_bwd, "(dx, dw, db)");

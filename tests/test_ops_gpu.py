@@ -37,7 +37,9 @@ class TestLinear:
         (100, 3136, 128),   # FEMNIST fc1
         (100, 128, 62),     # FEMNIST head
         (37, 100, 62),      # odd sizes / predication
-        (256, 512, 512),    # square-ish
+        (256, 512, 512),    # square-ish (256x256 double-buffered path)
+        (512, 2304, 256),   # resnet50-mid shape class: 256-path + split-K
+        (768, 320, 256),    # 256-path, K not a multiple of 256
         (2036, 5, 2),       # sponsor global test shape
     ])
     def test_fwd(self, m, k, n):
@@ -50,7 +52,9 @@ class TestLinear:
         assert_close(y, ref)
 
     @pytest.mark.parametrize("m,k,n", [(100, 5, 2), (64, 3136, 128),
-                                       (37, 100, 62), (128, 256, 192)])
+                                       (37, 100, 62), (128, 256, 192),
+                                       # 256-path TB (dx) / TA (dw) layouts
+                                       (256, 512, 512), (512, 2304, 256)])
     def test_bwd(self, m, k, n):
         torch.manual_seed(1)
         x, w, dy = torch.randn(m, k), torch.randn(k, n), torch.randn(m, n)
