@@ -151,22 +151,27 @@ class _Conv2dFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b, stride: int, padding: int):
-        ctx.save_for_backward(x, w)
         ctx.stride, ctx.padding = stride, padding
         ctx.has_bias = b is not None
         if x.is_cuda:
-            return hip_ops().conv2d_fwd(x, w, b if b is not None else
-                                        torch.zeros(w.shape[0], device=x.device,
-                                                    dtype=x.dtype),
-                                        stride, padding)
+            y, col = hip_ops().conv2d_fwd_col(
+                x, w, b if b is not None else
+                torch.zeros(w.shape[0], device=x.device, dtype=x.dtype),
+                stride, padding)
+            # keep col for wgrad: recomputing im2col cost ~12% of a round
+            ctx.save_for_backward(x, w, col)
+            return y
+        ctx.save_for_backward(x, w)
         return F.conv2d(x, w, b, stride=stride, padding=padding)
 
     @staticmethod
     def backward(ctx, dy):
-        x, w = ctx.saved_tensors
+        x, w = ctx.saved_tensors[:2]
         dy = dy.contiguous()
         if x.is_cuda:
-            dx, dw, db = hip_ops().conv2d_bwd(x, w, dy, ctx.stride, ctx.padding)
+            col = ctx.saved_tensors[2]
+            dx, dw, db = hip_ops().conv2d_bwd(x, w, dy, ctx.stride,
+                                              ctx.padding, col)
         else:
             dx = torch.nn.grad.conv2d_input(x.shape, w, dy, stride=ctx.stride,
                                             padding=ctx.padding)

@@ -38,11 +38,13 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy);
 
 // conv_im2col.hip
+std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride, long pad);
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          long stride, long pad);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
-    long pad);
+    long pad, c10::optional<torch::Tensor> col_cache);
 std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
                                                        long kernel,
                                                        long stride);
@@ -85,7 +87,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_raw", &bflc::gemm_raw, "raw GEMM (bench/ablation)");
   m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");
   m.def("conv2d_fwd", &bflc::conv2d_fwd, "im2col + MFMA GEMM, NCHW");
-  m.def("conv2d_bwd", &bflc::conv2d_bwd, "(dx, dw, db)");
+  m.def("conv2d_fwd_col", &bflc::conv2d_fwd_col, "(y, col) - col for bwd");
+  m.def("conv2d_bwd", &bflc::conv2d_bwd, "(dx, dw, db)",
+        py::arg("x"), py::arg("w"), py::arg("dy"), py::arg("stride"),
+        py::arg("pad"), py::arg("col_cache") = py::none());
   m.def("maxpool2d_fwd", &bflc::maxpool2d_fwd);
   m.def("maxpool2d_bwd", &bflc::maxpool2d_bwd);
   m.def("batchnorm_fwd", &bflc::batchnorm_fwd,

@@ -199,8 +199,21 @@ torch::Tensor dy_to_mk(const torch::Tensor& dy, const ConvShape& sh) {
 
 }  // namespace
 
+std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride,
+    long pad);
+
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          long stride, long pad) {
+  return std::get<0>(conv2d_fwd_col(x, w, b, stride, pad));
+}
+
+// Returns (y, col) so the autograd wrapper can hand col back to the
+// backward: wgrad needs the same im2col matrix, and recomputing it cost
+// ~12% of an FL round (288 GB HBM3E makes keeping it essentially free).
+std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride,
+    long pad) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv: bf16 only");
   auto sh = make_shape(x, w, stride, pad);
@@ -213,12 +226,12 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
   auto bc = b.contiguous();
   gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.CRS(), false, true, &bc,
                 false, EpStore::kConvNCHW, (long)sh.OH * sh.OW);
-  return y;
+  return {y, col};
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
-    long pad) {
+    long pad, c10::optional<torch::Tensor> col_cache) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w); CHECK_CONTIG(dy);
   auto sh = make_shape(x, w, stride, pad);
   auto w2 = w.view({(long)sh.Kout, sh.CRS()});
@@ -235,8 +248,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
                      (bf16*)dx.data_ptr(), sh, (int)dx.numel());
   HIP_CHECK(hipGetLastError());
 
-  // wgrad: dW[Kout, CRS] = dy2^T @ col
-  auto col = im2col(x, sh);
+  // wgrad: dW[Kout, CRS] = dy2^T @ col (col reused from fwd when given)
+  auto col = col_cache.has_value() ? *col_cache : im2col(x, sh);
   auto dw = torch::empty_like(w2);
   gemm_bf16_raw(dy2, col, dw, sh.Kout, sh.CRS(), sh.M(), true, false,
                 nullptr, false, EpStore::kPlain, 0);

@@ -483,25 +483,42 @@ torch::Tensor transpose_bf16(const torch::Tensor& X) {
 
 // Deterministic split-K reduce: out = sum_s partial[s] (ascending s),
 // then the fused epilogue (bias/relu/NCHW).
+// Block = 8 slice-lanes x 32 output elements; each lane sums slices
+// s === lane (mod 8) ascending, then a fixed-order 8-way combine — the
+// reduction tree is independent of timing/data, so replicas stay
+// bitwise identical (the old thread-per-element serial loop over up to
+// 512 slices was latency-bound at ~50us/call).
 __global__ void splitk_reduce_kernel(const float* __restrict__ Cpart, int S,
                                      long M, long N, bf16* __restrict__ C,
                                      const bf16* __restrict__ bias, int relu,
                                      int store_mode, long ohw) {
   const long total = M * N;
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (; i < total; i += stride) {
+  const int slane = threadIdx.x >> 5;          // 0..7
+  const int elane = threadIdx.x & 31;
+  long i = (long)blockIdx.x * 32 + elane;
+  const long stride = (long)gridDim.x * 32;
+  __shared__ float red[8][33];
+  for (; i - elane < total; i += stride) {
     float acc = 0.f;
-    for (int s = 0; s < S; ++s) acc += Cpart[(long)s * total + i];
-    const long row = i / N, col = i - row * N;
-    if (bias) acc += b2f(bias[col]);
-    if (relu) acc = fmaxf(acc, 0.f);
-    if (store_mode == (int)EpStore::kConvNCHW) {
-      const long img = row / ohw, sp = row % ohw;
-      C[(img * N + col) * ohw + sp] = f2b(acc);
-    } else {
-      C[i] = f2b(acc);
+    if (i < total)
+      for (int s = slane; s < S; s += 8) acc += Cpart[(long)s * total + i];
+    red[slane][elane] = acc;
+    __syncthreads();
+    if (slane == 0 && i < total) {
+      float v = 0.f;
+#pragma unroll
+      for (int r = 0; r < 8; ++r) v += red[r][elane];
+      const long row = i / N, col = i - row * N;
+      if (bias) v += b2f(bias[col]);
+      if (relu) v = fmaxf(v, 0.f);
+      if (store_mode == (int)EpStore::kConvNCHW) {
+        const long img = row / ohw, sp = row % ohw;
+        C[(img * N + col) * ohw + sp] = f2b(v);
+      } else {
+        C[i] = f2b(v);
+      }
     }
+    __syncthreads();
   }
 }
 
@@ -532,14 +549,26 @@ __global__ void colsum_part_kernel(const bf16* __restrict__ X, long M, long N,
   }
 }
 
+// Same fixed-order 8-lane tree as splitk_reduce (chunks can reach
+// M/1024 = 512: a serial per-column loop was ~35us/call).
 __global__ void colsum_final_kernel(const float* __restrict__ part,
                                     int chunks, long N,
                                     bf16* __restrict__ out) {
-  const long col = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= N) return;
+  const int clane = threadIdx.x >> 5;  // 0..7
+  const int elane = threadIdx.x & 31;
+  const long col = (long)blockIdx.x * 32 + elane;
   float acc = 0.f;
-  for (int c = 0; c < chunks; ++c) acc += part[(long)c * N + col];
-  out[col] = f2b(acc);
+  if (col < N)
+    for (int c = clane; c < chunks; c += 8) acc += part[(long)c * N + col];
+  __shared__ float red[8][33];
+  red[clane][elane] = acc;
+  __syncthreads();
+  if (clane == 0 && col < N) {
+    float v = 0.f;
+#pragma unroll
+    for (int r = 0; r < 8; ++r) v += red[r][elane];
+    out[col] = f2b(v);
+  }
 }
 
 struct TileCfg { int bm, bn, wr, wc; };
@@ -614,7 +643,7 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
     HIP_CHECK(hipGetLastError());
     if (S > 1) {
       const long total = M * N;
-      int blocks = (int)std::min<long>((total + 255) / 256, 8192);
+      int blocks = (int)std::min<long>((total + 31) / 32, 16384);
       hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
                          cur_stream(), part_ptr, (int)S, M, N, c, bs,
                          relu ? 1 : 0, (int)store, ohw);
@@ -697,7 +726,7 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
 
   if (S > 1) {
     const long total = M * N;
-    int blocks = (int)std::min<long>((total + 255) / 256, 8192);
+    int blocks = (int)std::min<long>((total + 31) / 32, 16384);
     hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
                        cur_stream(), part_ptr, (int)S, M, N, c, bs,
                        relu ? 1 : 0, (int)store, ohw);
@@ -716,7 +745,7 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
                      (const bf16*)X.data_ptr(), M, N,
                      part.data_ptr<float>());
   HIP_CHECK(hipGetLastError());
-  hipLaunchKernelGGL(colsum_final_kernel, dim3(ceil_div(N, 256)), dim3(256),
+  hipLaunchKernelGGL(colsum_final_kernel, dim3(ceil_div(N, 32)), dim3(256),
                      0, cur_stream(), part.data_ptr<float>(), chunks, N,
                      (bf16*)out.data_ptr());
   HIP_CHECK(hipGetLastError());
