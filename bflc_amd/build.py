@@ -77,7 +77,7 @@ HIP_SOURCES = [
     "elementwise.hip",
     "softmax_ce.hip",
     "gemm_bf16.hip",
-    "conv_im2col.hip",
+    "conv2d.hip",
     "batchnorm.hip",
     "reduce.hip",
     "bindings.cpp",

@@ -44,11 +44,12 @@ def make_tabular(n: int, n_features: int, n_class: int,
 
 def make_images(n: int, channels: int, hw: int, n_class: int,
                 gen: torch.Generator) -> Tuple[torch.Tensor, torch.Tensor]:
-    """FEMNIST/CIFAR/ImageNet-shaped synthetic images: class template +
+    """FEMNIST/CIFAR/ImageNet-shaped synthetic images (NHWC, the
+    channels-last layout the gfx950 conv kernels use): class template +
     noise, so CNNs can learn and accuracy is a meaningful signal."""
-    templates = torch.randn(n_class, channels, hw, hw, generator=gen)
+    templates = torch.randn(n_class, hw, hw, channels, generator=gen)
     y = torch.randint(0, n_class, (n,), generator=gen)
-    x = templates[y] * 0.5 + torch.randn(n, channels, hw, hw, generator=gen)
+    x = templates[y] * 0.5 + torch.randn(n, hw, hw, channels, generator=gen)
     return x, y
 
 

@@ -52,8 +52,9 @@ class FemnistCNN(FlatModel):
 
     def specs(self):
         c = self.cfg
-        return [("c1w", (32, 1, 3, 3), "kaiming"), ("c1b", (32,), "zeros"),
-                ("c2w", (64, 32, 3, 3), "kaiming"), ("c2b", (64,), "zeros"),
+        # NHWC: conv weights [Kout, R, S, C]; x is [N, 28, 28, 1]
+        return [("c1w", (32, 3, 3, 1), "kaiming"), ("c1b", (32,), "zeros"),
+                ("c2w", (64, 3, 3, 32), "kaiming"), ("c2b", (64,), "zeros"),
                 ("f1w", (64 * 7 * 7, 128), "kaiming"), ("f1b", (128,), "zeros"),
                 ("f2w", (128, c.n_class), "xavier"), ("f2b", (c.n_class,), "zeros")]
 

@@ -1,6 +1,7 @@
 """ResNet-20 (CIFAR, BASELINE config 3) and ResNet-50 (ImageNet-shape,
-BASELINE config 5) built from the gfx950 op set: im2col-MFMA conv,
-batch-stats BatchNorm, fused add+relu, global avgpool, linear head.
+BASELINE config 5) built from the gfx950 op set: NHWC MFMA conv (1x1 =
+pure GEMM), batch-stats BatchNorm, fused add+relu, global avgpool,
+linear head. All activations are channels-last [N, H, W, C].
 
 Flat-parameter models (models/base.py): conv weights, BN gamma/beta and
 the fc head all live in one fp32 master vector, so FL deltas/FedAvg stay
@@ -47,8 +48,8 @@ class ResNet(FlatModel):
         in_ch = 3
         s: List[Tuple[str, Tuple[int, ...], str]] = []
 
-        def conv(name, cin, cout, k):
-            s.append((f"{name}.w", (cout, cin, k, k), "kaiming"))
+        def conv(name, cin, cout, k):  # NHWC: weights [Kout, R, S, C]
+            s.append((f"{name}.w", (cout, k, k, cin), "kaiming"))
 
         def bn(name, ch):
             s.append((f"{name}.g", (ch,), "zeros"))  # filled with 1s below

@@ -146,8 +146,10 @@ def relu(x: torch.Tensor) -> torch.Tensor:
 
 
 class _Conv2dFn(torch.autograd.Function):
-    """NCHW conv via im2col + MFMA GEMM on GPU (csrc/hip/conv_im2col.hip);
-    torch fp32 oracle on CPU. Weight layout [K, C, R, S]."""
+    """NHWC conv (csrc/hip/conv2d.hip): channels-last makes 1x1 convs
+    pure MFMA GEMMs (no im2col) and RxS gathers 16-B vector copies.
+    x [N, H, W, C], w [Kout, R, S, C], y [N, OH, OW, Kout]. CPU oracle:
+    permute to NCHW, torch fp32 conv, permute back."""
 
     @staticmethod
     def forward(ctx, x, w, b, stride: int, padding: int):
@@ -162,7 +164,9 @@ class _Conv2dFn(torch.autograd.Function):
             ctx.save_for_backward(x, w, col)
             return y
         ctx.save_for_backward(x, w)
-        return F.conv2d(x, w, b, stride=stride, padding=padding)
+        y = F.conv2d(x.permute(0, 3, 1, 2), w.permute(0, 3, 1, 2), b,
+                     stride=stride, padding=padding)
+        return y.permute(0, 2, 3, 1).contiguous()
 
     @staticmethod
     def backward(ctx, dy):
@@ -173,11 +177,16 @@ class _Conv2dFn(torch.autograd.Function):
             dx, dw, db = hip_ops().conv2d_bwd(x, w, dy, ctx.stride,
                                               ctx.padding, col)
         else:
-            dx = torch.nn.grad.conv2d_input(x.shape, w, dy, stride=ctx.stride,
-                                            padding=ctx.padding)
-            dw = torch.nn.grad.conv2d_weight(x, w.shape, dy, stride=ctx.stride,
-                                             padding=ctx.padding)
-            db = dy.sum(dim=(0, 2, 3))
+            xn = x.permute(0, 3, 1, 2)
+            wn = w.permute(0, 3, 1, 2)
+            dyn = dy.permute(0, 3, 1, 2)
+            dx = torch.nn.grad.conv2d_input(
+                xn.shape, wn, dyn, stride=ctx.stride, padding=ctx.padding)
+            dw = torch.nn.grad.conv2d_weight(
+                xn, wn.shape, dyn, stride=ctx.stride, padding=ctx.padding)
+            dx = dx.permute(0, 2, 3, 1).contiguous()
+            dw = dw.permute(0, 2, 3, 1).contiguous()
+            db = dy.sum(dim=(0, 1, 2))
         return dx, dw, (db if ctx.has_bias else None), None, None
 
 
@@ -191,7 +200,9 @@ class _MaxPool2dFn(torch.autograd.Function):
         if x.is_cuda:
             y, idx = hip_ops().maxpool2d_fwd(x, kernel, stride)
         else:
-            y, idx = F.max_pool2d(x, kernel, stride, return_indices=True)
+            y, idx = F.max_pool2d(x.permute(0, 3, 1, 2), kernel, stride,
+                                  return_indices=True)
+            y = y.permute(0, 2, 3, 1).contiguous()
         ctx.save_for_backward(idx)
         ctx.in_shape = x.shape
         ctx.kernel, ctx.stride = kernel, stride
@@ -205,8 +216,10 @@ class _MaxPool2dFn(torch.autograd.Function):
             dx = hip_ops().maxpool2d_bwd(dy, idx, list(ctx.in_shape),
                                          ctx.kernel, ctx.stride)
         else:
-            dx = F.max_unpool2d(dy, idx, ctx.kernel, ctx.stride,
-                                output_size=ctx.in_shape[-2:])
+            dx = F.max_unpool2d(dy.permute(0, 3, 1, 2), idx, ctx.kernel,
+                                ctx.stride,
+                                output_size=ctx.in_shape[1:3])
+            dx = dx.permute(0, 2, 3, 1).contiguous()
         return dx, None, None
 
 
@@ -215,10 +228,10 @@ def maxpool2d(x, kernel: int = 2, stride: Optional[int] = None) -> torch.Tensor:
 
 
 class _BatchNormFn(torch.autograd.Function):
-    """BatchNorm2d in batch-stats mode (train AND eval — no running
+    """BatchNorm2d (NHWC, batch-stats mode: train AND eval — no running
     buffers; the FedBN-style simplification so the flat parameter vector
-    is exactly {gamma, beta}). GPU: hierarchical fixed-order reduction
-    kernels (csrc/hip/batchnorm.hip); CPU: fp32 oracle."""
+    is exactly {gamma, beta}). GPU: column-reduction kernels over
+    x viewed [N*H*W, C] (csrc/hip/batchnorm.hip); CPU: fp32 oracle."""
 
     @staticmethod
     def forward(ctx, x, gamma, beta, eps: float):
@@ -227,13 +240,11 @@ class _BatchNormFn(torch.autograd.Function):
                                                       False)
         else:
             xf = x.float()
-            mean = xf.mean(dim=(0, 2, 3))
-            var = xf.var(dim=(0, 2, 3), unbiased=False)
+            mean = xf.mean(dim=(0, 1, 2))
+            var = xf.var(dim=(0, 1, 2), unbiased=False)
             invstd = (var + eps).rsqrt()
-            y = ((xf - mean[None, :, None, None])
-                 * invstd[None, :, None, None]
-                 * gamma.float()[None, :, None, None]
-                 + beta.float()[None, :, None, None]).to(x.dtype)
+            y = ((xf - mean) * invstd * gamma.float()
+                 + beta.float()).to(x.dtype)
         ctx.save_for_backward(x, gamma, mean, invstd)
         return y
 
@@ -246,14 +257,11 @@ class _BatchNormFn(torch.autograd.Function):
                                                         gamma)
         else:
             xf, dyf = x.float(), dy.float()
-            n = x.numel() / x.shape[1]
-            xhat = (xf - mean[None, :, None, None]) * invstd[None, :, None,
-                                                            None]
-            sdy = dyf.sum(dim=(0, 2, 3))
-            sdyx = (dyf * xhat).sum(dim=(0, 2, 3))
-            dx = (gamma.float() * invstd)[None, :, None, None] * (
-                dyf - sdy[None, :, None, None] / n
-                - xhat * sdyx[None, :, None, None] / n)
+            n = x.numel() / x.shape[-1]
+            xhat = (xf - mean) * invstd
+            sdy = dyf.sum(dim=(0, 1, 2))
+            sdyx = (dyf * xhat).sum(dim=(0, 1, 2))
+            dx = (gamma.float() * invstd) * (dyf - sdy / n - xhat * sdyx / n)
             dx = dx.to(x.dtype)
             dgamma = sdyx.to(x.dtype)
             dbeta = sdy.to(x.dtype)
@@ -267,10 +275,10 @@ def batchnorm2d(x, gamma, beta, eps: float = 1e-5) -> torch.Tensor:
 class _GlobalAvgPoolFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
-        ctx.hw = (x.shape[2], x.shape[3])
+        ctx.hw = (x.shape[1], x.shape[2])
         if x.is_cuda:
             return hip_ops().global_avgpool_fwd(x)
-        return x.mean(dim=(2, 3))
+        return x.mean(dim=(1, 2))
 
     @staticmethod
     def backward(ctx, dy):
@@ -278,7 +286,7 @@ class _GlobalAvgPoolFn(torch.autograd.Function):
         dy = dy.contiguous()
         if dy.is_cuda:
             return hip_ops().global_avgpool_bwd(dy, h, w)
-        return (dy / (h * w))[:, :, None, None].expand(-1, -1, h, w) \
+        return (dy / (h * w))[:, None, None, :].expand(-1, h, w, -1) \
             .contiguous()
 
 

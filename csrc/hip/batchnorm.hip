@@ -1,5 +1,10 @@
-// BatchNorm2d (NCHW, bf16, batch-stats mode) + global avgpool +
+// BatchNorm2d (NHWC, bf16, batch-stats mode) + global avgpool +
 // fused residual add+ReLU — gfx950.
+//
+// NHWC makes BatchNorm a COLUMN reduction over x viewed [M, C]
+// (M = N*H*W): per-channel partials are contiguous-row sweeps with
+// 16-byte vector loads, the same shape as the GEMM colsum — no
+// per-channel strided plane walks like the NCHW formulation needed.
 //
 // The reference has no normalization (5x2 logistic regression); these
 // ops exist for the ResNet configs (BASELINE configs 3 and 5).
@@ -7,7 +12,7 @@
 // buffers) so the flat parameter vector is exactly {gamma, beta} and
 // committee scoring needs no buffer aggregation — the standard FedBN
 // simplification; deterministic because reductions are fixed-order
-// hierarchical (chunk partials reduced ascending), no atomics.
+// hierarchical (8-lane tree, chunks ascending), no atomics.
 
 #include "common.h"
 
@@ -15,110 +20,154 @@ namespace bflc {
 
 namespace {
 
-constexpr int kChunk = 4096;  // flattened (n,hw) elements per partial
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
 
-// pass 1: per-(channel, chunk) partial sum & sumsq over the N*HW domain
-__global__ void bn_stats_part_kernel(const bf16* __restrict__ x, int N,
-                                     int C, long HW, int chunks,
-                                     float* __restrict__ psum,
+constexpr int kChunkRows = 1024;  // rows of [M, C] per partial
+
+// pass 1: per-(chunk, channel) partial sum & sumsq. Block = 8 row-lanes
+// x 32 channels (fixed-order lane tree, colsum_part shape).
+__global__ void bn_stats_part_kernel(const bf16* __restrict__ x, long M,
+                                     int C, float* __restrict__ psum,
                                      float* __restrict__ psq) {
-  const int c = blockIdx.x;
-  const int chunk = blockIdx.y;
-  const long total = (long)N * HW;
-  const long j0 = (long)chunk * kChunk;
-  const long j1 = min(total, j0 + kChunk);
+  const int col = blockIdx.x * 32 + (threadIdx.x & 31);
+  const int rlane = threadIdx.x >> 5;  // 0..7
+  const long r0 = (long)blockIdx.y * kChunkRows;
+  const long r1 = min(M, r0 + kChunkRows);
   float s = 0.f, q = 0.f;
-  for (long j = j0 + threadIdx.x; j < j1; j += blockDim.x) {
-    const long n = j / HW, hw = j - n * HW;
-    const float v = b2f(x[(n * C + c) * HW + hw]);
-    s += v;
-    q += v * v;
-  }
-  s = wave_sum(s);
-  q = wave_sum(q);
-  __shared__ float ls[8], lq[8];
-  const int wid = threadIdx.x / kWave, lane = threadIdx.x % kWave;
-  if (lane == 0) { ls[wid] = s; lq[wid] = q; }
+  if (col < C)
+    for (long m = r0 + rlane; m < r1; m += 8) {
+      const float v = b2f(x[m * C + col]);
+      s += v;
+      q += v * v;
+    }
+  __shared__ float ls[8][33], lq[8][33];
+  ls[rlane][threadIdx.x & 31] = s;
+  lq[rlane][threadIdx.x & 31] = q;
   __syncthreads();
-  if (threadIdx.x == 0) {
+  if (rlane == 0 && col < C) {
     float ts = 0.f, tq = 0.f;
-    for (int w = 0; w < (int)blockDim.x / kWave; ++w) { ts += ls[w]; tq += lq[w]; }
-    psum[(long)chunk * C + c] = ts;
-    psq[(long)chunk * C + c] = tq;
+#pragma unroll
+    for (int r = 0; r < 8; ++r) {
+      ts += ls[r][threadIdx.x & 31];
+      tq += lq[r][threadIdx.x & 31];
+    }
+    psum[(long)blockIdx.y * C + col] = ts;
+    psq[(long)blockIdx.y * C + col] = tq;
   }
 }
 
-// pass 2: mean/invstd per channel (ascending chunk order: deterministic)
+// pass 2: mean/invstd per channel (8-lane tree over chunks, fixed order)
 __global__ void bn_stats_final_kernel(const float* __restrict__ psum,
                                       const float* __restrict__ psq,
                                       int chunks, int C, float count,
                                       float eps, float* __restrict__ mean,
                                       float* __restrict__ invstd) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  const int clane = threadIdx.x >> 5;
+  const int col = blockIdx.x * 32 + (threadIdx.x & 31);
   float s = 0.f, q = 0.f;
-  for (int k = 0; k < chunks; ++k) {
-    s += psum[(long)k * C + c];
-    q += psq[(long)k * C + c];
+  if (col < C)
+    for (int k = clane; k < chunks; k += 8) {
+      s += psum[(long)k * C + col];
+      q += psq[(long)k * C + col];
+    }
+  __shared__ float ls[8][33], lq[8][33];
+  ls[clane][threadIdx.x & 31] = s;
+  lq[clane][threadIdx.x & 31] = q;
+  __syncthreads();
+  if (clane == 0 && col < C) {
+    float ts = 0.f, tq = 0.f;
+#pragma unroll
+    for (int r = 0; r < 8; ++r) {
+      ts += ls[r][threadIdx.x & 31];
+      tq += lq[r][threadIdx.x & 31];
+    }
+    const float m = ts / count;
+    const float var = fmaxf(tq / count - m * m, 0.f);
+    mean[col] = m;
+    invstd[col] = rsqrtf(var + eps);
   }
-  const float m = s / count;
-  const float var = fmaxf(q / count - m * m, 0.f);
-  mean[c] = m;
-  invstd[c] = rsqrtf(var + eps);
 }
 
-// pass 3: y = (x - mean) * invstd * gamma + beta  (+optional relu)
+// pass 3: y = (x - mean) * invstd * gamma + beta (+optional relu),
+// one 16-B granule (8 consecutive channels) per iteration.
+__global__ void bn_norm_vec_kernel(const bf16* __restrict__ x,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ invstd,
+                                   const bf16* __restrict__ gamma,
+                                   const bf16* __restrict__ beta, long M,
+                                   int C, int relu, bf16* __restrict__ y) {
+  const int c8g = C / 8;
+  const long total_g = M * c8g;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total_g; g += stride) {
+    const int c8 = (int)(g % c8g) * 8;
+    const long i = (g / c8g) * C + c8;
+    const bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(&x[i]);
+    bf16x8_t out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c8 + j;
+      float f = (b2f(v[j]) - mean[c]) * invstd[c] * b2f(gamma[c]) +
+                b2f(beta[c]);
+      if (relu) f = fmaxf(f, 0.f);
+      out[j] = f2b(f);
+    }
+    *reinterpret_cast<bf16x8_t*>(&y[i]) = out;
+  }
+}
+
 __global__ void bn_norm_kernel(const bf16* __restrict__ x,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
                                const bf16* __restrict__ gamma,
-                               const bf16* __restrict__ beta, int N, int C,
-                               long HW, int relu, bf16* __restrict__ y) {
-  const long total = (long)N * C * HW;
+                               const bf16* __restrict__ beta, long M, int C,
+                               int relu, bf16* __restrict__ y) {
+  const long total = M * C;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (; i < total; i += stride) {
-    const int c = (int)((i / HW) % C);
-    float v = (b2f(x[i]) - mean[c]) * invstd[c] * b2f(gamma[c]) +
+    const int c = (int)(i % C);
+    float f = (b2f(x[i]) - mean[c]) * invstd[c] * b2f(gamma[c]) +
               b2f(beta[c]);
-    if (relu) v = fmaxf(v, 0.f);
-    y[i] = f2b(v);
+    if (relu) f = fmaxf(f, 0.f);
+    y[i] = f2b(f);
   }
 }
 
-// bwd pass 1: per-(channel, chunk) partials of sum(dy) and sum(dy*xhat)
+// bwd pass 1: per-(chunk, channel) partials of sum(dy), sum(dy*xhat)
 __global__ void bn_bwd_part_kernel(const bf16* __restrict__ x,
                                    const bf16* __restrict__ dy,
                                    const float* __restrict__ mean,
-                                   const float* __restrict__ invstd, int N,
-                                   int C, long HW, int chunks,
-                                   float* __restrict__ pdy,
+                                   const float* __restrict__ invstd, long M,
+                                   int C, float* __restrict__ pdy,
                                    float* __restrict__ pdyx) {
-  const int c = blockIdx.x;
-  const int chunk = blockIdx.y;
-  const long total = (long)N * HW;
-  const long j0 = (long)chunk * kChunk;
-  const long j1 = min(total, j0 + kChunk);
-  const float m = mean[c], is = invstd[c];
+  const int col = blockIdx.x * 32 + (threadIdx.x & 31);
+  const int rlane = threadIdx.x >> 5;
+  const long r0 = (long)blockIdx.y * kChunkRows;
+  const long r1 = min(M, r0 + kChunkRows);
   float s1 = 0.f, s2 = 0.f;
-  for (long j = j0 + threadIdx.x; j < j1; j += blockDim.x) {
-    const long n = j / HW, hw = j - n * HW;
-    const long i = (n * C + c) * HW + hw;
-    const float g = b2f(dy[i]);
-    s1 += g;
-    s2 += g * (b2f(x[i]) - m) * is;
+  if (col < C) {
+    const float mn = mean[col], is = invstd[col];
+    for (long m = r0 + rlane; m < r1; m += 8) {
+      const float g = b2f(dy[m * C + col]);
+      s1 += g;
+      s2 += g * (b2f(x[m * C + col]) - mn) * is;
+    }
   }
-  s1 = wave_sum(s1);
-  s2 = wave_sum(s2);
-  __shared__ float l1[8], l2[8];
-  const int wid = threadIdx.x / kWave, lane = threadIdx.x % kWave;
-  if (lane == 0) { l1[wid] = s1; l2[wid] = s2; }
+  __shared__ float l1[8][33], l2[8][33];
+  l1[rlane][threadIdx.x & 31] = s1;
+  l2[rlane][threadIdx.x & 31] = s2;
   __syncthreads();
-  if (threadIdx.x == 0) {
+  if (rlane == 0 && col < C) {
     float t1 = 0.f, t2 = 0.f;
-    for (int w = 0; w < (int)blockDim.x / kWave; ++w) { t1 += l1[w]; t2 += l2[w]; }
-    pdy[(long)chunk * C + c] = t1;
-    pdyx[(long)chunk * C + c] = t2;
+#pragma unroll
+    for (int r = 0; r < 8; ++r) {
+      t1 += l1[r][threadIdx.x & 31];
+      t2 += l2[r][threadIdx.x & 31];
+    }
+    pdy[(long)blockIdx.y * C + col] = t1;
+    pdyx[(long)blockIdx.y * C + col] = t2;
   }
 }
 
@@ -129,60 +178,114 @@ __global__ void bn_bwd_final_kernel(const float* __restrict__ pdy,
                                     float* __restrict__ sdyx,
                                     bf16* __restrict__ dgamma,
                                     bf16* __restrict__ dbeta) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  const int clane = threadIdx.x >> 5;
+  const int col = blockIdx.x * 32 + (threadIdx.x & 31);
   float t1 = 0.f, t2 = 0.f;
-  for (int k = 0; k < chunks; ++k) {
-    t1 += pdy[(long)k * C + c];
-    t2 += pdyx[(long)k * C + c];
+  if (col < C)
+    for (int k = clane; k < chunks; k += 8) {
+      t1 += pdy[(long)k * C + col];
+      t2 += pdyx[(long)k * C + col];
+    }
+  __shared__ float l1[8][33], l2[8][33];
+  l1[clane][threadIdx.x & 31] = t1;
+  l2[clane][threadIdx.x & 31] = t2;
+  __syncthreads();
+  if (clane == 0 && col < C) {
+    float a = 0.f, b = 0.f;
+#pragma unroll
+    for (int r = 0; r < 8; ++r) {
+      a += l1[r][threadIdx.x & 31];
+      b += l2[r][threadIdx.x & 31];
+    }
+    sdy[col] = a;
+    sdyx[col] = b;
+    dbeta[col] = f2b(a);
+    dgamma[col] = f2b(b);
   }
-  sdy[c] = t1;
-  sdyx[c] = t2;
-  dbeta[c] = f2b(t1);
-  dgamma[c] = f2b(t2);
 }
 
 // bwd pass 2: dx = gamma*invstd*(dy - sdy/cnt - xhat*sdyx/cnt)
+__global__ void bn_bwd_dx_vec_kernel(const bf16* __restrict__ x,
+                                     const bf16* __restrict__ dy,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     const bf16* __restrict__ gamma,
+                                     const float* __restrict__ sdy,
+                                     const float* __restrict__ sdyx, long M,
+                                     int C, float count,
+                                     bf16* __restrict__ dx) {
+  const int c8g = C / 8;
+  const long total_g = M * c8g;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total_g; g += stride) {
+    const int c8 = (int)(g % c8g) * 8;
+    const long i = (g / c8g) * C + c8;
+    const bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(&x[i]);
+    const bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(&dy[i]);
+    bf16x8_t out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c8 + j;
+      const float xhat = (b2f(xv[j]) - mean[c]) * invstd[c];
+      out[j] = f2b(b2f(gamma[c]) * invstd[c] *
+                   (b2f(gv[j]) - sdy[c] / count - xhat * sdyx[c] / count));
+    }
+    *reinterpret_cast<bf16x8_t*>(&dx[i]) = out;
+  }
+}
+
 __global__ void bn_bwd_dx_kernel(const bf16* __restrict__ x,
                                  const bf16* __restrict__ dy,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
                                  const bf16* __restrict__ gamma,
                                  const float* __restrict__ sdy,
-                                 const float* __restrict__ sdyx, int N,
-                                 int C, long HW, float count,
-                                 bf16* __restrict__ dx) {
-  const long total = (long)N * C * HW;
+                                 const float* __restrict__ sdyx, long M,
+                                 int C, float count, bf16* __restrict__ dx) {
+  const long total = M * C;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (; i < total; i += stride) {
-    const int c = (int)((i / HW) % C);
+    const int c = (int)(i % C);
     const float xhat = (b2f(x[i]) - mean[c]) * invstd[c];
-    const float v = b2f(gamma[c]) * invstd[c] *
-        (b2f(dy[i]) - sdy[c] / count - xhat * sdyx[c] / count);
-    dx[i] = f2b(v);
+    dx[i] = f2b(b2f(gamma[c]) * invstd[c] *
+                (b2f(dy[i]) - sdy[c] / count - xhat * sdyx[c] / count));
   }
 }
 
-// global average pool: y[n][c] = mean over HW (one wave per (n,c))
-__global__ void gap_fwd_kernel(const bf16* __restrict__ x, int NC, long HW,
+// global average pool (NHWC): y[n][c] = mean over HW. Block = 8 hw-lanes
+// x 32 channels per n (fixed-order tree).
+__global__ void gap_fwd_kernel(const bf16* __restrict__ x, long HW, int C,
                                bf16* __restrict__ y) {
-  const int nc = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
-  const int lane = threadIdx.x % kWave;
-  if (nc >= NC) return;
+  const int n = blockIdx.y;
+  const int col = blockIdx.x * 32 + (threadIdx.x & 31);
+  const int rlane = threadIdx.x >> 5;
   float s = 0.f;
-  for (long i = lane; i < HW; i += kWave) s += b2f(x[(long)nc * HW + i]);
-  s = wave_sum(s);
-  if (lane == 0) y[nc] = f2b(s / (float)HW);
+  if (col < C) {
+    const bf16* base = x + (long)n * HW * C;
+    for (long i = rlane; i < HW; i += 8) s += b2f(base[i * C + col]);
+  }
+  __shared__ float ls[8][33];
+  ls[rlane][threadIdx.x & 31] = s;
+  __syncthreads();
+  if (rlane == 0 && col < C) {
+    float t = 0.f;
+#pragma unroll
+    for (int r = 0; r < 8; ++r) t += ls[r][threadIdx.x & 31];
+    y[(long)n * C + col] = f2b(t / (float)HW);
+  }
 }
 
-__global__ void gap_bwd_kernel(const bf16* __restrict__ dy, int NC, long HW,
-                               bf16* __restrict__ dx) {
-  const long total = (long)NC * HW;
+__global__ void gap_bwd_kernel(const bf16* __restrict__ dy, long HW, int C,
+                               long total, bf16* __restrict__ dx) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (; i < total; i += stride)
-    dx[i] = f2b(b2f(dy[i / HW]) / (float)HW);
+  for (; i < total; i += stride) {
+    const int c = (int)(i % C);
+    const long n = i / (HW * C);
+    dx[i] = f2b(b2f(dy[n * C + c]) / (float)HW);
+  }
 }
 
 // fused residual add + relu: y = max(a+b, 0); bwd masks both branches
@@ -214,31 +317,37 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     torch::Tensor x, torch::Tensor gamma, torch::Tensor beta, double eps,
     bool relu) {
   CHECK_GPU(x); CHECK_CONTIG(x);
-  int N = (int)x.size(0), C = (int)x.size(1);
-  long HW = x.size(2) * x.size(3);
-  const long total = (long)N * HW;
-  const int chunks = (int)((total + kChunk - 1) / kChunk);
+  const int C = (int)x.size(-1);
+  const long M = x.numel() / C;
+  const int chunks = (int)((M + kChunkRows - 1) / kChunkRows);
   auto opts = x.options().dtype(at::kFloat);
   auto psum = torch::empty({chunks, C}, opts);
   auto psq = torch::empty({chunks, C}, opts);
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
-  hipLaunchKernelGGL(bn_stats_part_kernel, dim3(C, chunks), dim3(256), 0,
-                     cur_stream(), (const bf16*)x.data_ptr(), N, C, HW,
-                     chunks, psum.data_ptr<float>(), psq.data_ptr<float>());
-  hipLaunchKernelGGL(bn_stats_final_kernel, dim3(ceil_div(C, 256)),
-                     dim3(256), 0, cur_stream(), psum.data_ptr<float>(),
-                     psq.data_ptr<float>(), chunks, C, (float)total,
-                     (float)eps, mean.data_ptr<float>(),
-                     invstd.data_ptr<float>());
+  hipLaunchKernelGGL(bn_stats_part_kernel, dim3(ceil_div(C, 32), chunks),
+                     dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
+                     M, C, psum.data_ptr<float>(), psq.data_ptr<float>());
+  hipLaunchKernelGGL(bn_stats_final_kernel, dim3(ceil_div(C, 32)), dim3(256),
+                     0, cur_stream(), psum.data_ptr<float>(),
+                     psq.data_ptr<float>(), chunks, C, (float)M, (float)eps,
+                     mean.data_ptr<float>(), invstd.data_ptr<float>());
   auto y = torch::empty_like(x);
   auto gc = gamma.contiguous();
   auto bc = beta.contiguous();
-  hipLaunchKernelGGL(bn_norm_kernel, dim3(ew_grid(x.numel())), dim3(1024), 0,
-                     cur_stream(), (const bf16*)x.data_ptr(),
-                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     (const bf16*)gc.data_ptr(), (const bf16*)bc.data_ptr(),
-                     N, C, HW, relu ? 1 : 0, (bf16*)y.data_ptr());
+  if (C % 8 == 0)
+    hipLaunchKernelGGL(bn_norm_vec_kernel, dim3(ew_grid(x.numel() / 8)),
+                       dim3(1024), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), (const bf16*)gc.data_ptr(),
+                       (const bf16*)bc.data_ptr(), M, C, relu ? 1 : 0,
+                       (bf16*)y.data_ptr());
+  else
+    hipLaunchKernelGGL(bn_norm_kernel, dim3(ew_grid(x.numel())), dim3(1024),
+                       0, cur_stream(), (const bf16*)x.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       (const bf16*)gc.data_ptr(), (const bf16*)bc.data_ptr(),
+                       M, C, relu ? 1 : 0, (bf16*)y.data_ptr());
   HIP_CHECK(hipGetLastError());
   return {y, mean, invstd};
 }
@@ -247,10 +356,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
     torch::Tensor invstd, torch::Tensor gamma) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(dy);
-  int N = (int)x.size(0), C = (int)x.size(1);
-  long HW = x.size(2) * x.size(3);
-  const long total = (long)N * HW;
-  const int chunks = (int)((total + kChunk - 1) / kChunk);
+  const int C = (int)x.size(-1);
+  const long M = x.numel() / C;
+  const int chunks = (int)((M + kChunkRows - 1) / kChunkRows);
   auto opts = x.options().dtype(at::kFloat);
   auto pdy = torch::empty({chunks, C}, opts);
   auto pdyx = torch::empty({chunks, C}, opts);
@@ -259,36 +367,43 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
   auto dgamma = torch::empty({C}, x.options());
   auto dbeta = torch::empty({C}, x.options());
   auto gc = gamma.contiguous();
-  hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(C, chunks), dim3(256), 0,
-                     cur_stream(), (const bf16*)x.data_ptr(),
+  hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(ceil_div(C, 32), chunks),
+                     dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
                      (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
-                     invstd.data_ptr<float>(), N, C, HW, chunks,
-                     pdy.data_ptr<float>(), pdyx.data_ptr<float>());
-  hipLaunchKernelGGL(bn_bwd_final_kernel, dim3(ceil_div(C, 256)), dim3(256),
+                     invstd.data_ptr<float>(), M, C, pdy.data_ptr<float>(),
+                     pdyx.data_ptr<float>());
+  hipLaunchKernelGGL(bn_bwd_final_kernel, dim3(ceil_div(C, 32)), dim3(256),
                      0, cur_stream(), pdy.data_ptr<float>(),
                      pdyx.data_ptr<float>(), chunks, C,
                      sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
                      (bf16*)dgamma.data_ptr(), (bf16*)dbeta.data_ptr());
   auto dx = torch::empty_like(x);
-  hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(ew_grid(x.numel())), dim3(1024),
-                     0, cur_stream(), (const bf16*)x.data_ptr(),
-                     (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
-                     invstd.data_ptr<float>(), (const bf16*)gc.data_ptr(),
-                     sdy.data_ptr<float>(), sdyx.data_ptr<float>(), N, C,
-                     HW, (float)total, (bf16*)dx.data_ptr());
+  if (C % 8 == 0)
+    hipLaunchKernelGGL(bn_bwd_dx_vec_kernel, dim3(ew_grid(x.numel() / 8)),
+                       dim3(1024), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       (const bf16*)gc.data_ptr(), sdy.data_ptr<float>(),
+                       sdyx.data_ptr<float>(), M, C, (float)M,
+                       (bf16*)dx.data_ptr());
+  else
+    hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(ew_grid(x.numel())), dim3(1024),
+                       0, cur_stream(), (const bf16*)x.data_ptr(),
+                       (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), (const bf16*)gc.data_ptr(),
+                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(), M, C,
+                       (float)M, (bf16*)dx.data_ptr());
   HIP_CHECK(hipGetLastError());
   return {dx, dgamma, dbeta};
 }
 
 torch::Tensor global_avgpool_fwd(torch::Tensor x) {
   CHECK_GPU(x); CHECK_CONTIG(x);
-  int N = (int)x.size(0), C = (int)x.size(1);
-  long HW = x.size(2) * x.size(3);
+  const int N = (int)x.size(0), C = (int)x.size(-1);
+  const long HW = x.size(1) * x.size(2);
   auto y = torch::empty({N, C}, x.options());
-  const int wpb = 4;
-  hipLaunchKernelGGL(gap_fwd_kernel, dim3(ceil_div((long)N * C, wpb)),
-                     dim3(kWave * wpb), 0, cur_stream(),
-                     (const bf16*)x.data_ptr(), N * C, HW,
+  hipLaunchKernelGGL(gap_fwd_kernel, dim3(ceil_div(C, 32), N), dim3(256), 0,
+                     cur_stream(), (const bf16*)x.data_ptr(), HW, C,
                      (bf16*)y.data_ptr());
   HIP_CHECK(hipGetLastError());
   return y;
@@ -296,11 +411,11 @@ torch::Tensor global_avgpool_fwd(torch::Tensor x) {
 
 torch::Tensor global_avgpool_bwd(torch::Tensor dy, long H, long W) {
   CHECK_GPU(dy); CHECK_CONTIG(dy);
-  int N = (int)dy.size(0), C = (int)dy.size(1);
-  auto dx = torch::empty({N, C, H, W}, dy.options());
+  const int N = (int)dy.size(0), C = (int)dy.size(1);
+  auto dx = torch::empty({(long)N, H, W, (long)C}, dy.options());
   hipLaunchKernelGGL(gap_bwd_kernel, dim3(ew_grid(dx.numel())), dim3(1024),
-                     0, cur_stream(), (const bf16*)dy.data_ptr(), N * C,
-                     H * W, (bf16*)dx.data_ptr());
+                     0, cur_stream(), (const bf16*)dy.data_ptr(), H * W, C,
+                     dx.numel(), (bf16*)dx.data_ptr());
   HIP_CHECK(hipGetLastError());
   return dx;
 }

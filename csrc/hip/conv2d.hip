@@ -1,0 +1,475 @@
+// Conv2D (NHWC, bf16) for gfx950: channels-last is the MI355X-native
+// layout — the innermost C dimension makes every gather a contiguous
+// 16-byte vector access, so
+//   - 1x1 stride-1 convolutions are PURE GEMMs: x viewed [M, C] feeds
+//     the MFMA kernel directly, no im2col materialization at all
+//     (ResNet-50 is dominated by 1x1 convs; the NCHW design spent 25%
+//     of a round re-laying them out),
+//   - RxS convolutions stage through a vectorized NHWC im2col / col2im
+//     (C-contiguous copies, not scalar scatter),
+//   - dy.view(M, Kout) is free (no NCHW->MK permute kernel), and the
+//     GEMM epilogue stores NHWC output plainly (no scatter epilogue).
+// The reference has no conv at all (its model is a 5x2 logistic
+// regression, main.py:113-120); BASELINE configs 2/3/5 (FEMNIST CNN,
+// ResNet-20/50) demand Conv2D fwd/bwd as hand-written CDNA4 kernels.
+//
+// Layouts: x [N, H, W, C], w [Kout, R, S, C], y [N, OH, OW, Kout].
+// GEMM views: fwd   y2[M,Kout] = col[M,RSC] @ w2[Kout,RSC]^T
+//             dgrad dcol[M,RSC] = dy2[M,Kout] @ w2, dx = col2im(dcol)
+//             wgrad dw[Kout,RSC] = dy2^T @ col   (col cached from fwd)
+// where M = N*OH*OW, RSC = R*S*C, k = (r*S + s)*C + c.
+
+#include "common.h"
+#include "gemm_api.h"
+
+namespace bflc {
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+typedef __attribute__((ext_vector_type(8))) unsigned short u16x8_t;
+typedef __attribute__((ext_vector_type(8))) int i32x8_t;
+
+struct ConvShape {
+  int N, C, H, W, Kout, R, S, stride, pad, OH, OW;
+  __host__ __device__ long M() const { return (long)N * OH * OW; }
+  __host__ __device__ long RSC() const { return (long)R * S * C; }
+};
+
+// col[m][(r*S+s)*C + c8..] = x[n][ih][iw][c8..]  — one 16-B granule per
+// thread iteration; C % 8 == 0.
+__global__ void im2col_nhwc_vec_kernel(const bf16* __restrict__ x,
+                                       bf16* __restrict__ col, ConvShape sh,
+                                       long total_g) {
+  const int c8g = sh.C / 8;
+  const int rs = sh.R * sh.S;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total_g; g += stride) {
+    const int c8 = (int)(g % c8g) * 8;
+    const int k = (int)((g / c8g) % rs);
+    const long m = g / ((long)c8g * rs);
+    const int s = k % sh.S, r = k / sh.S;
+    const int ow = (int)(m % sh.OW), oh = (int)((m / sh.OW) % sh.OH);
+    const int n = (int)(m / ((long)sh.OW * sh.OH));
+    const int ih = oh * sh.stride - sh.pad + r;
+    const int iw = ow * sh.stride - sh.pad + s;
+    bf16x8_t v;
+    if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W) {
+      v = *reinterpret_cast<const bf16x8_t*>(
+          &x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C + c8]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = f2b(0.f);
+    }
+    *reinterpret_cast<bf16x8_t*>(&col[m * sh.RSC() + (long)k * sh.C + c8])
+        = v;
+  }
+}
+
+// Scalar variant for C % 8 != 0 (FEMNIST C=1, ResNet stems C=3): one
+// thread per (m, r, s) copies the C contiguous channels.
+__global__ void im2col_nhwc_kernel(const bf16* __restrict__ x,
+                                   bf16* __restrict__ col, ConvShape sh,
+                                   long total_mrs) {
+  const int rs = sh.R * sh.S;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const bf16 zero = f2b(0.f);
+  for (; g < total_mrs; g += stride) {
+    const int k = (int)(g % rs);
+    const long m = g / rs;
+    const int s = k % sh.S, r = k / sh.S;
+    const int ow = (int)(m % sh.OW), oh = (int)((m / sh.OW) % sh.OH);
+    const int n = (int)(m / ((long)sh.OW * sh.OH));
+    const int ih = oh * sh.stride - sh.pad + r;
+    const int iw = ow * sh.stride - sh.pad + s;
+    bf16* out = &col[m * sh.RSC() + (long)k * sh.C];
+    if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W) {
+      const bf16* src = &x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C];
+      for (int c = 0; c < sh.C; ++c) out[c] = src[c];
+    } else {
+      for (int c = 0; c < sh.C; ++c) out[c] = zero;
+    }
+  }
+}
+
+// Gather col2im (dgrad): dx[n][ih][iw][c8..] = sum over valid (r,s) of
+// dcol[m(oh,ow)][(r*S+s)*C + c8..]. Fixed (r,s) order => deterministic.
+__global__ void col2im_nhwc_vec_kernel(const bf16* __restrict__ dcol,
+                                       bf16* __restrict__ dx, ConvShape sh,
+                                       long total_g) {
+  const int c8g = sh.C / 8;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const long rsc = sh.RSC();
+  for (; g < total_g; g += stride) {
+    const int c8 = (int)(g % c8g) * 8;
+    const long i = g / c8g;  // (n, ih, iw)
+    const int iw = (int)(i % sh.W), ih = (int)((i / sh.W) % sh.H);
+    const int n = (int)(i / ((long)sh.W * sh.H));
+    float acc[8] = {};
+    for (int r = 0; r < sh.R; ++r) {
+      const int oh_num = ih + sh.pad - r;
+      if (oh_num < 0 || oh_num % sh.stride) continue;
+      const int oh = oh_num / sh.stride;
+      if (oh >= sh.OH) continue;
+      for (int s = 0; s < sh.S; ++s) {
+        const int ow_num = iw + sh.pad - s;
+        if (ow_num < 0 || ow_num % sh.stride) continue;
+        const int ow = ow_num / sh.stride;
+        if (ow >= sh.OW) continue;
+        const long m = ((long)n * sh.OH + oh) * sh.OW + ow;
+        const bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(
+            &dcol[m * rsc + ((long)(r * sh.S + s)) * sh.C + c8]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += b2f(v[j]);
+      }
+    }
+    bf16x8_t out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = f2b(acc[j]);
+    *reinterpret_cast<bf16x8_t*>(&dx[i * sh.C + c8]) = out;
+  }
+}
+
+__global__ void col2im_nhwc_kernel(const bf16* __restrict__ dcol,
+                                   bf16* __restrict__ dx, ConvShape sh,
+                                   long total) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const long rsc = sh.RSC();
+  for (; i < total; i += stride) {
+    const int c = (int)(i % sh.C);
+    const long p = i / sh.C;
+    const int iw = (int)(p % sh.W), ih = (int)((p / sh.W) % sh.H);
+    const int n = (int)(p / ((long)sh.W * sh.H));
+    float acc = 0.f;
+    for (int r = 0; r < sh.R; ++r) {
+      const int oh_num = ih + sh.pad - r;
+      if (oh_num < 0 || oh_num % sh.stride) continue;
+      const int oh = oh_num / sh.stride;
+      if (oh >= sh.OH) continue;
+      for (int s = 0; s < sh.S; ++s) {
+        const int ow_num = iw + sh.pad - s;
+        if (ow_num < 0 || ow_num % sh.stride) continue;
+        const int ow = ow_num / sh.stride;
+        if (ow >= sh.OW) continue;
+        const long m = ((long)n * sh.OH + oh) * sh.OW + ow;
+        acc += b2f(dcol[m * rsc + ((long)(r * sh.S + s)) * sh.C + c]);
+      }
+    }
+    dx[i] = f2b(acc);
+  }
+}
+
+// ---- maxpool (NHWC, C-vectorized, int32 argmax = ih*W+iw) ----
+__global__ void maxpool_nhwc_vec_fwd(const bf16* __restrict__ x,
+                                     bf16* __restrict__ y,
+                                     int* __restrict__ idx, ConvShape sh,
+                                     long total_g) {
+  const int c8g = sh.C / 8;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total_g; g += stride) {
+    const int c8 = (int)(g % c8g) * 8;
+    const long m = g / c8g;
+    const int ow = (int)(m % sh.OW), oh = (int)((m / sh.OW) % sh.OH);
+    const int n = (int)(m / ((long)sh.OW * sh.OH));
+    float best[8];
+    int bi[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { best[j] = -INFINITY; bi[j] = 0; }
+    for (int r = 0; r < sh.R; ++r) {
+      const int ih = oh * sh.stride + r;
+      if (ih >= sh.H) break;
+      for (int s = 0; s < sh.S; ++s) {
+        const int iw = ow * sh.stride + s;
+        if (iw >= sh.W) break;
+        const bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(
+            &x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C + c8]);
+        const int code = ih * sh.W + iw;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float f = b2f(v[j]);
+          if (f > best[j]) { best[j] = f; bi[j] = code; }
+        }
+      }
+    }
+    bf16x8_t out;
+    i32x8_t oidx;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { out[j] = f2b(best[j]); oidx[j] = bi[j]; }
+    *reinterpret_cast<bf16x8_t*>(&y[m * sh.C + c8]) = out;
+    *reinterpret_cast<i32x8_t*>(&idx[m * sh.C + c8]) = oidx;
+  }
+}
+
+__global__ void maxpool_nhwc_vec_bwd(const bf16* __restrict__ dy,
+                                     const int* __restrict__ idx,
+                                     bf16* __restrict__ dx, ConvShape sh,
+                                     long total_g) {
+  const int c8g = sh.C / 8;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total_g; g += stride) {
+    const int c8 = (int)(g % c8g) * 8;
+    const long i = g / c8g;  // (n, ih, iw)
+    const int iw = (int)(i % sh.W), ih = (int)((i / sh.W) % sh.H);
+    const int n = (int)(i / ((long)sh.W * sh.H));
+    const int code = ih * sh.W + iw;
+    const int oh_lo = max(0, (ih - sh.R + sh.stride) / sh.stride);
+    const int oh_hi = min(sh.OH - 1, ih / sh.stride);
+    const int ow_lo = max(0, (iw - sh.S + sh.stride) / sh.stride);
+    const int ow_hi = min(sh.OW - 1, iw / sh.stride);
+    float acc[8] = {};
+    for (int oh = oh_lo; oh <= oh_hi; ++oh)
+      for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+        const long m = ((long)n * sh.OH + oh) * sh.OW + ow;
+        const i32x8_t iv = *reinterpret_cast<const i32x8_t*>(
+            &idx[m * sh.C + c8]);
+        const bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(
+            &dy[m * sh.C + c8]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (iv[j] == code) acc[j] += b2f(dv[j]);
+      }
+    bf16x8_t out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = f2b(acc[j]);
+    *reinterpret_cast<bf16x8_t*>(&dx[i * sh.C + c8]) = out;
+  }
+}
+
+// Scalar maxpool for C % 8 != 0.
+__global__ void maxpool_nhwc_fwd(const bf16* __restrict__ x,
+                                 bf16* __restrict__ y, int* __restrict__ idx,
+                                 ConvShape sh, long total) {
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total; g += stride) {
+    const int c = (int)(g % sh.C);
+    const long m = g / sh.C;
+    const int ow = (int)(m % sh.OW), oh = (int)((m / sh.OW) % sh.OH);
+    const int n = (int)(m / ((long)sh.OW * sh.OH));
+    float best = -INFINITY;
+    int bi = 0;
+    for (int r = 0; r < sh.R; ++r) {
+      const int ih = oh * sh.stride + r;
+      if (ih >= sh.H) break;
+      for (int s = 0; s < sh.S; ++s) {
+        const int iw = ow * sh.stride + s;
+        if (iw >= sh.W) break;
+        const float f =
+            b2f(x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C + c]);
+        if (f > best) { best = f; bi = ih * sh.W + iw; }
+      }
+    }
+    y[g] = f2b(best);
+    idx[g] = bi;
+  }
+}
+
+__global__ void maxpool_nhwc_bwd(const bf16* __restrict__ dy,
+                                 const int* __restrict__ idx,
+                                 bf16* __restrict__ dx, ConvShape sh,
+                                 long total) {
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total; g += stride) {
+    const int c = (int)(g % sh.C);
+    const long i = g / sh.C;
+    const int iw = (int)(i % sh.W), ih = (int)((i / sh.W) % sh.H);
+    const int n = (int)(i / ((long)sh.W * sh.H));
+    const int code = ih * sh.W + iw;
+    const int oh_lo = max(0, (ih - sh.R + sh.stride) / sh.stride);
+    const int oh_hi = min(sh.OH - 1, ih / sh.stride);
+    const int ow_lo = max(0, (iw - sh.S + sh.stride) / sh.stride);
+    const int ow_hi = min(sh.OW - 1, iw / sh.stride);
+    float acc = 0.f;
+    for (int oh = oh_lo; oh <= oh_hi; ++oh)
+      for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+        const long m = ((long)n * sh.OH + oh) * sh.OW + ow;
+        if (idx[m * sh.C + c] == code) acc += b2f(dy[m * sh.C + c]);
+      }
+    dx[g] = f2b(acc);
+  }
+}
+
+inline int ew_grid(long n) {
+  return (int)std::min<long>((n + 255) / 256, 16384);
+}
+
+ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
+                     long stride, long pad) {
+  ConvShape sh;
+  sh.N = (int)x.size(0); sh.H = (int)x.size(1);
+  sh.W = (int)x.size(2); sh.C = (int)x.size(3);
+  sh.Kout = (int)w.size(0); sh.R = (int)w.size(1); sh.S = (int)w.size(2);
+  sh.stride = (int)stride; sh.pad = (int)pad;
+  sh.OH = (sh.H + 2 * sh.pad - sh.R) / sh.stride + 1;
+  sh.OW = (sh.W + 2 * sh.pad - sh.S) / sh.stride + 1;
+  TORCH_CHECK(w.size(3) == sh.C, "conv channel mismatch (NHWC x, KRSC w)");
+  return sh;
+}
+
+torch::Tensor im2col(const torch::Tensor& x, const ConvShape& sh) {
+  auto col = torch::empty({sh.M(), sh.RSC()}, x.options());
+  if (sh.C % 8 == 0) {
+    const long total_g = sh.M() * sh.R * sh.S * (sh.C / 8);
+    hipLaunchKernelGGL(im2col_nhwc_vec_kernel, dim3(ew_grid(total_g)),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
+                       total_g);
+  } else {
+    const long total_mrs = sh.M() * sh.R * sh.S;
+    hipLaunchKernelGGL(im2col_nhwc_kernel, dim3(ew_grid(total_mrs)),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
+                       total_mrs);
+  }
+  HIP_CHECK(hipGetLastError());
+  return col;
+}
+
+bool is_1x1_s1(const ConvShape& sh) {
+  return sh.R == 1 && sh.S == 1 && sh.stride == 1 && sh.pad == 0;
+}
+
+}  // namespace
+
+// Returns (y, col) so the autograd wrapper hands col back to wgrad
+// (recomputing im2col cost ~12% of an FL round; 288 GB HBM3E makes the
+// cache free). For 1x1 stride-1 convs col is just a VIEW of x.
+std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride,
+    long pad) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv: bf16 only");
+  auto sh = make_shape(x, w, stride, pad);
+  auto w2 = w.view({(long)sh.Kout, sh.RSC()});
+  auto bc = b.contiguous();
+  auto y = torch::empty({(long)sh.N, (long)sh.OH, (long)sh.OW,
+                         (long)sh.Kout}, x.options());
+  torch::Tensor col = is_1x1_s1(sh) ? x.view({sh.M(), (long)sh.C})
+                                    : im2col(x, sh);
+  gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &bc,
+                false, EpStore::kPlain, 0);
+  return {y, col};
+}
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         long stride, long pad) {
+  return std::get<0>(conv2d_fwd_col(x, w, b, stride, pad));
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
+    long pad, c10::optional<torch::Tensor> col_cache) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w); CHECK_CONTIG(dy);
+  auto sh = make_shape(x, w, stride, pad);
+  auto w2 = w.view({(long)sh.Kout, sh.RSC()});
+  auto dy2 = dy.view({sh.M(), (long)sh.Kout});  // NHWC: free view
+
+  // dgrad: dcol[M, RSC] = dy2[M, Kout] @ W[Kout, RSC]. Pre-transposing
+  // the (small) weight puts B in the vector-staging [N][K] layout for
+  // every tile path.
+  auto wT = transpose_bf16(w2);  // [RSC, Kout]
+  torch::Tensor dx;
+  if (is_1x1_s1(sh)) {
+    dx = torch::empty_like(x);
+    auto dxv = dx.view({sh.M(), (long)sh.C});
+    gemm_bf16_raw(dy2, wT, dxv, sh.M(), sh.RSC(), sh.Kout, false, true,
+                  nullptr, false, EpStore::kPlain, 0);
+  } else {
+    auto dcol = torch::empty({sh.M(), sh.RSC()}, x.options());
+    gemm_bf16_raw(dy2, wT, dcol, sh.M(), sh.RSC(), sh.Kout, false, true,
+                  nullptr, false, EpStore::kPlain, 0);
+    dx = torch::empty_like(x);
+    if (sh.C % 8 == 0) {
+      const long total_g = (long)sh.N * sh.H * sh.W * (sh.C / 8);
+      hipLaunchKernelGGL(col2im_nhwc_vec_kernel, dim3(ew_grid(total_g)),
+                         dim3(256), 0, cur_stream(),
+                         (const bf16*)dcol.data_ptr(),
+                         (bf16*)dx.data_ptr(), sh, total_g);
+    } else {
+      const long total = dx.numel();
+      hipLaunchKernelGGL(col2im_nhwc_kernel, dim3(ew_grid(total)), dim3(256),
+                         0, cur_stream(), (const bf16*)dcol.data_ptr(),
+                         (bf16*)dx.data_ptr(), sh, total);
+    }
+    HIP_CHECK(hipGetLastError());
+  }
+
+  // wgrad: dW[Kout, RSC] = dy2^T @ col (col reused from fwd when given)
+  auto col = col_cache.has_value()
+                 ? *col_cache
+                 : (is_1x1_s1(sh) ? x.view({sh.M(), (long)sh.C})
+                                  : im2col(x, sh));
+  auto dw = torch::empty_like(w2);
+  gemm_bf16_raw(dy2, col, dw, sh.Kout, sh.RSC(), sh.M(), true, false,
+                nullptr, false, EpStore::kPlain, 0);
+
+  auto db = colsum_bf16(dy2);
+  return {dx, dw.view(w.sizes()), db};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
+                                                       long kernel,
+                                                       long stride) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  ConvShape sh;
+  sh.N = (int)x.size(0); sh.H = (int)x.size(1);
+  sh.W = (int)x.size(2); sh.C = (int)x.size(3);
+  sh.R = sh.S = (int)kernel; sh.stride = (int)stride; sh.pad = 0;
+  sh.Kout = sh.C;
+  sh.OH = (sh.H - sh.R) / sh.stride + 1;
+  sh.OW = (sh.W - sh.S) / sh.stride + 1;
+  auto y = torch::empty({(long)sh.N, (long)sh.OH, (long)sh.OW, (long)sh.C},
+                        x.options());
+  auto idx = torch::empty_like(y, y.options().dtype(at::kInt));
+  if (sh.C % 8 == 0) {
+    const long total_g = sh.M() * (sh.C / 8);
+    hipLaunchKernelGGL(maxpool_nhwc_vec_fwd, dim3(ew_grid(total_g)),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), (bf16*)y.data_ptr(),
+                       idx.data_ptr<int>(), sh, total_g);
+  } else {
+    const long total = y.numel();
+    hipLaunchKernelGGL(maxpool_nhwc_fwd, dim3(ew_grid(total)), dim3(256), 0,
+                       cur_stream(), (const bf16*)x.data_ptr(),
+                       (bf16*)y.data_ptr(), idx.data_ptr<int>(), sh, total);
+  }
+  HIP_CHECK(hipGetLastError());
+  return {y, idx};
+}
+
+torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx,
+                            std::vector<long> in_shape, long kernel,
+                            long stride) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(idx);
+  ConvShape sh;
+  sh.N = (int)in_shape[0]; sh.H = (int)in_shape[1];
+  sh.W = (int)in_shape[2]; sh.C = (int)in_shape[3];
+  sh.R = sh.S = (int)kernel; sh.stride = (int)stride; sh.pad = 0;
+  sh.Kout = sh.C;
+  sh.OH = (int)dy.size(1); sh.OW = (int)dy.size(2);
+  auto dx = torch::empty({(long)sh.N, (long)sh.H, (long)sh.W, (long)sh.C},
+                         dy.options());
+  if (sh.C % 8 == 0) {
+    const long total_g = (long)sh.N * sh.H * sh.W * (sh.C / 8);
+    hipLaunchKernelGGL(maxpool_nhwc_vec_bwd, dim3(ew_grid(total_g)),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)dy.data_ptr(), idx.data_ptr<int>(),
+                       (bf16*)dx.data_ptr(), sh, total_g);
+  } else {
+    const long total = dx.numel();
+    hipLaunchKernelGGL(maxpool_nhwc_bwd, dim3(ew_grid(total)), dim3(256), 0,
+                       cur_stream(), (const bf16*)dy.data_ptr(),
+                       idx.data_ptr<int>(), (bf16*)dx.data_ptr(), sh, total);
+  }
+  HIP_CHECK(hipGetLastError());
+  return dx;
+}
+
+}  // namespace bflc

@@ -25,4 +25,8 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
 // colsum: out[n] = sum_m X[m,n]  (bias gradient)
 torch::Tensor colsum_bf16(const torch::Tensor& X);
 
+// LDS-tiled bf16 transpose: out[C][R] = X[R][C]^T (16-B coalesced both
+// sides). Used to put GEMM operands into the vector-staging layout.
+torch::Tensor transpose_bf16(const torch::Tensor& X);
+
 }  // namespace bflc

@@ -471,16 +471,6 @@ __global__ void transpose_bf16_kernel(const bf16* __restrict__ in,
   }
 }
 
-torch::Tensor transpose_bf16(const torch::Tensor& X) {
-  const long R = X.size(0), C = X.size(1);
-  auto out = torch::empty({C, R}, X.options());
-  dim3 grid((unsigned)ceil_div(C, 64), (unsigned)ceil_div(R, 64));
-  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, cur_stream(),
-                     (const bf16*)X.data_ptr(), (bf16*)out.data_ptr(), R, C);
-  HIP_CHECK(hipGetLastError());
-  return out;
-}
-
 // Deterministic split-K reduce: out = sum_s partial[s] (ascending s),
 // then the fused epilogue (bias/relu/NCHW).
 // Block = 8 slice-lanes x 32 output elements; each lane sums slices
@@ -592,6 +582,16 @@ TileCfg pick_tile(long M, long N) {
 }
 
 }  // namespace
+
+torch::Tensor transpose_bf16(const torch::Tensor& X) {
+  const long R = X.size(0), C = X.size(1);
+  auto out = torch::empty({C, R}, X.options());
+  dim3 grid((unsigned)ceil_div(C, 64), (unsigned)ceil_div(R, 64));
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const bf16*)X.data_ptr(), (bf16*)out.data_ptr(), R, C);
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
 
 void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
                    torch::Tensor& C, long M, long N, long K, bool ta, bool tb,

@@ -32,32 +32,35 @@ def test_softmax_ce_matches_torch():
 
 
 def test_conv2d_matches_torch():
+    # NHWC x [N,H,W,C], w [Kout,R,S,C] vs the torch NCHW reference
     torch.manual_seed(0)
-    x = torch.randn(4, 3, 14, 14, requires_grad=True)
+    x = torch.randn(4, 14, 14, 3, requires_grad=True)
     w = torch.randn(8, 3, 3, 3, requires_grad=True)
     b = torch.randn(8, requires_grad=True)
     y = O.conv2d(x, w, b, stride=1, padding=1)
-    ref = F.conv2d(x, w, b, stride=1, padding=1)
+    ref = F.conv2d(x.permute(0, 3, 1, 2), w.permute(0, 3, 1, 2), b,
+                   stride=1, padding=1).permute(0, 2, 3, 1)
     assert torch.allclose(y, ref, atol=1e-5)
     g = torch.randn_like(y)
     (y * g).sum().backward()
     x2 = x.detach().clone().requires_grad_(True)
     w2 = w.detach().clone().requires_grad_(True)
     b2 = b.detach().clone().requires_grad_(True)
-    (F.conv2d(x2, w2, b2, stride=1, padding=1) * g).sum().backward()
+    (F.conv2d(x2.permute(0, 3, 1, 2), w2.permute(0, 3, 1, 2), b2,
+              stride=1, padding=1).permute(0, 2, 3, 1) * g).sum().backward()
     assert torch.allclose(x.grad, x2.grad, atol=1e-4)
     assert torch.allclose(w.grad, w2.grad, atol=1e-4)
     assert torch.allclose(b.grad, b2.grad, atol=1e-4)
 
 
 def test_maxpool_matches_torch():
-    x = torch.randn(2, 4, 8, 8, requires_grad=True)
+    x = torch.randn(2, 8, 8, 4, requires_grad=True)  # NHWC
     y = O.maxpool2d(x, 2)
-    ref = F.max_pool2d(x, 2)
+    ref = F.max_pool2d(x.permute(0, 3, 1, 2), 2).permute(0, 2, 3, 1)
     assert torch.equal(y, ref)
     y.sum().backward()
     x2 = x.detach().clone().requires_grad_(True)
-    F.max_pool2d(x2, 2).sum().backward()
+    F.max_pool2d(x2.permute(0, 3, 1, 2), 2).sum().backward()
     assert torch.equal(x.grad, x2.grad)
 
 

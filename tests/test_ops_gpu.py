@@ -95,43 +95,47 @@ class TestConv:
         (2, 3, 32, 16, 7, 2, 3),   # resnet50 stem 7x7/2
     ])
     def test_fwd_bwd(self, n, c, h, k, r, stride, pad):
+        # NHWC x [N,H,W,C], w [Kout,R,S,C]; torch NCHW reference
         torch.manual_seed(3)
-        x = torch.randn(n, c, h, h)
-        w = torch.randn(k, c, r, r) * (1.0 / math.sqrt(c * r * r))
+        x = torch.randn(n, h, h, c)
+        w = torch.randn(k, r, r, c) * (1.0 / math.sqrt(c * r * r))
         b = torch.randn(k)
         y = hip().conv2d_fwd(bf(x), bf(w), bf(b), stride, pad)
         xf, wf, bfl = (bf(t).float().cpu() for t in (x, w, b))
-        ref = torch.nn.functional.conv2d(xf, wf, bfl, stride=stride,
-                                         padding=pad)
+        xn, wn = xf.permute(0, 3, 1, 2), wf.permute(0, 3, 1, 2)
+        ref = torch.nn.functional.conv2d(xn, wn, bfl, stride=stride,
+                                         padding=pad).permute(0, 2, 3, 1)
         assert y.shape == ref.shape
         assert_close(y, ref)
 
         dy = torch.randn_like(ref)
         dx, dw, db = hip().conv2d_bwd(bf(x), bf(w), bf(dy), stride, pad)
-        x2 = xf.clone().requires_grad_(True)
-        w2 = wf.clone().requires_grad_(True)
+        x2 = xn.clone().requires_grad_(True)
+        w2 = wn.clone().requires_grad_(True)
         b2 = bfl.clone().requires_grad_(True)
         out = torch.nn.functional.conv2d(x2, w2, b2, stride=stride,
                                          padding=pad)
-        (out * bf(dy).float().cpu()).sum().backward()
-        assert_close(dx, x2.grad, rel=0.03)
-        assert_close(dw, w2.grad, rel=0.03)
+        (out * bf(dy).float().cpu().permute(0, 3, 1, 2)).sum().backward()
+        assert_close(dx, x2.grad.permute(0, 2, 3, 1), rel=0.03)
+        assert_close(dw, w2.grad.permute(0, 2, 3, 1), rel=0.03)
         assert_close(db, b2.grad, rel=0.03)
 
 
 class TestPoolReluAcc:
     def test_maxpool(self):
         torch.manual_seed(4)
-        x = torch.randn(3, 8, 14, 14)
+        x = torch.randn(3, 14, 14, 8)  # NHWC
         y, idx = hip().maxpool2d_fwd(bf(x), 2, 2)
+        xn = bf(x).float().cpu().permute(0, 3, 1, 2)
         ref, ridx = torch.nn.functional.max_pool2d(
-            bf(x).float().cpu(), 2, 2, return_indices=True)
-        assert_close(y, ref, rel=0.01)
+            xn, 2, 2, return_indices=True)
+        assert_close(y, ref.permute(0, 2, 3, 1), rel=0.01)
         dy = torch.randn_like(ref)
-        dx = hip().maxpool2d_bwd(bf(dy), idx, [3, 8, 14, 14], 2, 2)
+        dx = hip().maxpool2d_bwd(bf(dy.permute(0, 2, 3, 1).contiguous()),
+                                 idx, [3, 14, 14, 8], 2, 2)
         ref_dx = torch.nn.functional.max_unpool2d(
             bf(dy).float().cpu(), ridx, 2, 2, output_size=(14, 14))
-        assert_close(dx, ref_dx, rel=0.01)
+        assert_close(dx, ref_dx.permute(0, 2, 3, 1), rel=0.01)
 
     def test_relu(self):
         x = torch.randn(1000)

@@ -11,27 +11,29 @@ from bflc_amd.ops import functional as O
 class TestBatchNormOracle:
     def test_fwd_bwd_matches_torch(self):
         torch.manual_seed(0)
-        x = torch.randn(4, 8, 7, 7, requires_grad=True)
+        x = torch.randn(4, 7, 7, 8, requires_grad=True)  # NHWC
         g = torch.randn(8, requires_grad=True)
         b = torch.randn(8, requires_grad=True)
         y = O.batchnorm2d(x, g, b)
-        ref = F.batch_norm(x, None, None, g, b, training=True, eps=1e-5)
+        ref = F.batch_norm(x.permute(0, 3, 1, 2), None, None, g, b,
+                           training=True, eps=1e-5).permute(0, 2, 3, 1)
         assert torch.allclose(y, ref, atol=1e-5)
         dy = torch.randn_like(y)
         (y * dy).sum().backward()
         x2 = x.detach().clone().requires_grad_(True)
         g2 = g.detach().clone().requires_grad_(True)
         b2 = b.detach().clone().requires_grad_(True)
-        (F.batch_norm(x2, None, None, g2, b2, training=True) * dy).sum() \
+        (F.batch_norm(x2.permute(0, 3, 1, 2), None, None, g2, b2,
+                      training=True).permute(0, 2, 3, 1) * dy).sum() \
             .backward()
         assert torch.allclose(x.grad, x2.grad, atol=1e-4)
         assert torch.allclose(g.grad, g2.grad, atol=1e-4)
         assert torch.allclose(b.grad, b2.grad, atol=1e-4)
 
     def test_gap_and_add_relu(self):
-        x = torch.randn(2, 4, 5, 5, requires_grad=True)
+        x = torch.randn(2, 5, 5, 4, requires_grad=True)  # NHWC
         y = O.global_avgpool(x)
-        assert torch.allclose(y, x.mean(dim=(2, 3)))
+        assert torch.allclose(y, x.mean(dim=(1, 2)))
         y.sum().backward()
         assert torch.allclose(x.grad, torch.full_like(x, 1 / 25.0))
 
@@ -51,7 +53,7 @@ class TestResNet20:
                        comm_count=1, needed_update_count=1,
                        aggregate_count=1)
         m = build_model(cfg, torch.device("cpu"))
-        x = torch.randn(4, 3, 32, 32)
+        x = torch.randn(4, 32, 32, 3)
         y = torch.randint(0, 10, (4,))
         logits = m.forward(x)
         assert logits.shape == (4, 10)
@@ -68,7 +70,7 @@ class TestResNet20:
                        aggregate_count=1, learning_rate=0.05)
         m = build_model(cfg, torch.device("cpu"))
         torch.manual_seed(0)
-        x = torch.randn(16, 3, 32, 32)
+        x = torch.randn(16, 32, 32, 3)
         y = torch.randint(0, 4, (16,))
         losses = []
         for _ in range(6):
@@ -88,7 +90,7 @@ class TestResNet50:
         m = build_model(cfg, torch.device("cpu"))
         # 64x64 input keeps the CPU test fast; the conv/pool stack and
         # global avgpool handle any spatial size
-        x = torch.randn(2, 3, 64, 64)
+        x = torch.randn(2, 64, 64, 3)
         y = torch.randint(0, 10, (2,))
         loss = m.loss(x, y)
         loss.backward()

@@ -27,15 +27,16 @@ class TestBatchNormGPU:
     @pytest.mark.parametrize("n,c,h", [(4, 8, 7), (16, 64, 8), (2, 3, 32)])
     def test_fwd_bwd(self, n, c, h):
         torch.manual_seed(0)
-        x = torch.randn(n, c, h, h)
+        x = torch.randn(n, h, h, c)  # NHWC
         g = torch.rand(c) + 0.5
         b = torch.randn(c)
         y, mean, invstd = hip().batchnorm_fwd(bf(x), bf(g), bf(b), 1e-5,
                                               False)
         xf = bf(x).float().cpu()
         gf, bfl = bf(g).float().cpu(), bf(b).float().cpu()
-        ref = torch.nn.functional.batch_norm(xf, None, None, gf, bfl,
-                                             training=True, eps=1e-5)
+        ref = torch.nn.functional.batch_norm(
+            xf.permute(0, 3, 1, 2), None, None, gf, bfl, training=True,
+            eps=1e-5).permute(0, 2, 3, 1)
         assert_close(y, ref)
 
         dy = torch.randn_like(ref)
@@ -44,15 +45,16 @@ class TestBatchNormGPU:
         x2 = xf.clone().requires_grad_(True)
         g2 = gf.clone().requires_grad_(True)
         b2 = bfl.clone().requires_grad_(True)
-        out = torch.nn.functional.batch_norm(x2, None, None, g2, b2,
-                                             training=True, eps=1e-5)
+        out = torch.nn.functional.batch_norm(
+            x2.permute(0, 3, 1, 2), None, None, g2, b2, training=True,
+            eps=1e-5).permute(0, 2, 3, 1)
         (out * bf(dy).float().cpu()).sum().backward()
         assert_close(dx, x2.grad, rel=0.05)
         assert_close(dgamma, g2.grad, rel=0.05)
         assert_close(dbeta, b2.grad, rel=0.05)
 
     def test_fwd_deterministic(self):
-        x = bf(torch.randn(8, 16, 14, 14))
+        x = bf(torch.randn(8, 14, 14, 16))
         g, b = bf(torch.rand(16)), bf(torch.randn(16))
         y1, m1, i1 = hip().batchnorm_fwd(x, g, b, 1e-5, False)
         y2, m2, i2 = hip().batchnorm_fwd(x, g, b, 1e-5, False)
@@ -61,13 +63,13 @@ class TestBatchNormGPU:
 
 class TestPoolAddRelu:
     def test_gap(self):
-        x = torch.randn(3, 5, 9, 9)
+        x = torch.randn(3, 9, 9, 5)  # NHWC
         y = hip().global_avgpool_fwd(bf(x))
-        assert_close(y, bf(x).float().cpu().mean(dim=(2, 3)), rel=0.01)
+        assert_close(y, bf(x).float().cpu().mean(dim=(1, 2)), rel=0.01)
         dy = torch.randn(3, 5)
         dx = hip().global_avgpool_bwd(bf(dy), 9, 9)
-        ref = (bf(dy).float().cpu() / 81)[:, :, None, None] \
-            .expand(-1, -1, 9, 9)
+        ref = (bf(dy).float().cpu() / 81)[:, None, None, :] \
+            .expand(-1, 9, 9, -1)
         assert_close(dx, ref, rel=0.01)
 
     def test_add_relu(self):
@@ -81,18 +83,21 @@ class TestPoolAddRelu:
         assert_close(da, refd, rel=0.01)
 
     def test_overlapping_maxpool(self):
-        # ResNet-50 stem: k=3 s=2 overlapping windows
+        # ResNet-50 stem: k=3 s=2 overlapping windows (NHWC)
         torch.manual_seed(1)
-        x = torch.randn(2, 4, 13, 13)
+        x = torch.randn(2, 13, 13, 4)
         y, idx = hip().maxpool2d_fwd(bf(x), 3, 2)
-        ref = torch.nn.functional.max_pool2d(bf(x).float().cpu(), 3, 2)
+        ref = torch.nn.functional.max_pool2d(
+            bf(x).float().cpu().permute(0, 3, 1, 2), 3, 2) \
+            .permute(0, 2, 3, 1)
         assert_close(y, ref, rel=0.01)
         dy = torch.randn_like(ref)
-        dx = hip().maxpool2d_bwd(bf(dy), idx, [2, 4, 13, 13], 3, 2)
-        x2 = bf(x).float().cpu().requires_grad_(True)
+        dx = hip().maxpool2d_bwd(bf(dy), idx, [2, 13, 13, 4], 3, 2)
+        x2 = bf(x).float().cpu().permute(0, 3, 1, 2).detach() \
+            .requires_grad_(True)
         (torch.nn.functional.max_pool2d(x2, 3, 2) *
-         bf(dy).float().cpu()).sum().backward()
-        assert_close(dx, x2.grad, rel=0.01)
+         bf(dy).float().cpu().permute(0, 3, 1, 2)).sum().backward()
+        assert_close(dx, x2.grad.permute(0, 2, 3, 1), rel=0.01)
 
 
 class TestResNetGPU:
@@ -111,7 +116,7 @@ class TestResNetGPU:
                        aggregate_count=1, learning_rate=lr)
         m = build_model(cfg, torch.device(DEV))
         torch.manual_seed(0)
-        x = torch.randn(16, 3, hw, hw)
+        x = torch.randn(16, hw, hw, 3)
         y = torch.randint(0, nclass, (16,), device=DEV)
         losses = []
         for _ in range(steps):
