@@ -3,10 +3,13 @@
 FL operates on flat parameter vectors: delta extraction, candidate
 reconstruction, FedAvg, and the RCCL all-gather all want ONE contiguous
 buffer per model. So every model here owns a single
-fp32 master buffer `flat`; layer weights are autograd views into it.
-loss.backward() therefore accumulates the WHOLE gradient into
-`flat.grad` — one fused SGD/Adam kernel updates the entire model, and
-`(flat0 - flat) / lr` is one AXPY (reference main.py:153-154).
+fp32 master buffer `flat`; layer weights are per-parameter autograd
+LEAVES whose storage aliases the flat compute buffer (parameter-sized
+grads — a single flat leaf made autograd materialize full-model-sized
+slice-backward grads per parameter). grad_flat() cats the per-parameter
+grads into one contiguous vector, one fused SGD/Adam kernel updates the
+entire model, and `(flat0 - flat) / lr` is one AXPY (reference
+main.py:153-154).
 """
 from __future__ import annotations
 
@@ -44,12 +47,24 @@ class FlatModel:
         # update master + shadow in one pass — no per-parameter casts.
         self.flat = torch.zeros(off, dtype=torch.float32, device=self.device)
         if self.compute_dtype == torch.float32:
-            self.flat.requires_grad_(True)
             self.cflat = self.flat
         else:
             self.cflat = torch.zeros(off, dtype=self.compute_dtype,
                                      device=self.device)
-            self.cflat.requires_grad_(True)
+        # Per-parameter autograd LEAVES aliasing the flat compute buffer
+        # (set_ shares storage). With ONE flat leaf, every parameter's
+        # slice-backward materialized a full-numel zeros + add per
+        # parameter per backward (~160 x 25M-element passes on
+        # ResNet-50, ~20% of a round); per-parameter leaves make each
+        # grad parameter-sized, and grad_flat() cats them back into one
+        # contiguous vector matching the offset layout.
+        self._params: Dict[str, torch.Tensor] = {}
+        for name, shape, _ in self._specs:
+            o, _ = self._offsets[name]
+            t = torch.empty(0, dtype=self.compute_dtype, device=self.device)
+            t.set_(self.cflat, o, shape)
+            t.requires_grad_(True)
+            self._params[name] = t
 
     # -- subclass interface -------------------------------------------------
     def specs(self) -> Sequence[Tuple[str, Tuple[int, ...], str]]:
@@ -99,11 +114,9 @@ class FlatModel:
                 self.cflat.data.copy_(self.flat.detach())
 
     def p(self, name: str) -> torch.Tensor:
-        """Autograd view of a parameter in compute dtype (fresh per call:
-        views must be re-derived after in-place updates)."""
-        off, shape = self._offsets[name]
-        n = int(math.prod(shape))
-        return self.cflat[off:off + n].view(shape)
+        """Leaf parameter tensor (compute dtype) aliasing the flat
+        buffer; in-place flat updates are visible through it."""
+        return self._params[name]
 
     def get_flat(self) -> torch.Tensor:
         return self.flat.detach().clone()
@@ -114,14 +127,22 @@ class FlatModel:
         self._sync_shadow()
 
     def zero_grad(self) -> None:
-        if self.cflat.grad is not None:
-            self.cflat.grad.detach_()
-            self.cflat.grad.zero_()
+        for t in self._params.values():
+            t.grad = None
+
+    def grad_flat(self) -> torch.Tensor:
+        """All parameter grads as ONE contiguous vector (spec order ==
+        offset order), built with a single cat kernel."""
+        gs = []
+        for name, shape, _ in self._specs:
+            g = self._params[name].grad
+            assert g is not None, f"no grad for {name}"
+            gs.append(g.reshape(-1))
+        return torch.cat(gs)
 
     # -- fused optimizer steps (fp32 master + shadow refresh) ----------
     def sgd_step(self, lr: float) -> None:
-        g = self.cflat.grad
-        assert g is not None
+        g = self.grad_flat()
         if self.cflat is self.flat:
             from bflc_amd.ops.functional import sgd_step_
             sgd_step_(self.flat.data, g, lr)
@@ -136,8 +157,7 @@ class FlatModel:
 
     def adam_step(self, m: torch.Tensor, v: torch.Tensor, step: int,
                   lr: float) -> None:
-        g = self.cflat.grad
-        assert g is not None
+        g = self.grad_flat()
         if self.cflat is self.flat:
             from bflc_amd.ops.functional import adam_step_
             adam_step_(self.flat.data, g, m, v, step, lr)

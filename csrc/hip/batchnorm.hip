@@ -24,8 +24,118 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
 
 constexpr int kChunkRows = 1024;  // rows of [M, C] per partial
 
-// pass 1: per-(chunk, channel) partial sum & sumsq. Block = 8 row-lanes
-// x 32 channels (fixed-order lane tree, colsum_part shape).
+// Vectorized pass 1 (C % 8 == 0): 256 threads = RL row-lanes x G
+// 16-B channel granules (G = pow2 <= min(C/8, 256), RL = 256/G), each
+// thread accumulating 8 channel partials from vector loads; fixed
+// pairwise halving tree over the row-lanes (structure-deterministic).
+__global__ void bn_stats_part_vec_kernel(const bf16* __restrict__ x, long M,
+                                         int C, int G,
+                                         float* __restrict__ psum,
+                                         float* __restrict__ psq) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 v8;
+  const int gi = threadIdx.x % G, rl = threadIdx.x / G;
+  const int RL = blockDim.x / G;
+  const int c8 = (blockIdx.x * G + gi) * 8;
+  const long r0 = (long)blockIdx.y * kChunkRows;
+  const long r1 = min(M, r0 + kChunkRows);
+  float sj[8] = {}, qj[8] = {};
+  if (c8 < C)
+    for (long m = r0 + rl; m < r1; m += RL) {
+      const v8 v = *reinterpret_cast<const v8*>(&x[m * C + c8]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = b2f(v[j]);
+        sj[j] += f;
+        qj[j] += f * f;
+      }
+    }
+  __shared__ float rs[256][8], rq[256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    rs[threadIdx.x][j] = sj[j];
+    rq[threadIdx.x][j] = qj[j];
+  }
+  __syncthreads();
+  for (int h = RL >> 1; h > 0; h >>= 1) {
+    if (rl < h) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        rs[rl * G + gi][j] += rs[(rl + h) * G + gi][j];
+        rq[rl * G + gi][j] += rq[(rl + h) * G + gi][j];
+      }
+    }
+    __syncthreads();
+  }
+  if (rl == 0 && c8 < C) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      psum[(long)blockIdx.y * C + c8 + j] = rs[gi][j];
+      psq[(long)blockIdx.y * C + c8 + j] = rq[gi][j];
+    }
+  }
+}
+
+// bwd vectorized pass 1: partials of sum(dy), sum(dy*xhat)
+__global__ void bn_bwd_part_vec_kernel(const bf16* __restrict__ x,
+                                       const bf16* __restrict__ dy,
+                                       const float* __restrict__ mean,
+                                       const float* __restrict__ invstd,
+                                       long M, int C, int G,
+                                       float* __restrict__ pdy,
+                                       float* __restrict__ pdyx) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 v8;
+  const int gi = threadIdx.x % G, rl = threadIdx.x / G;
+  const int RL = blockDim.x / G;
+  const int c8 = (blockIdx.x * G + gi) * 8;
+  const long r0 = (long)blockIdx.y * kChunkRows;
+  const long r1 = min(M, r0 + kChunkRows);
+  float s1[8] = {}, s2[8] = {};
+  float mn[8], is[8];
+  if (c8 < C) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mn[j] = mean[c8 + j];
+      is[j] = invstd[c8 + j];
+    }
+    for (long m = r0 + rl; m < r1; m += RL) {
+      const v8 xv = *reinterpret_cast<const v8*>(&x[m * C + c8]);
+      const v8 gv = *reinterpret_cast<const v8*>(&dy[m * C + c8]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float g = b2f(gv[j]);
+        s1[j] += g;
+        s2[j] += g * (b2f(xv[j]) - mn[j]) * is[j];
+      }
+    }
+  }
+  __shared__ float rs[256][8], rq[256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    rs[threadIdx.x][j] = s1[j];
+    rq[threadIdx.x][j] = s2[j];
+  }
+  __syncthreads();
+  for (int h = RL >> 1; h > 0; h >>= 1) {
+    if (rl < h) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        rs[rl * G + gi][j] += rs[(rl + h) * G + gi][j];
+        rq[rl * G + gi][j] += rq[(rl + h) * G + gi][j];
+      }
+    }
+    __syncthreads();
+  }
+  if (rl == 0 && c8 < C) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      pdy[(long)blockIdx.y * C + c8 + j] = rs[gi][j];
+      pdyx[(long)blockIdx.y * C + c8 + j] = rq[gi][j];
+    }
+  }
+}
+
+// pass 1 (scalar fallback, C % 8 != 0): per-(chunk, channel) partial
+// sum & sumsq. Block = 8 row-lanes x 32 channels.
 __global__ void bn_stats_part_kernel(const bf16* __restrict__ x, long M,
                                      int C, float* __restrict__ psum,
                                      float* __restrict__ psq) {
@@ -311,6 +421,14 @@ inline int ew_grid(long n) {
   return (int)std::min<long>((n + 1023) / 1024, 8192);
 }
 
+// largest power of two <= min(C/8, 256) — the granule-lane count per
+// block for the vectorized column-reduction kernels
+inline int granule_lanes(int C) {
+  int g = 1;
+  while (g * 2 <= std::min(C / 8, 256)) g *= 2;
+  return g;
+}
+
 }  // namespace
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
@@ -325,9 +443,17 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
   auto psq = torch::empty({chunks, C}, opts);
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
-  hipLaunchKernelGGL(bn_stats_part_kernel, dim3(ceil_div(C, 32), chunks),
-                     dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
-                     M, C, psum.data_ptr<float>(), psq.data_ptr<float>());
+  if (C % 8 == 0) {
+    const int G = granule_lanes(C);
+    hipLaunchKernelGGL(bn_stats_part_vec_kernel,
+                       dim3(ceil_div(C / 8, G), chunks), dim3(256), 0,
+                       cur_stream(), (const bf16*)x.data_ptr(), M, C, G,
+                       psum.data_ptr<float>(), psq.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(bn_stats_part_kernel, dim3(ceil_div(C, 32), chunks),
+                       dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
+                       M, C, psum.data_ptr<float>(), psq.data_ptr<float>());
+  }
   hipLaunchKernelGGL(bn_stats_final_kernel, dim3(ceil_div(C, 32)), dim3(256),
                      0, cur_stream(), psum.data_ptr<float>(),
                      psq.data_ptr<float>(), chunks, C, (float)M, (float)eps,
@@ -367,11 +493,21 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
   auto dgamma = torch::empty({C}, x.options());
   auto dbeta = torch::empty({C}, x.options());
   auto gc = gamma.contiguous();
-  hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(ceil_div(C, 32), chunks),
-                     dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
-                     (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
-                     invstd.data_ptr<float>(), M, C, pdy.data_ptr<float>(),
-                     pdyx.data_ptr<float>());
+  if (C % 8 == 0) {
+    const int G = granule_lanes(C);
+    hipLaunchKernelGGL(bn_bwd_part_vec_kernel,
+                       dim3(ceil_div(C / 8, G), chunks), dim3(256), 0,
+                       cur_stream(), (const bf16*)x.data_ptr(),
+                       (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), M, C, G,
+                       pdy.data_ptr<float>(), pdyx.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(ceil_div(C, 32), chunks),
+                       dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
+                       (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), M, C,
+                       pdy.data_ptr<float>(), pdyx.data_ptr<float>());
+  }
   hipLaunchKernelGGL(bn_bwd_final_kernel, dim3(ceil_div(C, 32)), dim3(256),
                      0, cur_stream(), pdy.data_ptr<float>(),
                      pdyx.data_ptr<float>(), chunks, C,

@@ -518,6 +518,43 @@ __global__ void splitk_reduce_kernel(const float* __restrict__ Cpart, int S,
 // row-walk per thread was latency-bound at ~300us/call).
 constexpr int kColsumRows = 1024;
 
+// Vectorized pass 1 (N % 8 == 0): RL row-lanes x G 16-B column
+// granules per 256-thread block, fixed pairwise halving tree over the
+// row-lanes (structure-deterministic).
+__global__ void colsum_part_vec_kernel(const bf16* __restrict__ X, long M,
+                                       long N, int G,
+                                       float* __restrict__ part) {
+  const int gi = threadIdx.x % G, rl = threadIdx.x / G;
+  const int RL = (int)blockDim.x / G;
+  const long c8 = ((long)blockIdx.x * G + gi) * 8;
+  const long r0 = (long)blockIdx.y * kColsumRows;
+  const long r1 = min(M, r0 + kColsumRows);
+  float sj[8] = {};
+  if (c8 < N)
+    for (long m = r0 + rl; m < r1; m += RL) {
+      const bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(&X[m * N + c8]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) sj[j] += b2f(v[j]);
+    }
+  __shared__ float rs[256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) rs[threadIdx.x][j] = sj[j];
+  __syncthreads();
+  for (int h = RL >> 1; h > 0; h >>= 1) {
+    if (rl < h) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        rs[rl * G + gi][j] += rs[(rl + h) * G + gi][j];
+    }
+    __syncthreads();
+  }
+  if (rl == 0 && c8 < N) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      part[(long)blockIdx.y * N + c8 + j] = rs[gi][j];
+  }
+}
+
 __global__ void colsum_part_kernel(const bf16* __restrict__ X, long M, long N,
                                    float* __restrict__ part) {
   const long col = (long)blockIdx.x * 32 + (threadIdx.x & 31);
@@ -660,8 +697,10 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   long S = 1;
   if (tiles < 512 && K >= 4 * BK) {
     // cap slices by the partial-buffer budget (fp32 [S, M, N] <= 256 MB)
+    // and at 64: past that the fixed-order reduce pass costs more than
+    // the occupancy buys (wgrad outputs are small, S*M*N reads dominate)
     const long budget = (256L << 20) / std::max<long>(M * N * 4, 1);
-    S = std::min<long>({(512 + tiles - 1) / tiles, ksteps, 512, budget});
+    S = std::min<long>({(512 + tiles - 1) / tiles, ksteps, 64, budget});
     S = std::max<long>(S, 1);
   }
   const long kslice = ((ksteps + S - 1) / S) * BK;
@@ -740,10 +779,19 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
   auto out = torch::empty({N}, X.options());
   const int chunks = (int)((M + kColsumRows - 1) / kColsumRows);
   auto part = torch::empty({chunks, N}, X.options().dtype(at::kFloat));
-  dim3 grid(ceil_div(N, 32), chunks);
-  hipLaunchKernelGGL(colsum_part_kernel, grid, dim3(256), 0, cur_stream(),
-                     (const bf16*)X.data_ptr(), M, N,
-                     part.data_ptr<float>());
+  if (N % 8 == 0) {
+    int G = 1;
+    while (G * 2 <= std::min<long>(N / 8, 256)) G *= 2;
+    hipLaunchKernelGGL(colsum_part_vec_kernel,
+                       dim3(ceil_div(N / 8, (long)G), chunks), dim3(256), 0,
+                       cur_stream(), (const bf16*)X.data_ptr(), M, N, G,
+                       part.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(colsum_part_kernel, dim3(ceil_div(N, 32), chunks),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)X.data_ptr(), M, N,
+                       part.data_ptr<float>());
+  }
   HIP_CHECK(hipGetLastError());
   hipLaunchKernelGGL(colsum_final_kernel, dim3(ceil_div(N, 32)), dim3(256),
                      0, cur_stream(), part.data_ptr<float>(), chunks, N,

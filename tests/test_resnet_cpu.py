@@ -59,8 +59,7 @@ class TestResNet20:
         assert logits.shape == (4, 10)
         loss = m.loss(x, y)
         loss.backward()
-        assert m.flat.grad is not None
-        assert float(m.flat.grad.abs().sum()) > 0
+        assert float(m.grad_flat().abs().sum()) > 0
         # param count sanity: resnet20 ~= 0.27M params
         assert 0.25e6 < m.numel < 0.30e6
 
@@ -94,7 +93,7 @@ class TestResNet50:
         y = torch.randint(0, 10, (2,))
         loss = m.loss(x, y)
         loss.backward()
-        assert float(m.flat.grad.abs().sum()) > 0
+        assert float(m.grad_flat().abs().sum()) > 0
         # ~25.5M params at 1000 classes; ~23.5M at 10
         assert 20e6 < m.numel < 27e6
 
