@@ -29,15 +29,15 @@ constexpr int kChunkRows = 1024;  // rows of [M, C] per partial
 // thread accumulating 8 channel partials from vector loads; fixed
 // pairwise halving tree over the row-lanes (structure-deterministic).
 __global__ void bn_stats_part_vec_kernel(const bf16* __restrict__ x, long M,
-                                         int C, int G,
+                                         int C, int G, long chunk_rows,
                                          float* __restrict__ psum,
                                          float* __restrict__ psq) {
   typedef __attribute__((ext_vector_type(8))) __bf16 v8;
   const int gi = threadIdx.x % G, rl = threadIdx.x / G;
   const int RL = blockDim.x / G;
   const int c8 = (blockIdx.x * G + gi) * 8;
-  const long r0 = (long)blockIdx.y * kChunkRows;
-  const long r1 = min(M, r0 + kChunkRows);
+  const long r0 = (long)blockIdx.y * chunk_rows;
+  const long r1 = min(M, r0 + chunk_rows);
   float sj[8] = {}, qj[8] = {};
   if (c8 < C)
     for (long m = r0 + rl; m < r1; m += RL) {
@@ -80,15 +80,15 @@ __global__ void bn_bwd_part_vec_kernel(const bf16* __restrict__ x,
                                        const bf16* __restrict__ dy,
                                        const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
-                                       long M, int C, int G,
+                                       long M, int C, int G, long chunk_rows,
                                        float* __restrict__ pdy,
                                        float* __restrict__ pdyx) {
   typedef __attribute__((ext_vector_type(8))) __bf16 v8;
   const int gi = threadIdx.x % G, rl = threadIdx.x / G;
   const int RL = blockDim.x / G;
   const int c8 = (blockIdx.x * G + gi) * 8;
-  const long r0 = (long)blockIdx.y * kChunkRows;
-  const long r1 = min(M, r0 + kChunkRows);
+  const long r0 = (long)blockIdx.y * chunk_rows;
+  const long r1 = min(M, r0 + chunk_rows);
   float s1[8] = {}, s2[8] = {};
   float mn[8], is[8];
   if (c8 < C) {
@@ -137,12 +137,13 @@ __global__ void bn_bwd_part_vec_kernel(const bf16* __restrict__ x,
 // pass 1 (scalar fallback, C % 8 != 0): per-(chunk, channel) partial
 // sum & sumsq. Block = 8 row-lanes x 32 channels.
 __global__ void bn_stats_part_kernel(const bf16* __restrict__ x, long M,
-                                     int C, float* __restrict__ psum,
+                                     int C, long chunk_rows,
+                                     float* __restrict__ psum,
                                      float* __restrict__ psq) {
   const int col = blockIdx.x * 32 + (threadIdx.x & 31);
   const int rlane = threadIdx.x >> 5;  // 0..7
-  const long r0 = (long)blockIdx.y * kChunkRows;
-  const long r1 = min(M, r0 + kChunkRows);
+  const long r0 = (long)blockIdx.y * chunk_rows;
+  const long r1 = min(M, r0 + chunk_rows);
   float s = 0.f, q = 0.f;
   if (col < C)
     for (long m = r0 + rlane; m < r1; m += 8) {
@@ -250,12 +251,13 @@ __global__ void bn_bwd_part_kernel(const bf16* __restrict__ x,
                                    const bf16* __restrict__ dy,
                                    const float* __restrict__ mean,
                                    const float* __restrict__ invstd, long M,
-                                   int C, float* __restrict__ pdy,
+                                   int C, long chunk_rows,
+                                   float* __restrict__ pdy,
                                    float* __restrict__ pdyx) {
   const int col = blockIdx.x * 32 + (threadIdx.x & 31);
   const int rlane = threadIdx.x >> 5;
-  const long r0 = (long)blockIdx.y * kChunkRows;
-  const long r1 = min(M, r0 + kChunkRows);
+  const long r0 = (long)blockIdx.y * chunk_rows;
+  const long r1 = min(M, r0 + chunk_rows);
   float s1 = 0.f, s2 = 0.f;
   if (col < C) {
     const float mn = mean[col], is = invstd[col];
@@ -429,6 +431,14 @@ inline int granule_lanes(int C) {
   return g;
 }
 
+// chunk rows so (col-blocks x chunks) >= ~768 blocks (wide-C layers
+// collapse the channel grid dimension, so the row dimension must
+// supply the parallelism)
+inline long pick_chunk_rows(long M, int cblocks) {
+  const long target = std::max<long>(1, 768 / std::max(cblocks, 1));
+  return std::max<long>(64, (M + target - 1) / target);
+}
+
 }  // namespace
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
@@ -437,22 +447,26 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
   CHECK_GPU(x); CHECK_CONTIG(x);
   const int C = (int)x.size(-1);
   const long M = x.numel() / C;
-  const int chunks = (int)((M + kChunkRows - 1) / kChunkRows);
+  const int G = (C % 8 == 0) ? granule_lanes(C) : 0;
+  const int cblocks = (C % 8 == 0) ? (int)ceil_div(C / 8, G)
+                                   : (int)ceil_div(C, 32);
+  const long rows = pick_chunk_rows(M, cblocks);
+  const int chunks = (int)((M + rows - 1) / rows);
   auto opts = x.options().dtype(at::kFloat);
   auto psum = torch::empty({chunks, C}, opts);
   auto psq = torch::empty({chunks, C}, opts);
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
   if (C % 8 == 0) {
-    const int G = granule_lanes(C);
-    hipLaunchKernelGGL(bn_stats_part_vec_kernel,
-                       dim3(ceil_div(C / 8, G), chunks), dim3(256), 0,
-                       cur_stream(), (const bf16*)x.data_ptr(), M, C, G,
+    hipLaunchKernelGGL(bn_stats_part_vec_kernel, dim3(cblocks, chunks),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), M, C, G, rows,
                        psum.data_ptr<float>(), psq.data_ptr<float>());
   } else {
-    hipLaunchKernelGGL(bn_stats_part_kernel, dim3(ceil_div(C, 32), chunks),
+    hipLaunchKernelGGL(bn_stats_part_kernel, dim3(cblocks, chunks),
                        dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
-                       M, C, psum.data_ptr<float>(), psq.data_ptr<float>());
+                       M, C, rows, psum.data_ptr<float>(),
+                       psq.data_ptr<float>());
   }
   hipLaunchKernelGGL(bn_stats_final_kernel, dim3(ceil_div(C, 32)), dim3(256),
                      0, cur_stream(), psum.data_ptr<float>(),
@@ -484,7 +498,11 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(dy);
   const int C = (int)x.size(-1);
   const long M = x.numel() / C;
-  const int chunks = (int)((M + kChunkRows - 1) / kChunkRows);
+  const int G = (C % 8 == 0) ? granule_lanes(C) : 0;
+  const int cblocks = (C % 8 == 0) ? (int)ceil_div(C / 8, G)
+                                   : (int)ceil_div(C, 32);
+  const long rows = pick_chunk_rows(M, cblocks);
+  const int chunks = (int)((M + rows - 1) / rows);
   auto opts = x.options().dtype(at::kFloat);
   auto pdy = torch::empty({chunks, C}, opts);
   auto pdyx = torch::empty({chunks, C}, opts);
@@ -494,18 +512,17 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
   auto dbeta = torch::empty({C}, x.options());
   auto gc = gamma.contiguous();
   if (C % 8 == 0) {
-    const int G = granule_lanes(C);
-    hipLaunchKernelGGL(bn_bwd_part_vec_kernel,
-                       dim3(ceil_div(C / 8, G), chunks), dim3(256), 0,
-                       cur_stream(), (const bf16*)x.data_ptr(),
+    hipLaunchKernelGGL(bn_bwd_part_vec_kernel, dim3(cblocks, chunks),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(),
                        (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), M, C, G,
+                       invstd.data_ptr<float>(), M, C, G, rows,
                        pdy.data_ptr<float>(), pdyx.data_ptr<float>());
   } else {
-    hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(ceil_div(C, 32), chunks),
+    hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(cblocks, chunks),
                        dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
                        (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), M, C,
+                       invstd.data_ptr<float>(), M, C, rows,
                        pdy.data_ptr<float>(), pdyx.data_ptr<float>());
   }
   hipLaunchKernelGGL(bn_bwd_final_kernel, dim3(ceil_div(C, 32)), dim3(256),

@@ -522,13 +522,13 @@ constexpr int kColsumRows = 1024;
 // granules per 256-thread block, fixed pairwise halving tree over the
 // row-lanes (structure-deterministic).
 __global__ void colsum_part_vec_kernel(const bf16* __restrict__ X, long M,
-                                       long N, int G,
+                                       long N, int G, long chunk_rows,
                                        float* __restrict__ part) {
   const int gi = threadIdx.x % G, rl = threadIdx.x / G;
   const int RL = (int)blockDim.x / G;
   const long c8 = ((long)blockIdx.x * G + gi) * 8;
-  const long r0 = (long)blockIdx.y * kColsumRows;
-  const long r1 = min(M, r0 + kColsumRows);
+  const long r0 = (long)blockIdx.y * chunk_rows;
+  const long r1 = min(M, r0 + chunk_rows);
   float sj[8] = {};
   if (c8 < N)
     for (long m = r0 + rl; m < r1; m += RL) {
@@ -556,12 +556,13 @@ __global__ void colsum_part_vec_kernel(const bf16* __restrict__ X, long M,
 }
 
 __global__ void colsum_part_kernel(const bf16* __restrict__ X, long M, long N,
+                                   long chunk_rows,
                                    float* __restrict__ part) {
   const long col = (long)blockIdx.x * 32 + (threadIdx.x & 31);
   const int rlane = threadIdx.x >> 5;  // 0..7
   const int chunk = blockIdx.y;
-  const long r0 = (long)chunk * kColsumRows;
-  const long r1 = min(M, r0 + kColsumRows);
+  const long r0 = (long)chunk * chunk_rows;
+  const long r1 = min(M, r0 + chunk_rows);
   float acc = 0.f;
   if (col < N)
     for (long m = r0 + rlane; m < r1; m += 8) acc += b2f(X[m * N + col]);
@@ -697,10 +698,12 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   long S = 1;
   if (tiles < 512 && K >= 4 * BK) {
     // cap slices by the partial-buffer budget (fp32 [S, M, N] <= 256 MB)
-    // and at 64: past that the fixed-order reduce pass costs more than
-    // the occupancy buys (wgrad outputs are small, S*M*N reads dominate)
+    // and by reduce traffic: allow large S only while the reduce pass
+    // stays under ~16 MB of partial reads (tiny-output wgrads want
+    // S~512 for occupancy; mid-size outputs drown in the reduce)
     const long budget = (256L << 20) / std::max<long>(M * N * 4, 1);
-    S = std::min<long>({(512 + tiles - 1) / tiles, ksteps, 64, budget});
+    const long rcap = std::max<long>(64, (16L << 20) / std::max<long>(M * N * 4, 1));
+    S = std::min<long>({(512 + tiles - 1) / tiles, ksteps, rcap, budget});
     S = std::max<long>(S, 1);
   }
   const long kslice = ((ksteps + S - 1) / S) * BK;
@@ -777,19 +780,23 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
   CHECK_GPU(X); CHECK_CONTIG(X);
   long M = X.size(0), N = X.size(1);
   auto out = torch::empty({N}, X.options());
-  const int chunks = (int)((M + kColsumRows - 1) / kColsumRows);
+  int G = 1;
+  while (G * 2 <= std::min<long>(N / 8, 256)) G *= 2;
+  const int cblocks = (N % 8 == 0) ? (int)ceil_div(N / 8, (long)G)
+                                   : (int)ceil_div(N, (long)32);
+  const long target = std::max<long>(1, 768 / std::max(cblocks, 1));
+  const long rows = std::max<long>(64, (M + target - 1) / target);
+  const int chunks = (int)((M + rows - 1) / rows);
   auto part = torch::empty({chunks, N}, X.options().dtype(at::kFloat));
   if (N % 8 == 0) {
-    int G = 1;
-    while (G * 2 <= std::min<long>(N / 8, 256)) G *= 2;
-    hipLaunchKernelGGL(colsum_part_vec_kernel,
-                       dim3(ceil_div(N / 8, (long)G), chunks), dim3(256), 0,
-                       cur_stream(), (const bf16*)X.data_ptr(), M, N, G,
+    hipLaunchKernelGGL(colsum_part_vec_kernel, dim3(cblocks, chunks),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)X.data_ptr(), M, N, G, rows,
                        part.data_ptr<float>());
   } else {
-    hipLaunchKernelGGL(colsum_part_kernel, dim3(ceil_div(N, 32), chunks),
+    hipLaunchKernelGGL(colsum_part_kernel, dim3(cblocks, chunks),
                        dim3(256), 0, cur_stream(),
-                       (const bf16*)X.data_ptr(), M, N,
+                       (const bf16*)X.data_ptr(), M, N, rows,
                        part.data_ptr<float>());
   }
   HIP_CHECK(hipGetLastError());
