@@ -167,7 +167,101 @@ __global__ void bn_stats_part_kernel(const bf16* __restrict__ x, long M,
   }
 }
 
-// pass 2: mean/invstd per channel (8-lane tree over chunks, fixed order)
+// Vectorized pass 2 (C % 4 == 0): float4 granules x KL chunk-lanes,
+// fixed pairwise tree (the serial per-channel chunk walk was
+// latency-bound at ~19us/call with ~768 chunks at wide C).
+__global__ void bn_stats_final_vec_kernel(const float* __restrict__ psum,
+                                          const float* __restrict__ psq,
+                                          int chunks, int C, int G,
+                                          float count, float eps,
+                                          float* __restrict__ mean,
+                                          float* __restrict__ invstd) {
+  const int gi = threadIdx.x % G, kl = threadIdx.x / G;
+  const int KL = blockDim.x / G;
+  const int c4 = (blockIdx.x * G + gi) * 4;
+  float sj[4] = {}, qj[4] = {};
+  if (c4 < C)
+    for (int k = kl; k < chunks; k += KL) {
+      const float4 a = *reinterpret_cast<const float4*>(&psum[(long)k * C + c4]);
+      const float4 b = *reinterpret_cast<const float4*>(&psq[(long)k * C + c4]);
+      sj[0] += a.x; sj[1] += a.y; sj[2] += a.z; sj[3] += a.w;
+      qj[0] += b.x; qj[1] += b.y; qj[2] += b.z; qj[3] += b.w;
+    }
+  __shared__ float rs[256][4], rq[256][4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    rs[threadIdx.x][j] = sj[j];
+    rq[threadIdx.x][j] = qj[j];
+  }
+  __syncthreads();
+  for (int h = KL >> 1; h > 0; h >>= 1) {
+    if (kl < h) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        rs[kl * G + gi][j] += rs[(kl + h) * G + gi][j];
+        rq[kl * G + gi][j] += rq[(kl + h) * G + gi][j];
+      }
+    }
+    __syncthreads();
+  }
+  if (kl == 0 && c4 < C) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float m = rs[gi][j] / count;
+      const float var = fmaxf(rq[gi][j] / count - m * m, 0.f);
+      mean[c4 + j] = m;
+      invstd[c4 + j] = rsqrtf(var + eps);
+    }
+  }
+}
+
+__global__ void bn_bwd_final_vec_kernel(const float* __restrict__ pdy,
+                                        const float* __restrict__ pdyx,
+                                        int chunks, int C, int G,
+                                        float* __restrict__ sdy,
+                                        float* __restrict__ sdyx,
+                                        bf16* __restrict__ dgamma,
+                                        bf16* __restrict__ dbeta) {
+  const int gi = threadIdx.x % G, kl = threadIdx.x / G;
+  const int KL = blockDim.x / G;
+  const int c4 = (blockIdx.x * G + gi) * 4;
+  float sj[4] = {}, qj[4] = {};
+  if (c4 < C)
+    for (int k = kl; k < chunks; k += KL) {
+      const float4 a = *reinterpret_cast<const float4*>(&pdy[(long)k * C + c4]);
+      const float4 b = *reinterpret_cast<const float4*>(&pdyx[(long)k * C + c4]);
+      sj[0] += a.x; sj[1] += a.y; sj[2] += a.z; sj[3] += a.w;
+      qj[0] += b.x; qj[1] += b.y; qj[2] += b.z; qj[3] += b.w;
+    }
+  __shared__ float rs[256][4], rq[256][4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    rs[threadIdx.x][j] = sj[j];
+    rq[threadIdx.x][j] = qj[j];
+  }
+  __syncthreads();
+  for (int h = KL >> 1; h > 0; h >>= 1) {
+    if (kl < h) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        rs[kl * G + gi][j] += rs[(kl + h) * G + gi][j];
+        rq[kl * G + gi][j] += rq[(kl + h) * G + gi][j];
+      }
+    }
+    __syncthreads();
+  }
+  if (kl == 0 && c4 < C) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      sdy[c4 + j] = rs[gi][j];
+      sdyx[c4 + j] = rq[gi][j];
+      dbeta[c4 + j] = f2b(rs[gi][j]);
+      dgamma[c4 + j] = f2b(rq[gi][j]);
+    }
+  }
+}
+
+// pass 2 (scalar fallback): mean/invstd per channel (8-lane tree)
 __global__ void bn_stats_final_kernel(const float* __restrict__ psum,
                                       const float* __restrict__ psq,
                                       int chunks, int C, float count,
@@ -393,6 +487,19 @@ __global__ void gap_bwd_kernel(const bf16* __restrict__ dy, long HW, int C,
                                long total, bf16* __restrict__ dx) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
+  if (C % 8 == 0) {
+    const long tg = total / 8;
+    const int c8g = C / 8;
+    for (long g = i; g < tg; g += stride) {
+      const int c8 = (int)(g % c8g) * 8;
+      const long n = g / ((long)c8g * HW);
+      bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(&dy[n * C + c8]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = f2b(b2f(v[j]) / (float)HW);
+      reinterpret_cast<bf16x8_t*>(dx)[g] = v;
+    }
+    return;
+  }
   for (; i < total; i += stride) {
     const int c = (int)(i % C);
     const long n = i / (HW * C);
@@ -400,14 +507,25 @@ __global__ void gap_bwd_kernel(const bf16* __restrict__ dy, long HW, int C,
   }
 }
 
-// fused residual add + relu: y = max(a+b, 0); bwd masks both branches
+// fused residual add + relu: y = max(a+b, 0); bwd masks both branches.
+// 16-B granule main loop + scalar tail.
 __global__ void add_relu_fwd_kernel(const bf16* __restrict__ a,
                                     const bf16* __restrict__ b,
                                     bf16* __restrict__ y, long n) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (; i < n; i += stride)
-    y[i] = f2b(fmaxf(b2f(a[i]) + b2f(b[i]), 0.f));
+  const long n8 = n / 8;
+  for (long k = i; k < n8; k += stride) {
+    const bf16x8_t av = reinterpret_cast<const bf16x8_t*>(a)[k];
+    const bf16x8_t bv = reinterpret_cast<const bf16x8_t*>(b)[k];
+    bf16x8_t out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      out[j] = f2b(fmaxf(b2f(av[j]) + b2f(bv[j]), 0.f));
+    reinterpret_cast<bf16x8_t*>(y)[k] = out;
+  }
+  for (long k = n8 * 8 + i; k < n; k += stride)
+    y[k] = f2b(fmaxf(b2f(a[k]) + b2f(b[k]), 0.f));
 }
 
 __global__ void add_relu_bwd_kernel(const bf16* __restrict__ y,
@@ -415,8 +533,17 @@ __global__ void add_relu_bwd_kernel(const bf16* __restrict__ y,
                                     bf16* __restrict__ da, long n) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (; i < n; i += stride)
-    da[i] = (b2f(y[i]) > 0.f) ? dy[i] : f2b(0.f);
+  const long n8 = n / 8;
+  for (long k = i; k < n8; k += stride) {
+    const bf16x8_t yv = reinterpret_cast<const bf16x8_t*>(y)[k];
+    bf16x8_t gv = reinterpret_cast<const bf16x8_t*>(dy)[k];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      if (!(b2f(yv[j]) > 0.f)) gv[j] = f2b(0.f);
+    reinterpret_cast<bf16x8_t*>(da)[k] = gv;
+  }
+  for (long k = n8 * 8 + i; k < n; k += stride)
+    da[k] = (b2f(y[k]) > 0.f) ? dy[k] : f2b(0.f);
 }
 
 inline int ew_grid(long n) {
@@ -468,10 +595,22 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
                        M, C, rows, psum.data_ptr<float>(),
                        psq.data_ptr<float>());
   }
-  hipLaunchKernelGGL(bn_stats_final_kernel, dim3(ceil_div(C, 32)), dim3(256),
-                     0, cur_stream(), psum.data_ptr<float>(),
-                     psq.data_ptr<float>(), chunks, C, (float)M, (float)eps,
-                     mean.data_ptr<float>(), invstd.data_ptr<float>());
+  if (C % 4 == 0) {
+    int G4 = 1;
+    while (G4 * 2 <= std::min(C / 4, 256)) G4 *= 2;
+    hipLaunchKernelGGL(bn_stats_final_vec_kernel,
+                       dim3(ceil_div(C / 4, G4)), dim3(256), 0,
+                       cur_stream(), psum.data_ptr<float>(),
+                       psq.data_ptr<float>(), chunks, C, G4, (float)M,
+                       (float)eps, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(bn_stats_final_kernel, dim3(ceil_div(C, 32)),
+                       dim3(256), 0, cur_stream(), psum.data_ptr<float>(),
+                       psq.data_ptr<float>(), chunks, C, (float)M,
+                       (float)eps, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>());
+  }
   auto y = torch::empty_like(x);
   auto gc = gamma.contiguous();
   auto bc = beta.contiguous();
@@ -525,11 +664,22 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
                        invstd.data_ptr<float>(), M, C, rows,
                        pdy.data_ptr<float>(), pdyx.data_ptr<float>());
   }
-  hipLaunchKernelGGL(bn_bwd_final_kernel, dim3(ceil_div(C, 32)), dim3(256),
-                     0, cur_stream(), pdy.data_ptr<float>(),
-                     pdyx.data_ptr<float>(), chunks, C,
-                     sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
-                     (bf16*)dgamma.data_ptr(), (bf16*)dbeta.data_ptr());
+  if (C % 4 == 0) {
+    int G4 = 1;
+    while (G4 * 2 <= std::min(C / 4, 256)) G4 *= 2;
+    hipLaunchKernelGGL(bn_bwd_final_vec_kernel,
+                       dim3(ceil_div(C / 4, G4)), dim3(256), 0,
+                       cur_stream(), pdy.data_ptr<float>(),
+                       pdyx.data_ptr<float>(), chunks, C, G4,
+                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
+                       (bf16*)dgamma.data_ptr(), (bf16*)dbeta.data_ptr());
+  } else {
+    hipLaunchKernelGGL(bn_bwd_final_kernel, dim3(ceil_div(C, 32)),
+                       dim3(256), 0, cur_stream(), pdy.data_ptr<float>(),
+                       pdyx.data_ptr<float>(), chunks, C,
+                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
+                       (bf16*)dgamma.data_ptr(), (bf16*)dbeta.data_ptr());
+  }
   auto dx = torch::empty_like(x);
   if (C % 8 == 0)
     hipLaunchKernelGGL(bn_bwd_dx_vec_kernel, dim3(ew_grid(x.numel() / 8)),

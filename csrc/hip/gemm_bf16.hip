@@ -577,8 +577,41 @@ __global__ void colsum_part_kernel(const bf16* __restrict__ X, long M, long N,
   }
 }
 
-// Same fixed-order 8-lane tree as splitk_reduce (chunks can reach
-// M/1024 = 512: a serial per-column loop was ~35us/call).
+// Vectorized final (N % 4 == 0): float4 granules x KL chunk-lanes,
+// fixed pairwise tree.
+__global__ void colsum_final_vec_kernel(const float* __restrict__ part,
+                                        int chunks, long N, int G,
+                                        bf16* __restrict__ out) {
+  const int gi = threadIdx.x % G, kl = threadIdx.x / G;
+  const int KL = (int)blockDim.x / G;
+  const long c4 = ((long)blockIdx.x * G + gi) * 4;
+  float sj[4] = {};
+  if (c4 < N)
+    for (int k = kl; k < chunks; k += KL) {
+      const float4 a =
+          *reinterpret_cast<const float4*>(&part[(long)k * N + c4]);
+      sj[0] += a.x; sj[1] += a.y; sj[2] += a.z; sj[3] += a.w;
+    }
+  __shared__ float rs[256][4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) rs[threadIdx.x][j] = sj[j];
+  __syncthreads();
+  for (int h = KL >> 1; h > 0; h >>= 1) {
+    if (kl < h) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        rs[kl * G + gi][j] += rs[(kl + h) * G + gi][j];
+    }
+    __syncthreads();
+  }
+  if (kl == 0 && c4 < N) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) out[c4 + j] = f2b(rs[gi][j]);
+  }
+}
+
+// Scalar final fallback: fixed-order 8-lane tree (chunks can reach
+// hundreds: a serial per-column loop was ~35us/call).
 __global__ void colsum_final_kernel(const float* __restrict__ part,
                                     int chunks, long N,
                                     bf16* __restrict__ out) {
@@ -597,6 +630,29 @@ __global__ void colsum_final_kernel(const float* __restrict__ part,
     for (int r = 0; r < 8; ++r) v += red[r][elane];
     out[col] = f2b(v);
   }
+}
+
+// Choose the split-K slice count minimizing estimated GEMM + reduce
+// time: gemm ~ 2MNK/tf scaled by chip fill (block_target blocks fill
+// the 256 CUs at this path's occupancy), reduce ~ (S+1)*M*N*4B at
+// ~6 TB/s. Replaces the fixed-target heuristics that either starved
+// tiny-output wgrads of occupancy or drowned mid-size outputs in
+// partial-buffer traffic.
+long pick_splitk(long M, long N, long K, long tiles, long ksteps,
+                 int block_target, double tf) {
+  const long budget = (256L << 20) / std::max<long>(M * N * 4, 1);
+  const long smax = std::min<long>({ksteps, budget, 512});
+  const double work = 2.0 * (double)M * N * K / tf;
+  double best_t = 1e30;
+  long best_s = 1;
+  for (long s = 1; s <= smax; s *= 2) {
+    const double fill =
+        std::min(1.0, (double)(tiles * s) / block_target);
+    const double t = work / std::max(fill, 1e-3) +
+                     (s > 1 ? (s + 1.0) * M * N * 4 / 6.0e12 : 0.0);
+    if (t < best_t) { best_t = t; best_s = s; }
+  }
+  return best_s;
 }
 
 struct TileCfg { int bm, bn, wr, wc; };
@@ -647,12 +703,7 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   if (M % 256 == 0 && N % 256 == 0 && K % BK2 == 0 && M >= 256 && N >= 256) {
     const long tiles = (M / 256) * (N / 256);
     const long ksteps64 = K / BK2;
-    long S = 1;
-    if (tiles < 256 && K >= 4 * BK2) {
-      const long budget = (256L << 20) / std::max<long>(M * N * 4, 1);
-      S = std::min<long>({(256 + tiles - 1) / tiles, ksteps64, 64, budget});
-      S = std::max<long>(S, 1);
-    }
+    long S = pick_splitk(M, N, K, tiles, ksteps64, 256, 800.0e12);
     const long kslice = ((ksteps64 + S - 1) / S) * BK2;
     S = (K + kslice - 1) / kslice;
 
@@ -695,17 +746,7 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
 
   // split-K when the tile grid cannot fill the chip and K is deep
   const long ksteps = (K + BK - 1) / BK;
-  long S = 1;
-  if (tiles < 512 && K >= 4 * BK) {
-    // cap slices by the partial-buffer budget (fp32 [S, M, N] <= 256 MB)
-    // and by reduce traffic: allow large S only while the reduce pass
-    // stays under ~16 MB of partial reads (tiny-output wgrads want
-    // S~512 for occupancy; mid-size outputs drown in the reduce)
-    const long budget = (256L << 20) / std::max<long>(M * N * 4, 1);
-    const long rcap = std::max<long>(64, (16L << 20) / std::max<long>(M * N * 4, 1));
-    S = std::min<long>({(512 + tiles - 1) / tiles, ksteps, rcap, budget});
-    S = std::max<long>(S, 1);
-  }
+  long S = pick_splitk(M, N, K, tiles, ksteps, 512, 150.0e12);
   const long kslice = ((ksteps + S - 1) / S) * BK;
   S = (K + kslice - 1) / kslice;  // actual slices after rounding
 
@@ -800,9 +841,18 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
                        part.data_ptr<float>());
   }
   HIP_CHECK(hipGetLastError());
-  hipLaunchKernelGGL(colsum_final_kernel, dim3(ceil_div(N, 32)), dim3(256),
-                     0, cur_stream(), part.data_ptr<float>(), chunks, N,
-                     (bf16*)out.data_ptr());
+  if (N % 4 == 0) {
+    int G4 = 1;
+    while (G4 * 2 <= std::min<long>(N / 4, 256)) G4 *= 2;
+    hipLaunchKernelGGL(colsum_final_vec_kernel,
+                       dim3(ceil_div(N / 4, (long)G4)), dim3(256), 0,
+                       cur_stream(), part.data_ptr<float>(), chunks, N, G4,
+                       (bf16*)out.data_ptr());
+  } else {
+    hipLaunchKernelGGL(colsum_final_kernel, dim3(ceil_div(N, 32)),
+                       dim3(256), 0, cur_stream(), part.data_ptr<float>(),
+                       chunks, N, (bf16*)out.data_ptr());
+  }
   HIP_CHECK(hipGetLastError());
   return out;
 }
