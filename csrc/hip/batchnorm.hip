@@ -562,7 +562,10 @@ inline int granule_lanes(int C) {
 // collapse the channel grid dimension, so the row dimension must
 // supply the parallelism)
 inline long pick_chunk_rows(long M, int cblocks) {
-  const long target = std::max<long>(1, 768 / std::max(cblocks, 1));
+  // <= 256 chunks keeps the final pass small; >= 768 total blocks
+  // fills the chip when the channel dimension alone cannot
+  const long target =
+      std::max<long>(1, std::min<long>(768 / std::max(cblocks, 1), 256));
   return std::max<long>(64, (M + target - 1) / target);
 }
 
@@ -596,8 +599,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
                        psq.data_ptr<float>());
   }
   if (C % 4 == 0) {
-    int G4 = 1;
-    while (G4 * 2 <= std::min(C / 4, 256)) G4 *= 2;
+    int G4 = 1;  // <= 16 so every block keeps >= 16 chunk-lanes
+    while (G4 * 2 <= std::min(C / 4, 16)) G4 *= 2;
     hipLaunchKernelGGL(bn_stats_final_vec_kernel,
                        dim3(ceil_div(C / 4, G4)), dim3(256), 0,
                        cur_stream(), psum.data_ptr<float>(),
@@ -665,8 +668,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
                        pdy.data_ptr<float>(), pdyx.data_ptr<float>());
   }
   if (C % 4 == 0) {
-    int G4 = 1;
-    while (G4 * 2 <= std::min(C / 4, 256)) G4 *= 2;
+    int G4 = 1;  // <= 16 so every block keeps >= 16 chunk-lanes
+    while (G4 * 2 <= std::min(C / 4, 16)) G4 *= 2;
     hipLaunchKernelGGL(bn_bwd_final_vec_kernel,
                        dim3(ceil_div(C / 4, G4)), dim3(256), 0,
                        cur_stream(), pdy.data_ptr<float>(),

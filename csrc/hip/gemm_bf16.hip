@@ -825,7 +825,8 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
   while (G * 2 <= std::min<long>(N / 8, 256)) G *= 2;
   const int cblocks = (N % 8 == 0) ? (int)ceil_div(N / 8, (long)G)
                                    : (int)ceil_div(N, (long)32);
-  const long target = std::max<long>(1, 768 / std::max(cblocks, 1));
+  const long target =
+      std::max<long>(1, std::min<long>(768 / std::max(cblocks, 1), 256));
   const long rows = std::max<long>(64, (M + target - 1) / target);
   const int chunks = (int)((M + rows - 1) / rows);
   auto part = torch::empty({chunks, N}, X.options().dtype(at::kFloat));
@@ -842,8 +843,8 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
   }
   HIP_CHECK(hipGetLastError());
   if (N % 4 == 0) {
-    int G4 = 1;
-    while (G4 * 2 <= std::min<long>(N / 4, 256)) G4 *= 2;
+    int G4 = 1;  // <= 16 so every block keeps >= 16 chunk-lanes
+    while (G4 * 2 <= std::min<long>(N / 4, 16)) G4 *= 2;
     hipLaunchKernelGGL(colsum_final_vec_kernel,
                        dim3(ceil_div(N / 4, (long)G4)), dim3(256), 0,
                        cur_stream(), part.data_ptr<float>(), chunks, N, G4,
