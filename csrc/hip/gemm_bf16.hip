@@ -255,7 +255,10 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
 constexpr int BK2 = 64;
 constexpr int BKP2 = BK2 + 8;
 
-template <bool TA, bool TB>
+// Both operands in the vector-staging layout (A [M][K], B [N][K]; the
+// host pre-transposes anything else). M is bounds-guarded, so only
+// N % BN == 0 and K % 64 == 0 gate this path.
+template <int BM, int BN>
 __launch_bounds__(512, 1)
 __global__ void gemm256_kernel(const bf16* __restrict__ A,
                                const bf16* __restrict__ B,
@@ -263,13 +266,13 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
                                const bf16* __restrict__ bias, long M, long N,
                                long K, long kslice, int relu, int store_mode,
                                long ohw) {
-  constexpr int BM = 256, BN = 256;
-  constexpr int FM = 8, FN = 4;  // per-wave 128x64 output
+  constexpr int FM = BM / 32, FN = BN / 64;  // per-wave (BM/2)x(BN/4)
+  constexpr int GA = BM / 64, GB = BN / 64;  // 16-B granules per thread
   __shared__ __align__(16) bf16 As[2][BM][BKP2];
   __shared__ __align__(16) bf16 Bs[2][BN][BKP2];
 
   const int ntn = (int)(N / BN);
-  const int ntm = (int)(M / BM);
+  const int ntm = (int)((M + BM - 1) / BM);
   const int nwg = ntm * ntn;
   int bid = blockIdx.x;
   {  // XCD-bijective remap (guide T1)
@@ -291,59 +294,37 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
   const int wn0 = (wave & 3) * (BN / 4);
   const int l15 = lane & 15, l4 = lane >> 4;
 
-  // staging registers: 4 granules of 16 B per operand per thread
-  bf16x8_t ra[4], rb[4];
-  // granule coords (constant per thread across K-steps)
-  //   contiguous-k layout (!T): granule g -> row g/8, k8 (g%8)*8
-  //   transposed layout   (T) : granule g -> k g/32, row8 (g%32)*8
-  int am_[4], ak_[4], bm_[4], bk_[4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int g = tid + i * 512;
-    if (!TA) { am_[i] = g >> 3;          ak_[i] = (g & 7) * 8; }
-    else     { am_[i] = (g & 31) * 8;    ak_[i] = g >> 5; }
-    if (!TB) { bm_[i] = (g & 31) * 8;    bk_[i] = g >> 5; }
-    else     { bm_[i] = g >> 3;          bk_[i] = (g & 7) * 8; }
-  }
-
+  // staging registers; granule g -> tile row g/8, k8 (g%8)*8
+  bf16x8_t ra[GA], rb[GB];
   auto load_tiles = [&](long k0) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      if (!TA)
+    for (int i = 0; i < GA; ++i) {
+      const int g = tid + i * 512;
+      const long gm = tile_m + (g >> 3);
+      if (gm < M)
         ra[i] = *reinterpret_cast<const bf16x8_t*>(
-            &A[(tile_m + am_[i]) * K + k0 + ak_[i]]);
+            &A[gm * K + k0 + (g & 7) * 8]);
       else
-        ra[i] = *reinterpret_cast<const bf16x8_t*>(
-            &A[(k0 + ak_[i]) * M + tile_m + am_[i]]);
-      if (!TB)
-        rb[i] = *reinterpret_cast<const bf16x8_t*>(
-            &B[(k0 + bk_[i]) * N + tile_n + bm_[i]]);
-      else
-        rb[i] = *reinterpret_cast<const bf16x8_t*>(
-            &B[(tile_n + bm_[i]) * K + k0 + bk_[i]]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) ra[i][j] = 0.f;
+    }
+#pragma unroll
+    for (int i = 0; i < GB; ++i) {
+      const int g = tid + i * 512;
+      rb[i] = *reinterpret_cast<const bf16x8_t*>(
+          &B[(tile_n + (g >> 3)) * K + k0 + (g & 7) * 8]);
     }
   };
   auto write_tiles = [&](int buf) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      if (!TA) {
-        *reinterpret_cast<bf16x8_t*>(&As[buf][am_[i]][ak_[i]]) = ra[i];
-      } else {
-        const u16x8_t v = *reinterpret_cast<const u16x8_t*>(&ra[i]);
+    for (int i = 0; i < GA; ++i) {
+      const int g = tid + i * 512;
+      *reinterpret_cast<bf16x8_t*>(&As[buf][g >> 3][(g & 7) * 8]) = ra[i];
+    }
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<unsigned short*>(&As[buf][am_[i] + j][ak_[i]]) =
-              v[j];
-      }
-      if (!TB) {
-        const u16x8_t v = *reinterpret_cast<const u16x8_t*>(&rb[i]);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<unsigned short*>(&Bs[buf][bm_[i] + j][bk_[i]]) =
-              v[j];
-      } else {
-        *reinterpret_cast<bf16x8_t*>(&Bs[buf][bm_[i]][bk_[i]]) = rb[i];
-      }
+    for (int i = 0; i < GB; ++i) {
+      const int g = tid + i * 512;
+      *reinterpret_cast<bf16x8_t*>(&Bs[buf][g >> 3][(g & 7) * 8]) = rb[i];
     }
   };
 
@@ -392,7 +373,7 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const long row = tile_m + wm0 + fm * 16 + l4 * 4 + r;
-          out[row * N + col] = acc[fm][fn][r];
+          if (row < M) out[row * N + col] = acc[fm][fn][r];
         }
       }
     return;
@@ -407,6 +388,7 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const long row = tile_m + wm0 + fm * 16 + l4 * 4 + r;
+        if (row >= M) continue;
         float v = acc[fm][fn][r] + bv;
         if (relu) v = fmaxf(v, 0.f);
         if (store_mode == (int)EpStore::kConvNCHW) {
@@ -639,7 +621,7 @@ __global__ void colsum_final_kernel(const float* __restrict__ part,
 // tiny-output wgrads of occupancy or drowned mid-size outputs in
 // partial-buffer traffic.
 long pick_splitk(long M, long N, long K, long tiles, long ksteps,
-                 int block_target, double tf) {
+                 int block_target, double tf, double* time_out = nullptr) {
   const long budget = (256L << 20) / std::max<long>(M * N * 4, 1);
   const long smax = std::min<long>({ksteps, budget, 512});
   const double work = 2.0 * (double)M * N * K / tf;
@@ -652,6 +634,7 @@ long pick_splitk(long M, long N, long K, long tiles, long ksteps,
                      (s > 1 ? (s + 1.0) * M * N * 4 / 6.0e12 : 0.0);
     if (t < best_t) { best_t = t; best_s = s; }
   }
+  if (time_out) *time_out = best_t;
   return best_s;
 }
 
@@ -699,11 +682,28 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   bf16* c = (bf16*)C.data_ptr();
   const bf16* bs = bias ? (const bf16*)bias->data_ptr() : nullptr;
 
-  // ---- 256x256 double-buffered path for large aligned shapes ----
-  if (M % 256 == 0 && N % 256 == 0 && K % BK2 == 0 && M >= 256 && N >= 256) {
-    const long tiles = (M / 256) * (N / 256);
+  // ---- double-buffered BMxBN path (M bounds-guarded; needs N % 64
+  // and K % 64). Tile + split-K chosen together by the cost model;
+  // per-tile efficiency from the measured ladder (64-tile structures
+  // run far below the 256 ones).
+  if (N % 64 == 0 && K % BK2 == 0 && M >= 48 && N >= 64) {
     const long ksteps64 = K / BK2;
-    long S = pick_splitk(M, N, K, tiles, ksteps64, 256, 800.0e12);
+    const int bn2 = (N % 256 == 0) ? 256 : (N % 128 == 0 ? 128 : 64);
+    const long ntn2 = N / bn2;
+    int bm2 = 256;
+    long S = 1;
+    {
+      double best_t = 1e30;
+      for (int bm : {256, 128, 64}) {
+        const double tf = bm == 256 ? 800.0e12
+                                    : (bm == 128 ? 600.0e12 : 320.0e12);
+        const long tiles_c = ((M + bm - 1) / bm) * ntn2;
+        double t;
+        const long s = pick_splitk(M, N, K, tiles_c, ksteps64, 256, tf, &t);
+        if (t < best_t) { best_t = t; bm2 = bm; S = s; }
+      }
+    }
+    const long tiles = ((M + bm2 - 1) / bm2) * ntn2;
     const long kslice = ((ksteps64 + S - 1) / S) * BK2;
     S = (K + kslice - 1) / kslice;
 
@@ -726,9 +726,26 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
     dim3 grid((unsigned)tiles, (unsigned)S);
     dim3 block(512);
     const int sm = (int)store;
-    hipLaunchKernelGGL((gemm256_kernel<false, true>), grid, block, 0,
-                       cur_stream(), a2, b2, c, part_ptr, bs, M, N, K, kslice,
-                       relu, sm, ohw);
+    auto launch2 = [&](auto bmv, auto bnv) {
+      hipLaunchKernelGGL(
+          (gemm256_kernel<decltype(bmv)::value, decltype(bnv)::value>),
+          grid, block, 0, cur_stream(), a2, b2, c, part_ptr, bs, M, N, K,
+          kslice, relu, sm, ohw);
+    };
+    using c64i = std::integral_constant<int, 64>;
+    using c128i = std::integral_constant<int, 128>;
+    using c256i = std::integral_constant<int, 256>;
+    switch (bm2 * 1000 + bn2) {
+      case 256256: launch2(c256i{}, c256i{}); break;
+      case 256128: launch2(c256i{}, c128i{}); break;
+      case 256064: launch2(c256i{}, c64i{}); break;
+      case 128256: launch2(c128i{}, c256i{}); break;
+      case 128128: launch2(c128i{}, c128i{}); break;
+      case 128064: launch2(c128i{}, c64i{}); break;
+      case  64256: launch2(c64i{}, c256i{}); break;
+      case  64128: launch2(c64i{}, c128i{}); break;
+      case  64064: launch2(c64i{}, c64i{}); break;
+    }
     HIP_CHECK(hipGetLastError());
     if (S > 1) {
       const long total = M * N;
