@@ -630,8 +630,12 @@ long pick_splitk(long M, long N, long K, long tiles, long ksteps,
   for (long s = 1; s <= smax; s *= 2) {
     const double fill =
         std::min(1.0, (double)(tiles * s) / block_target);
-    const double t = work / std::max(fill, 1e-3) +
-                     (s > 1 ? (s + 1.0) * M * N * 4 / 6.0e12 : 0.0);
+    // short per-slice K chains never amortize the staging prologue;
+    // the measured reduce pass runs ~2.5 TB/s, not peak HBM
+    const double steps = (double)ksteps / s;
+    const double eff = steps / (steps + 4.0);
+    const double t = work / std::max(fill * eff, 1e-3) +
+                     (s > 1 ? (s + 1.0) * M * N * 4 / 2.5e12 : 0.0);
     if (t < best_t) { best_t = t; best_s = s; }
   }
   if (time_out) *time_out = best_t;
