@@ -256,8 +256,8 @@ constexpr int BK2 = 64;
 constexpr int BKP2 = BK2 + 8;
 
 // Both operands in the vector-staging layout (A [M][K], B [N][K]; the
-// host pre-transposes anything else). M is bounds-guarded, so only
-// N % BN == 0 and K % 64 == 0 gate this path.
+// host pre-transposes anything else). M and K edges are bounds-guarded,
+// so only N % BN == 0 and K % 8 == 0 gate this path.
 template <int BM, int BN>
 __launch_bounds__(512, 1)
 __global__ void gemm256_kernel(const bf16* __restrict__ A,
@@ -266,8 +266,12 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
                                const bf16* __restrict__ bias, long M, long N,
                                long K, long kslice, int relu, int store_mode,
                                long ohw) {
-  constexpr int FM = BM / 32, FN = BN / 64;  // per-wave (BM/2)x(BN/4)
-  constexpr int GA = BM / 64, GB = BN / 64;  // 16-B granules per thread
+  // wave grid: 2(M) x 4(N) for BN >= 64; 4(M) x 2(N) for BN == 32
+  constexpr int WR = BN >= 64 ? 2 : 4, WC = BN >= 64 ? 4 : 2;
+  constexpr int FM = BM / WR / 16, FN = BN / WC / 16;
+  constexpr int GA = BM / 64;                    // A granules per thread
+  constexpr int GBT = BN * 8;                    // total B granules
+  constexpr int GB = GBT >= 512 ? GBT / 512 : 1;
   __shared__ __align__(16) bf16 As[2][BM][BKP2];
   __shared__ __align__(16) bf16 Bs[2][BN][BKP2];
 
@@ -285,34 +289,56 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
   const long tile_n = (long)(bid % ntn) * BN;
   const long k_begin = (long)blockIdx.y * kslice;
   const long k_end = min(K, k_begin + kslice);
-  const int nt = (int)((k_end - k_begin) / BK2);
+  const int nt = (int)((k_end - k_begin + BK2 - 1) / BK2);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;             // 8 waves: 2(M) x 4(N)
-  const int wm0 = (wave >> 2) * (BM / 2);
-  const int wn0 = (wave & 3) * (BN / 4);
+  const int wave = tid >> 6;             // 8 waves: WR(M) x WC(N)
+  const int wm0 = (wave / WC) * (BM / WR);
+  const int wn0 = (wave % WC) * (BN / WC);
   const int l15 = lane & 15, l4 = lane >> 4;
 
-  // staging registers; granule g -> tile row g/8, k8 (g%8)*8
+  // staging registers; granule g -> tile row g/8, k8 (g%8)*8.
+  // M rows beyond the matrix and K columns beyond k_end stage zeros
+  // (zeros are MFMA-neutral), so edge tiles need no special kernel.
   bf16x8_t ra[GA], rb[GB];
   auto load_tiles = [&](long k0) {
 #pragma unroll
     for (int i = 0; i < GA; ++i) {
       const int g = tid + i * 512;
       const long gm = tile_m + (g >> 3);
-      if (gm < M)
-        ra[i] = *reinterpret_cast<const bf16x8_t*>(
-            &A[gm * K + k0 + (g & 7) * 8]);
-      else
+      const long gk = k0 + (g & 7) * 8;
+      if (gm < M && gk + 8 <= k_end) {
+        ra[i] = *reinterpret_cast<const bf16x8_t*>(&A[gm * K + gk]);
+      } else {
+        u16x8_t v;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) ra[i][j] = 0.f;
+        for (int j = 0; j < 8; ++j)
+          v[j] = (gm < M && gk + j < k_end)
+                     ? *reinterpret_cast<const unsigned short*>(
+                           &A[gm * K + gk + j])
+                     : (unsigned short)0;
+        ra[i] = *reinterpret_cast<const bf16x8_t*>(&v);
+      }
     }
 #pragma unroll
     for (int i = 0; i < GB; ++i) {
       const int g = tid + i * 512;
-      rb[i] = *reinterpret_cast<const bf16x8_t*>(
-          &B[(tile_n + (g >> 3)) * K + k0 + (g & 7) * 8]);
+      if (GBT < 512 && g >= GBT) break;
+      const long gn = tile_n + (g >> 3);
+      const long gk = k0 + (g & 7) * 8;
+      if (gk + 8 <= k_end) {
+        rb[i] = *reinterpret_cast<const bf16x8_t*>(&B[gn * K + gk]);
+      } else {
+        u16x8_t v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = (gk + j < k_end)
+                     ? *reinterpret_cast<const unsigned short*>(
+                           &B[gn * K + gk + j])
+                     : (unsigned short)0;
+        rb[i] = *reinterpret_cast<const bf16x8_t*>(&v);
+      }
     }
   };
   auto write_tiles = [&](int buf) {
@@ -324,6 +350,7 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
 #pragma unroll
     for (int i = 0; i < GB; ++i) {
       const int g = tid + i * 512;
+      if (GBT < 512 && g >= GBT) break;
       *reinterpret_cast<bf16x8_t*>(&Bs[buf][g >> 3][(g & 7) * 8]) = rb[i];
     }
   };
@@ -690,17 +717,22 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   // and K % 64). Tile + split-K chosen together by the cost model;
   // per-tile efficiency from the measured ladder (64-tile structures
   // run far below the 256 ones).
-  if (N % 64 == 0 && K % BK2 == 0 && M >= 48 && N >= 64) {
-    const long ksteps64 = K / BK2;
-    const int bn2 = (N % 256 == 0) ? 256 : (N % 128 == 0 ? 128 : 64);
+  if (N % 32 == 0 && K % 8 == 0 && M >= 48 && N >= 32 && K >= 32) {
+    const long ksteps64 = (K + BK2 - 1) / BK2;
+    const int bn2 = (N % 256 == 0) ? 256
+                    : (N % 128 == 0 ? 128 : (N % 64 == 0 ? 64 : 32));
     const long ntn2 = N / bn2;
     int bm2 = 256;
     long S = 1;
     {
       double best_t = 1e30;
+      // rough per-tile efficiency from the measured ladder; narrow
+      // tiles are staging-heavier
+      const double bn_pen = bn2 >= 128 ? 1.0 : (bn2 == 64 ? 0.75 : 0.5);
       for (int bm : {256, 128, 64}) {
-        const double tf = bm == 256 ? 800.0e12
-                                    : (bm == 128 ? 600.0e12 : 320.0e12);
+        const double tf = (bm == 256 ? 800.0e12
+                                     : (bm == 128 ? 600.0e12 : 320.0e12)) *
+                          bn_pen;
         const long tiles_c = ((M + bm - 1) / bm) * ntn2;
         double t;
         const long s = pick_splitk(M, N, K, tiles_c, ksteps64, 256, tf, &t);
@@ -739,16 +771,20 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
     using c64i = std::integral_constant<int, 64>;
     using c128i = std::integral_constant<int, 128>;
     using c256i = std::integral_constant<int, 256>;
+    using c32i = std::integral_constant<int, 32>;
     switch (bm2 * 1000 + bn2) {
       case 256256: launch2(c256i{}, c256i{}); break;
       case 256128: launch2(c256i{}, c128i{}); break;
       case 256064: launch2(c256i{}, c64i{}); break;
+      case 256032: launch2(c256i{}, c32i{}); break;
       case 128256: launch2(c128i{}, c256i{}); break;
       case 128128: launch2(c128i{}, c128i{}); break;
       case 128064: launch2(c128i{}, c64i{}); break;
+      case 128032: launch2(c128i{}, c32i{}); break;
       case  64256: launch2(c64i{}, c256i{}); break;
       case  64128: launch2(c64i{}, c128i{}); break;
       case  64064: launch2(c64i{}, c64i{}); break;
+      case  64032: launch2(c64i{}, c32i{}); break;
     }
     HIP_CHECK(hipGetLastError());
     if (S > 1) {
