@@ -717,10 +717,12 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   // and K % 64). Tile + split-K chosen together by the cost model;
   // per-tile efficiency from the measured ladder (64-tile structures
   // run far below the 256 ones).
-  if (N % 32 == 0 && K % 8 == 0 && M >= 48 && N >= 32 && K >= 32) {
+  // (a BN=32 config measured slower than the synchronous 128x32 path
+  // on the shallow-K shapes it would serve; N%64 stays the floor)
+  if (N % 64 == 0 && K % 8 == 0 && M >= 48 && N >= 64 && K >= 32) {
     const long ksteps64 = (K + BK2 - 1) / BK2;
     const int bn2 = (N % 256 == 0) ? 256
-                    : (N % 128 == 0 ? 128 : (N % 64 == 0 ? 64 : 32));
+                    : (N % 128 == 0 ? 128 : 64);
     const long ntn2 = N / bn2;
     int bm2 = 256;
     long S = 1;
