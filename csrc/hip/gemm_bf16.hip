@@ -253,7 +253,12 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
 // ---------------------------------------------------------------------------
 
 constexpr int BK2 = 64;
-constexpr int BKP2 = BK2 + 8;
+// +16-element pad: ds_read_b128 lane groups mix both 32-lane halves
+// ((l15, l4) pairs), and a row walk of 10 mod 16 16-B slots puts all 16
+// lanes of a group on distinct slots (72 was 2-way conflicted: 37.5%
+// of LDS cycles were conflict stalls in rocprof). 2*(256+256)*80*2 B =
+// exactly the 160 KiB LDS.
+constexpr int BKP2 = BK2 + 16;
 
 // Both operands in the vector-staging layout (A [M][K], B [N][K]; the
 // host pre-transposes anything else). M and K edges are bounds-guarded,
