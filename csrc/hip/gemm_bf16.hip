@@ -434,6 +434,176 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
   }
 }
 
+// ---------------------------------------------------------------------------
+// 8-phase 256x256 GEMM (cdna_hip_programming.md §5 "8-phase template"):
+// global_load_lds direct staging (no staging registers, no ds_writes),
+// counted vmcnt so prefetches stay in flight ACROSS raw barriers, two
+// barriers per phase, setprio(1) around each MFMA cluster. Per K-tile
+// (BK=64): 4 phases, each issuing one half-tile prefetch (2 glds) for
+// tile t+1 and computing one (kk, fn-pair) quadrant = 16 MFMAs; one
+// s_waitcnt vmcnt(2) per K-tile (tile t fully landed, the just-issued
+// half of t+1 still in flight). LDS image is lane-linear [256][64]
+// per operand (glds requirement); the bank swizzle therefore moves to
+// the per-lane SOURCE address plus the same XOR on every ds_read
+// (guide rule 21): byte ^= bit8->bit5 ^ bit10->bit6, which puts all 16
+// lanes of a ds_read_b128 group on 16 distinct 16-B slots (verified by
+// enumeration; the 37.5% SQ_LDS_BANK_CONFLICT of the padded layout
+// came from groups mixing both 32-lane halves). Gated to M%256, N%256,
+// K%64 (glds cannot zero-fill, so no edge guards here).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int swz8p(int byte) {
+  return byte ^ (((byte >> 8) & 1) << 5) ^ (((byte >> 10) & 1) << 6);
+}
+
+__launch_bounds__(512, 1)
+__global__ void gemm8p_kernel(const bf16* __restrict__ A,
+                              const bf16* __restrict__ B,
+                              bf16* __restrict__ C, float* __restrict__ Cpart,
+                              const bf16* __restrict__ bias, long M, long N,
+                              long K, long kslice, int relu) {
+  constexpr int FM = 8, FN = 4;
+  // ONE shared object (a second one makes hipcc drain vmcnt before
+  // every ds_read of a glds pipeline — guide §5 trap 4a):
+  // [dbuf][op][row][k] bf16, 128 KiB.
+  __shared__ __align__(16) bf16 lds[2][2][256][64];
+
+  const int ntn = (int)(N / 256);
+  const int nwg = (int)((M / 256) * ntn);
+  int bid = blockIdx.x;
+  {
+    const int nxcd = 8;
+    const int q = nwg / nxcd, r = nwg % nxcd;
+    const int xcd = bid % nxcd, idx = bid / nxcd;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const long tile_m = (long)(bid / ntn) * 256;
+  const long tile_n = (long)(bid % ntn) * 256;
+  const long k_begin = (long)blockIdx.y * kslice;
+  const long k_end = min(K, k_begin + kslice);
+  const int nt = (int)((k_end - k_begin) / 64);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm0 = (wave >> 2) * 128;
+  const int wn0 = (wave & 3) * 64;
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  // per-thread glds source coordinates (2 passes of 16 B per half-tile;
+  // the linear LDS offset o maps to the swizzled logical element)
+  int srow[2], scol[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int o = (tid + i * 512) * 16;
+    srow[i] = o >> 7;                       // 0..127 within the half
+    scol[i] = (swz8p(o) & 127) >> 1;        // element column 0..63
+  }
+  const char* lds_base = (const char*)&lds[0][0][0][0];
+
+  // h: 0 = A rows 0-127, 1 = B rows 0-127, 2 = A rows 128-255,
+  //    3 = B rows 128-255
+  auto issue_half = [&](int buf, long kt, int h) {
+    const int op = h & 1, half = h >> 1;
+    const bf16* src = op == 0 ? A : B;
+    const long brow = (op == 0 ? tile_m : tile_n) + half * 128;
+    const long k0 = k_begin + kt * 64;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const bf16* gp = &src[(brow + srow[i]) * K + k0 + scol[i]];
+      char* lp = (char*)&lds[buf][op][half * 128][0] + wave * 1024 +
+                 i * 8192;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gp,
+          (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
+    }
+  };
+
+  f32x4_t acc[FM][FN] = {};
+  bf16x8_t a8[FM];
+
+  // prologue: stage all 4 halves of tile 0
+#pragma unroll
+  for (int h = 0; h < 4; ++h) issue_half(0, 0, h);
+
+  for (int t = 0; t < nt; ++t) {
+    const int buf = t & 1;
+    const char* abase = (const char*)&lds[buf][0][0][0];
+    const char* bbase = (const char*)&lds[buf][1][0][0];
+    auto do_phase = [&](auto qc) {
+      constexpr int q = decltype(qc)::value;
+      constexpr int kk = q >> 1, p = q & 1;
+      if (t + 1 < nt) issue_half(buf ^ 1, t + 1, q);
+      if (q == 0) {
+        if (t + 1 < nt)
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      if (p == 0) {  // (re)load the kk A-fragments
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm) {
+          const int byte = (wm0 + fm * 16 + l15) * 128 + kk * 64 + l4 * 16;
+          a8[fm] = *reinterpret_cast<const bf16x8_t*>(abase + swz8p(byte));
+        }
+      }
+      bf16x8_t b2[2];
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int byte =
+            (wn0 + (2 * p + j) * 16 + l15) * 128 + kk * 64 + l4 * 16;
+        b2[j] = *reinterpret_cast<const bf16x8_t*>(bbase + swz8p(byte));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[fm][2 * p + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a8[fm], b2[j], acc[fm][2 * p + j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    };
+    do_phase(std::integral_constant<int, 0>{});
+    do_phase(std::integral_constant<int, 1>{});
+    do_phase(std::integral_constant<int, 2>{});
+    do_phase(std::integral_constant<int, 3>{});
+  }
+  (void)lds_base;
+
+  if (Cpart) {
+    float* out = Cpart + (long)blockIdx.y * M * N;
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn) {
+        const long col = tile_n + wn0 + fn * 16 + l15;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const long row = tile_m + wm0 + fm * 16 + l4 * 4 + r;
+          out[row * N + col] = acc[fm][fn][r];
+        }
+      }
+    return;
+  }
+#pragma unroll
+  for (int fm = 0; fm < FM; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      const long col = tile_n + wn0 + fn * 16 + l15;
+      const float bv = bias ? b2f(bias[col]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = tile_m + wm0 + fm * 16 + l4 * 4 + r;
+        float v = acc[fm][fn][r] + bv;
+        if (relu) v = fmaxf(v, 0.f);
+        C[row * N + col] = f2b(v);
+      }
+    }
+  }
+}
+
 // LDS-tiled bf16 transpose: out[C][R] = in[R][C]^T. 64x64 tiles, 16-B
 // coalesced loads AND stores (the scatter happens inside LDS where the
 // +8-element row pad keeps consecutive-row column reads conflict-free).
@@ -769,6 +939,23 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
     dim3 grid((unsigned)tiles, (unsigned)S);
     dim3 block(512);
     const int sm = (int)store;
+    // fully aligned large shapes take the 8-phase glds schedule
+    if (bm2 == 256 && bn2 == 256 && M % 256 == 0 && K % 64 == 0 &&
+        store == EpStore::kPlain) {
+      hipLaunchKernelGGL(gemm8p_kernel, grid, block, 0, cur_stream(), a2,
+                         b2, c, part_ptr, bs, M, N, K, kslice,
+                         relu ? 1 : 0);
+      HIP_CHECK(hipGetLastError());
+      if (S > 1) {
+        const long total = M * N;
+        int blocks = (int)std::min<long>((total + 31) / 32, 16384);
+        hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256),
+                           0, cur_stream(), part_ptr, (int)S, M, N, c, bs,
+                           relu ? 1 : 0, (int)store, ohw);
+        HIP_CHECK(hipGetLastError());
+      }
+      return;
+    }
     auto launch2 = [&](auto bmv, auto bnv) {
       hipLaunchKernelGGL(
           (gemm256_kernel<decltype(bmv)::value, decltype(bnv)::value>),
