@@ -540,21 +540,30 @@ __global__ void gemm8p_kernel(const bf16* __restrict__ A,
         else
           asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
-      __builtin_amdgcn_s_barrier();
-      if (p == 0) {  // (re)load the kk A-fragments
-#pragma unroll
-        for (int fm = 0; fm < FM; ++fm) {
-          const int byte = (wm0 + fm * 16 + l15) * 128 + kk * 64 + l4 * 16;
-          a8[fm] = *reinterpret_cast<const bf16x8_t*>(abase + swz8p(byte));
-        }
-      }
+      // issue this phase's ds_reads BEFORE the barrier (phases 1-3:
+      // the tile landed at phase 0, and the read latency then hides
+      // under the barrier wait); phase 0 must read after the
+      // vmcnt+barrier pair that publishes the tile.
       bf16x8_t b2[2];
+      auto load_frags = [&]() {
+        if (p == 0) {  // (re)load the kk A-fragments
 #pragma unroll
-      for (int j = 0; j < 2; ++j) {
-        const int byte =
-            (wn0 + (2 * p + j) * 16 + l15) * 128 + kk * 64 + l4 * 16;
-        b2[j] = *reinterpret_cast<const bf16x8_t*>(bbase + swz8p(byte));
-      }
+          for (int fm = 0; fm < FM; ++fm) {
+            const int byte =
+                (wm0 + fm * 16 + l15) * 128 + kk * 64 + l4 * 16;
+            a8[fm] = *reinterpret_cast<const bf16x8_t*>(abase + swz8p(byte));
+          }
+        }
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          const int byte =
+              (wn0 + (2 * p + j) * 16 + l15) * 128 + kk * 64 + l4 * 16;
+          b2[j] = *reinterpret_cast<const bf16x8_t*>(bbase + swz8p(byte));
+        }
+      };
+      if (q > 0) load_frags();
+      __builtin_amdgcn_s_barrier();
+      if (q == 0) load_frags();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
