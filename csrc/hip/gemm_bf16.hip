@@ -562,6 +562,11 @@ __global__ void gemm8p_kernel(const bf16* __restrict__ A,
         }
       };
       if (q > 0) load_frags();
+      // ONE barrier per phase: reads are issued pre-barrier, and the
+      // MFMA cluster touches no LDS, so the next phase's pre-barrier
+      // reads are ordered against this phase's by this barrier alone
+      // (cross-tile glds writes stay >= one barrier behind the last
+      // reads of the buffer they overwrite).
       __builtin_amdgcn_s_barrier();
       if (q == 0) load_frags();
       __builtin_amdgcn_s_setprio(1);
@@ -572,7 +577,6 @@ __global__ void gemm8p_kernel(const bf16* __restrict__ A,
           acc[fm][2 * p + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a8[fm], b2[j], acc[fm][2 * p + j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
-      __builtin_amdgcn_s_barrier();
     };
     do_phase(std::integral_constant<int, 0>{});
     do_phase(std::integral_constant<int, 1>{});
