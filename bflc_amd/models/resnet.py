@@ -92,9 +92,7 @@ class ResNet(FlatModel):
     def _cbr(self, x, conv, bnname, stride=1, pad=1, relu=True):
         h = O.conv2d(x, self.p(f"{conv}.w"), None, stride, pad)
         g, b = self.p(f"{bnname}.g"), self.p(f"{bnname}.b")
-        if relu:
-            return O.relu(O.batchnorm2d(h, g, b))
-        return O.batchnorm2d(h, g, b)
+        return O.batchnorm2d(h, g, b, relu=relu)  # fused BN(+relu)
 
     def _basic_block(self, x, p, stride):
         h = self._cbr(x, f"{p}.c1", f"{p}.bn1", stride, 1, relu=True)

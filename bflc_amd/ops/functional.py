@@ -234,10 +234,10 @@ class _BatchNormFn(torch.autograd.Function):
     x viewed [N*H*W, C] (csrc/hip/batchnorm.hip); CPU: fp32 oracle."""
 
     @staticmethod
-    def forward(ctx, x, gamma, beta, eps: float):
+    def forward(ctx, x, gamma, beta, eps: float, relu: bool):
         if x.is_cuda:
             y, mean, invstd = hip_ops().batchnorm_fwd(x, gamma, beta, eps,
-                                                      False)
+                                                      relu)
         else:
             xf = x.float()
             mean = xf.mean(dim=(0, 1, 2))
@@ -245,18 +245,27 @@ class _BatchNormFn(torch.autograd.Function):
             invstd = (var + eps).rsqrt()
             y = ((xf - mean) * invstd * gamma.float()
                  + beta.float()).to(x.dtype)
-        ctx.save_for_backward(x, gamma, mean, invstd)
+            if relu:
+                y = torch.relu(y)
+        ctx.relu = relu
+        if relu:
+            ctx.save_for_backward(x, gamma, mean, invstd, y)
+        else:
+            ctx.save_for_backward(x, gamma, mean, invstd)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, gamma, mean, invstd = ctx.saved_tensors
+        x, gamma, mean, invstd = ctx.saved_tensors[:4]
+        yr = ctx.saved_tensors[4] if ctx.relu else None
         dy = dy.contiguous()
         if x.is_cuda:
             dx, dgamma, dbeta = hip_ops().batchnorm_bwd(x, dy, mean, invstd,
-                                                        gamma)
+                                                        gamma, yr)
         else:
             xf, dyf = x.float(), dy.float()
+            if yr is not None:
+                dyf = dyf * (yr.float() > 0)
             n = x.numel() / x.shape[-1]
             xhat = (xf - mean) * invstd
             sdy = dyf.sum(dim=(0, 1, 2))
@@ -265,11 +274,14 @@ class _BatchNormFn(torch.autograd.Function):
             dx = dx.to(x.dtype)
             dgamma = sdyx.to(x.dtype)
             dbeta = sdy.to(x.dtype)
-        return dx, dgamma, dbeta, None
+        return dx, dgamma, dbeta, None, None
 
 
-def batchnorm2d(x, gamma, beta, eps: float = 1e-5) -> torch.Tensor:
-    return _BatchNormFn.apply(x, gamma, beta, eps)
+def batchnorm2d(x, gamma, beta, eps: float = 1e-5,
+                relu: bool = False) -> torch.Tensor:
+    """Batch-stats BN with optionally FUSED relu (one kernel fwd, the
+    relu mask folded into the backward reduction kernels)."""
+    return _BatchNormFn.apply(x, gamma, beta, eps, relu)
 
 
 class _GlobalAvgPoolFn(torch.autograd.Function):

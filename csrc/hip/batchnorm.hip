@@ -78,6 +78,7 @@ __global__ void bn_stats_part_vec_kernel(const bf16* __restrict__ x, long M,
 // bwd vectorized pass 1: partials of sum(dy), sum(dy*xhat)
 __global__ void bn_bwd_part_vec_kernel(const bf16* __restrict__ x,
                                        const bf16* __restrict__ dy,
+                                       const bf16* __restrict__ yr,
                                        const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
                                        long M, int C, int G, long chunk_rows,
@@ -99,7 +100,13 @@ __global__ void bn_bwd_part_vec_kernel(const bf16* __restrict__ x,
     }
     for (long m = r0 + rl; m < r1; m += RL) {
       const v8 xv = *reinterpret_cast<const v8*>(&x[m * C + c8]);
-      const v8 gv = *reinterpret_cast<const v8*>(&dy[m * C + c8]);
+      v8 gv = *reinterpret_cast<const v8*>(&dy[m * C + c8]);
+      if (yr) {
+        const v8 yv = *reinterpret_cast<const v8*>(&yr[m * C + c8]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (!(b2f(yv[j]) > 0.f)) gv[j] = (__bf16)0.f;
+      }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const float g = b2f(gv[j]);
@@ -343,6 +350,7 @@ __global__ void bn_norm_kernel(const bf16* __restrict__ x,
 // bwd pass 1: per-(chunk, channel) partials of sum(dy), sum(dy*xhat)
 __global__ void bn_bwd_part_kernel(const bf16* __restrict__ x,
                                    const bf16* __restrict__ dy,
+                                   const bf16* __restrict__ yr,
                                    const float* __restrict__ mean,
                                    const float* __restrict__ invstd, long M,
                                    int C, long chunk_rows,
@@ -356,7 +364,8 @@ __global__ void bn_bwd_part_kernel(const bf16* __restrict__ x,
   if (col < C) {
     const float mn = mean[col], is = invstd[col];
     for (long m = r0 + rlane; m < r1; m += 8) {
-      const float g = b2f(dy[m * C + col]);
+      float g = b2f(dy[m * C + col]);
+      if (yr && !(b2f(yr[m * C + col]) > 0.f)) g = 0.f;
       s1 += g;
       s2 += g * (b2f(x[m * C + col]) - mn) * is;
     }
@@ -413,6 +422,7 @@ __global__ void bn_bwd_final_kernel(const float* __restrict__ pdy,
 // bwd pass 2: dx = gamma*invstd*(dy - sdy/cnt - xhat*sdyx/cnt)
 __global__ void bn_bwd_dx_vec_kernel(const bf16* __restrict__ x,
                                      const bf16* __restrict__ dy,
+                                     const bf16* __restrict__ yr,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
                                      const bf16* __restrict__ gamma,
@@ -428,7 +438,13 @@ __global__ void bn_bwd_dx_vec_kernel(const bf16* __restrict__ x,
     const int c8 = (int)(g % c8g) * 8;
     const long i = (g / c8g) * C + c8;
     const bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(&x[i]);
-    const bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(&dy[i]);
+    bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(&dy[i]);
+    if (yr) {
+      const bf16x8_t yv = *reinterpret_cast<const bf16x8_t*>(&yr[i]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (!(b2f(yv[j]) > 0.f)) gv[j] = (__bf16)0.f;
+    }
     bf16x8_t out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -443,6 +459,7 @@ __global__ void bn_bwd_dx_vec_kernel(const bf16* __restrict__ x,
 
 __global__ void bn_bwd_dx_kernel(const bf16* __restrict__ x,
                                  const bf16* __restrict__ dy,
+                                 const bf16* __restrict__ yr,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
                                  const bf16* __restrict__ gamma,
@@ -454,9 +471,11 @@ __global__ void bn_bwd_dx_kernel(const bf16* __restrict__ x,
   const long stride = (long)gridDim.x * blockDim.x;
   for (; i < total; i += stride) {
     const int c = (int)(i % C);
+    float g = b2f(dy[i]);
+    if (yr && !(b2f(yr[i]) > 0.f)) g = 0.f;
     const float xhat = (b2f(x[i]) - mean[c]) * invstd[c];
     dx[i] = f2b(b2f(gamma[c]) * invstd[c] *
-                (b2f(dy[i]) - sdy[c] / count - xhat * sdyx[c] / count));
+                (g - sdy[c] / count - xhat * sdyx[c] / count));
   }
 }
 
@@ -636,8 +655,11 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
-    torch::Tensor invstd, torch::Tensor gamma) {
+    torch::Tensor invstd, torch::Tensor gamma,
+    c10::optional<torch::Tensor> y_relu) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(dy);
+  const bf16* yr =
+      y_relu.has_value() ? (const bf16*)y_relu->data_ptr() : nullptr;
   const int C = (int)x.size(-1);
   const long M = x.numel() / C;
   const int G = (C % 8 == 0) ? granule_lanes(C) : 0;
@@ -657,13 +679,15 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     hipLaunchKernelGGL(bn_bwd_part_vec_kernel, dim3(cblocks, chunks),
                        dim3(256), 0, cur_stream(),
                        (const bf16*)x.data_ptr(),
-                       (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
+                       (const bf16*)dy.data_ptr(), yr,
+                       mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), M, C, G, rows,
                        pdy.data_ptr<float>(), pdyx.data_ptr<float>());
   } else {
     hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(cblocks, chunks),
                        dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
-                       (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
+                       (const bf16*)dy.data_ptr(), yr,
+                       mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), M, C, rows,
                        pdy.data_ptr<float>(), pdyx.data_ptr<float>());
   }
@@ -688,14 +712,15 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     hipLaunchKernelGGL(bn_bwd_dx_vec_kernel, dim3(ew_grid(x.numel() / 8)),
                        dim3(1024), 0, cur_stream(),
                        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       yr, mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        (const bf16*)gc.data_ptr(), sdy.data_ptr<float>(),
                        sdyx.data_ptr<float>(), M, C, (float)M,
                        (bf16*)dx.data_ptr());
   else
     hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(ew_grid(x.numel())), dim3(1024),
                        0, cur_stream(), (const bf16*)x.data_ptr(),
-                       (const bf16*)dy.data_ptr(), mean.data_ptr<float>(),
+                       (const bf16*)dy.data_ptr(), yr,
+                       mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), (const bf16*)gc.data_ptr(),
                        sdy.data_ptr<float>(), sdyx.data_ptr<float>(), M, C,
                        (float)M, (bf16*)dx.data_ptr());

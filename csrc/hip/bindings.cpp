@@ -58,7 +58,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     bool relu);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
-    torch::Tensor invstd, torch::Tensor gamma);
+    torch::Tensor invstd, torch::Tensor gamma,
+    c10::optional<torch::Tensor> y_relu);
 torch::Tensor global_avgpool_fwd(torch::Tensor x);
 torch::Tensor global_avgpool_bwd(torch::Tensor dy, long H, long W);
 torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b);
@@ -95,7 +96,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2d_bwd", &bflc::maxpool2d_bwd);
   m.def("batchnorm_fwd", &bflc::batchnorm_fwd,
         "(y, mean, invstd) — batch-stats BN, optional fused relu");
-  m.def("batchnorm_bwd", &bflc::batchnorm_bwd, "(dx, dgamma, dbeta)");
+  m.def("batchnorm_bwd", &bflc::batchnorm_bwd, "(dx, dgamma, dbeta)",
+        py::arg("x"), py::arg("dy"), py::arg("mean"), py::arg("invstd"),
+        py::arg("gamma"), py::arg("y_relu") = py::none());
   m.def("global_avgpool_fwd", &bflc::global_avgpool_fwd);
   m.def("global_avgpool_bwd", &bflc::global_avgpool_bwd);
   m.def("add_relu_fwd", &bflc::add_relu_fwd, "fused residual add + relu");

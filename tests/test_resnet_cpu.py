@@ -30,6 +30,24 @@ class TestBatchNormOracle:
         assert torch.allclose(g.grad, g2.grad, atol=1e-4)
         assert torch.allclose(b.grad, b2.grad, atol=1e-4)
 
+    def test_fused_bn_relu_matches_unfused(self):
+        torch.manual_seed(2)
+        x = torch.randn(4, 7, 7, 8, requires_grad=True)
+        g = torch.rand(8, requires_grad=True) + 0.5
+        b = torch.randn(8, requires_grad=True)
+        y = O.batchnorm2d(x, g, b, relu=True)
+        dy = torch.randn_like(y)
+        (y * dy).sum().backward()
+        x2 = x.detach().clone().requires_grad_(True)
+        g2 = g.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        y2 = torch.relu(O.batchnorm2d(x2, g2, b2))
+        assert torch.allclose(y, y2)
+        (y2 * dy).sum().backward()
+        assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+        assert torch.allclose(g.grad, g2.grad, atol=1e-5)
+        assert torch.allclose(b.grad, b2.grad, atol=1e-5)
+
     def test_gap_and_add_relu(self):
         x = torch.randn(2, 5, 5, 4, requires_grad=True)  # NHWC
         y = O.global_avgpool(x)

@@ -53,6 +53,29 @@ class TestBatchNormGPU:
         assert_close(dgamma, g2.grad, rel=0.05)
         assert_close(dbeta, b2.grad, rel=0.05)
 
+    def test_fused_relu_bwd_mask(self):
+        torch.manual_seed(3)
+        x = torch.randn(8, 6, 6, 16)
+        g = torch.rand(16) + 0.5
+        b = torch.randn(16)
+        y, mean, invstd = hip().batchnorm_fwd(bf(x), bf(g), bf(b), 1e-5,
+                                              True)
+        assert float(y.min()) >= 0.0
+        dy = torch.randn_like(y.float().cpu())
+        dx, dg, db = hip().batchnorm_bwd(bf(x), bf(dy), mean, invstd,
+                                         bf(g), y)
+        # CPU oracle: relu mask then plain BN backward
+        xf = bf(x).float().cpu().requires_grad_(True)
+        gf = bf(g).float().cpu().requires_grad_(True)
+        bfl = bf(b).float().cpu().requires_grad_(True)
+        out = torch.relu(torch.nn.functional.batch_norm(
+            xf.permute(0, 3, 1, 2), None, None, gf, bfl, training=True,
+            eps=1e-5).permute(0, 2, 3, 1))
+        (out * bf(dy).float().cpu()).sum().backward()
+        assert_close(dx, xf.grad, rel=0.05)
+        assert_close(dg, gf.grad, rel=0.05)
+        assert_close(db, bfl.grad, rel=0.05)
+
     def test_fwd_deterministic(self):
         x = bf(torch.randn(8, 14, 14, 16))
         g, b = bf(torch.rand(16)), bf(torch.randn(16))
