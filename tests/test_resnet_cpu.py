@@ -33,7 +33,7 @@ class TestBatchNormOracle:
     def test_fused_bn_relu_matches_unfused(self):
         torch.manual_seed(2)
         x = torch.randn(4, 7, 7, 8, requires_grad=True)
-        g = torch.rand(8, requires_grad=True) + 0.5
+        g = (torch.rand(8) + 0.5).requires_grad_(True)
         b = torch.randn(8, requires_grad=True)
         y = O.batchnorm2d(x, g, b, relu=True)
         dy = torch.randn_like(y)
