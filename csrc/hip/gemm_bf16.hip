@@ -280,7 +280,7 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
   __shared__ __align__(16) bf16 As[2][BM][BKP2];
   __shared__ __align__(16) bf16 Bs[2][BN][BKP2];
 
-  const int ntn = (int)(N / BN);
+  const int ntn = (int)((N + BN - 1) / BN);
   const int ntm = (int)((M + BM - 1) / BM);
   const int nwg = ntm * ntn;
   int bid = blockIdx.x;
@@ -332,13 +332,13 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
       if (GBT < 512 && g >= GBT) break;
       const long gn = tile_n + (g >> 3);
       const long gk = k0 + (g & 7) * 8;
-      if (gk + 8 <= k_end) {
+      if (gn < N && gk + 8 <= k_end) {
         rb[i] = *reinterpret_cast<const bf16x8_t*>(&B[gn * K + gk]);
       } else {
         u16x8_t v;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          v[j] = (gk + j < k_end)
+          v[j] = (gn < N && gk + j < k_end)
                      ? *reinterpret_cast<const unsigned short*>(
                            &B[gn * K + gk + j])
                      : (unsigned short)0;
@@ -402,6 +402,7 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
 #pragma unroll
       for (int fn = 0; fn < FN; ++fn) {
         const long col = tile_n + wn0 + fn * 16 + l15;
+        if (col >= N) continue;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const long row = tile_m + wm0 + fm * 16 + l4 * 4 + r;
@@ -416,6 +417,7 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn) {
       const long col = tile_n + wn0 + fn * 16 + l15;
+      if (col >= N) continue;
       const float bv = bias ? b2f(bias[col]) : 0.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -906,12 +908,13 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   // per-tile efficiency from the measured ladder (64-tile structures
   // run far below the 256 ones).
   // (a BN=32 config measured slower than the synchronous 128x32 path
-  // on the shallow-K shapes it would serve; N%64 stays the floor)
-  if (N % 64 == 0 && K % 8 == 0 && M >= 48 && N >= 64 && K >= 32) {
+  // on the shallow-K shapes it would serve; BN=64 stays the floor, and
+  // N edges are zero-staged + epilogue-guarded like M edges)
+  if (N % 8 == 0 && K % 8 == 0 && M >= 48 && N >= 48 && K >= 32) {
     const long ksteps64 = (K + BK2 - 1) / BK2;
     const int bn2 = (N % 256 == 0) ? 256
                     : (N % 128 == 0 ? 128 : 64);
-    const long ntn2 = N / bn2;
+    const long ntn2 = (N + bn2 - 1) / bn2;
     int bm2 = 256;
     long S = 1;
     {
@@ -953,8 +956,8 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
     dim3 block(512);
     const int sm = (int)store;
     // fully aligned large shapes take the 8-phase glds schedule
-    if (bm2 == 256 && bn2 == 256 && M % 256 == 0 && K % 64 == 0 &&
-        store == EpStore::kPlain) {
+    if (bm2 == 256 && bn2 == 256 && M % 256 == 0 && N % 256 == 0 &&
+        K % 64 == 0 && store == EpStore::kPlain) {
       hipLaunchKernelGGL(gemm8p_kernel, grid, block, 0, cur_stream(), a2,
                          b2, c, part_ptr, bs, M, N, K, kslice,
                          relu ? 1 : 0);
