@@ -910,7 +910,10 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   // (a BN=32 config measured slower than the synchronous 128x32 path
   // on the shallow-K shapes it would serve; BN=64 stays the floor, and
   // N edges are zero-staged + epilogue-guarded like M edges)
-  if (N % 8 == 0 && K % 8 == 0 && M >= 48 && N >= 48 && K >= 32) {
+  // N%64 floor: widening to N%8 routed mid shapes onto narrow-BN
+  // configs that measured slower than the synchronous path (the edge
+  // guards stay for M and the K tail).
+  if (N % 64 == 0 && K % 8 == 0 && M >= 48 && N >= 64 && K >= 32) {
     const long ksteps64 = (K + BK2 - 1) / BK2;
     const int bn2 = (N % 256 == 0) ? 256
                     : (N % 128 == 0 ? 128 : 64);
