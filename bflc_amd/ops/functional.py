@@ -160,7 +160,9 @@ class _Conv2dFn(torch.autograd.Function):
                 x, w, b if b is not None else
                 torch.zeros(w.shape[0], device=x.device, dtype=x.dtype),
                 stride, padding)
-            # keep col for wgrad: recomputing im2col cost ~12% of a round
+            # keep col for wgrad when the fwd materialized one (the
+            # implicit-GEMM path returns an empty marker; bwd then
+            # builds its own)
             ctx.save_for_backward(x, w, col)
             return y
         ctx.save_for_backward(x, w)
@@ -174,6 +176,8 @@ class _Conv2dFn(torch.autograd.Function):
         dy = dy.contiguous()
         if x.is_cuda:
             col = ctx.saved_tensors[2]
+            if col.numel() == 0:
+                col = None
             dx, dw, db = hip_ops().conv2d_bwd(x, w, dy, ctx.stride,
                                               ctx.padding, col)
         else:

@@ -30,12 +30,6 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
 typedef __attribute__((ext_vector_type(8))) unsigned short u16x8_t;
 typedef __attribute__((ext_vector_type(8))) int i32x8_t;
 
-struct ConvShape {
-  int N, C, H, W, Kout, R, S, stride, pad, OH, OW;
-  __host__ __device__ long M() const { return (long)N * OH * OW; }
-  __host__ __device__ long RSC() const { return (long)R * S * C; }
-};
-
 // col[m][(r*S+s)*C + c8..] = x[n][ih][iw][c8..]  — one 16-B granule per
 // thread iteration; C % 8 == 0.
 __global__ void im2col_nhwc_vec_kernel(const bf16* __restrict__ x,
@@ -351,8 +345,17 @@ std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
   auto bc = b.contiguous();
   auto y = torch::empty({(long)sh.N, (long)sh.OH, (long)sh.OW,
                          (long)sh.Kout}, x.options());
-  torch::Tensor col = is_1x1_s1(sh) ? x.view({sh.M(), (long)sh.C})
-                                    : im2col(x, sh);
+  if (is_1x1_s1(sh)) {
+    auto col = x.view({sh.M(), (long)sh.C});
+    gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &bc,
+                  false, EpStore::kPlain, 0);
+    return {y, col};
+  }
+  // implicit-GEMM: the im2col gather runs inside the GEMM's A staging;
+  // no col matrix exists (wgrad materializes its own in the backward).
+  if (gemm_conv_fwd_raw(x, w2, y, sh, &bc, false))
+    return {y, torch::empty({0}, x.options())};
+  auto col = im2col(x, sh);
   gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &bc,
                 false, EpStore::kPlain, 0);
   return {y, col};

@@ -6,6 +6,20 @@
 
 namespace bflc {
 
+// NHWC convolution geometry (shared by conv2d.hip and the implicit-
+// GEMM staging in gemm_bf16.hip).
+#if defined(__HIPCC__) || defined(__HIP__)
+#define BFLC_HD __host__ __device__
+#else
+#define BFLC_HD
+#endif
+
+struct ConvShape {
+  int N, C, H, W, Kout, R, S, stride, pad, OH, OW;
+  BFLC_HD long M() const { return (long)N * OH * OW; }
+  BFLC_HD long RSC() const { return (long)R * S * C; }
+};
+
 // Epilogue store modes.
 enum class EpStore : int {
   kPlain = 0,   // C[m * N + n]
@@ -21,6 +35,14 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
                    torch::Tensor& C, long M, long N, long K, bool ta, bool tb,
                    const torch::Tensor* bias, bool relu, EpStore store,
                    long ohw);
+
+// Implicit-GEMM NHWC conv forward: y[M, Kout] = im2col(x) @ w2^T with
+// the im2col gather fused into the GEMM's A staging (no col matrix).
+// Requires sh.C % 8 == 0 and Kout % 64 == 0; returns false if the
+// shape cannot take this path (caller materializes col instead).
+bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
+                       torch::Tensor& y, const ConvShape& sh,
+                       const torch::Tensor* bias, bool relu);
 
 // colsum: out[n] = sum_m X[m,n]  (bias gradient)
 torch::Tensor colsum_bf16(const torch::Tensor& X);

@@ -263,14 +263,18 @@ constexpr int BKP2 = BK2 + 16;
 // Both operands in the vector-staging layout (A [M][K], B [N][K]; the
 // host pre-transposes anything else). M and K edges are bounds-guarded,
 // so only N % BN == 0 and K % 8 == 0 gate this path.
-template <int BM, int BN>
+// CONV = implicit-GEMM NHWC convolution forward: A is x [N,H,W,C] and
+// the im2col gather (k = (r*S+s)*C + c, C % 8 == 0 so a 16-B granule
+// stays inside one (r,s) window) happens in the A staging — the col
+// matrix is never materialized.
+template <int BM, int BN, bool CONV = false>
 __launch_bounds__(512, 1)
 __global__ void gemm256_kernel(const bf16* __restrict__ A,
                                const bf16* __restrict__ B,
                                bf16* __restrict__ C, float* __restrict__ Cpart,
                                const bf16* __restrict__ bias, long M, long N,
                                long K, long kslice, int relu, int store_mode,
-                               long ohw) {
+                               long ohw, ConvShape csh) {
   // wave grid: 2(M) x 4(N) for BN >= 64; 4(M) x 2(N) for BN == 32
   constexpr int WR = BN >= 64 ? 2 : 4, WC = BN >= 64 ? 4 : 2;
   constexpr int FM = BM / WR / 16, FN = BN / WC / 16;
@@ -307,12 +311,51 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
   // M rows beyond the matrix and K columns beyond k_end stage zeros
   // (zeros are MFMA-neutral), so edge tiles need no special kernel.
   bf16x8_t ra[GA], rb[GB];
-  auto load_tiles = [&](long k0) {
+  // implicit-conv per-granule row geometry (fixed across K-steps)
+  long cv_rowbase[CONV ? GA : 1];
+  int cv_ih0[CONV ? GA : 1], cv_iw0[CONV ? GA : 1];
+  bool cv_mok[CONV ? GA : 1];
+  if (CONV) {
 #pragma unroll
     for (int i = 0; i < GA; ++i) {
       const int g = tid + i * 512;
       const long gm = tile_m + (g >> 3);
+      cv_mok[i] = gm < M;
+      const long m = cv_mok[i] ? gm : 0;
+      const int ow = (int)(m % csh.OW);
+      const int oh = (int)((m / csh.OW) % csh.OH);
+      const int n = (int)(m / ((long)csh.OW * csh.OH));
+      cv_rowbase[i] = (long)n * csh.H * csh.W * csh.C;
+      cv_ih0[i] = oh * csh.stride - csh.pad;
+      cv_iw0[i] = ow * csh.stride - csh.pad;
+    }
+  }
+  auto load_tiles = [&](long k0) {
+#pragma unroll
+    for (int i = 0; i < GA; ++i) {
+      const int g = tid + i * 512;
       const long gk = k0 + (g & 7) * 8;
+      if (CONV) {
+        bool ok = cv_mok[i] && gk < k_end;
+        int ih = 0, iw = 0, c8 = 0;
+        if (ok) {
+          const int rs = (int)(gk / csh.C);
+          c8 = (int)(gk - (long)rs * csh.C);
+          const int rr = rs / csh.S, ss = rs - rr * csh.S;
+          ih = cv_ih0[i] + rr;
+          iw = cv_iw0[i] + ss;
+          ok = ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W;
+        }
+        if (ok) {
+          ra[i] = *reinterpret_cast<const bf16x8_t*>(
+              &A[cv_rowbase[i] + ((long)ih * csh.W + iw) * csh.C + c8]);
+        } else {
+          u16x8_t z = {};
+          ra[i] = *reinterpret_cast<const bf16x8_t*>(&z);
+        }
+        continue;
+      }
+      const long gm = tile_m + (g >> 3);
       if (gm < M && gk + 8 <= k_end) {
         ra[i] = *reinterpret_cast<const bf16x8_t*>(&A[gm * K + gk]);
       } else {
@@ -979,7 +1022,7 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
       hipLaunchKernelGGL(
           (gemm256_kernel<decltype(bmv)::value, decltype(bnv)::value>),
           grid, block, 0, cur_stream(), a2, b2, c, part_ptr, bs, M, N, K,
-          kslice, relu, sm, ohw);
+          kslice, relu, sm, ohw, ConvShape{});
     };
     using c64i = std::integral_constant<int, 64>;
     using c128i = std::integral_constant<int, 128>;
@@ -1129,6 +1172,76 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
 }
 
 // Raw GEMM entry (benchmark/ablation): C[M,N] = op(A) @ op(B).
+bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
+                       torch::Tensor& y, const ConvShape& sh,
+                       const torch::Tensor* bias, bool relu) {
+  const long M = sh.M(), N = sh.Kout, K = sh.RSC();
+  if (sh.C % 8 != 0 || N % 64 != 0 || N < 64 || M < 48) return false;
+  const long ksteps64 = (K + BK2 - 1) / BK2;
+  const int bn2 = (N % 256 == 0) ? 256 : (N % 128 == 0 ? 128 : 64);
+  const long ntn2 = (N + bn2 - 1) / bn2;
+  int bm2 = 256;
+  long S = 1;
+  {
+    double best_t = 1e30;
+    const double bn_pen = bn2 >= 128 ? 1.0 : 0.75;
+    for (int bm : {256, 128, 64}) {
+      const double tf = (bm == 256 ? 800.0e12
+                                   : (bm == 128 ? 600.0e12 : 320.0e12)) *
+                        bn_pen;
+      const long tiles_c = ((M + bm - 1) / bm) * ntn2;
+      double t;
+      const long ss = pick_splitk(M, N, K, tiles_c, ksteps64, 256, tf, &t);
+      if (t < best_t) { best_t = t; bm2 = bm; S = ss; }
+    }
+  }
+  const long tiles = ((M + bm2 - 1) / bm2) * ntn2;
+  const long kslice = ((ksteps64 + S - 1) / S) * BK2;
+  S = (K + kslice - 1) / kslice;
+  torch::Tensor part;
+  float* part_ptr = nullptr;
+  if (S > 1) {
+    part = torch::empty({S, M, N}, x.options().dtype(at::kFloat));
+    part_ptr = part.data_ptr<float>();
+  }
+  const bf16* a = (const bf16*)x.data_ptr();
+  const bf16* b = (const bf16*)w2.data_ptr();
+  bf16* c = (bf16*)y.data_ptr();
+  const bf16* bs = bias ? (const bf16*)bias->data_ptr() : nullptr;
+  dim3 grid((unsigned)tiles, (unsigned)S);
+  dim3 block(512);
+  auto launchc = [&](auto bmv, auto bnv) {
+    hipLaunchKernelGGL(
+        (gemm256_kernel<decltype(bmv)::value, decltype(bnv)::value, true>),
+        grid, block, 0, cur_stream(), a, b, c, part_ptr, bs, M, N, K,
+        kslice, relu ? 1 : 0, (int)EpStore::kPlain, 0, sh);
+  };
+  using c64i = std::integral_constant<int, 64>;
+  using c128i = std::integral_constant<int, 128>;
+  using c256i = std::integral_constant<int, 256>;
+  switch (bm2 * 1000 + bn2) {
+    case 256256: launchc(c256i{}, c256i{}); break;
+    case 256128: launchc(c256i{}, c128i{}); break;
+    case 256064: launchc(c256i{}, c64i{}); break;
+    case 128256: launchc(c128i{}, c256i{}); break;
+    case 128128: launchc(c128i{}, c128i{}); break;
+    case 128064: launchc(c128i{}, c64i{}); break;
+    case  64256: launchc(c64i{}, c256i{}); break;
+    case  64128: launchc(c64i{}, c128i{}); break;
+    case  64064: launchc(c64i{}, c64i{}); break;
+  }
+  HIP_CHECK(hipGetLastError());
+  if (S > 1) {
+    const long total = M * N;
+    int blocks = (int)std::min<long>((total + 31) / 32, 16384);
+    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
+                       cur_stream(), part_ptr, (int)S, M, N, c, bs,
+                       relu ? 1 : 0, (int)EpStore::kPlain, 0);
+    HIP_CHECK(hipGetLastError());
+  }
+  return true;
+}
+
 torch::Tensor gemm_raw(torch::Tensor A, torch::Tensor B, bool ta, bool tb) {
   CHECK_GPU(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
   const long M = ta ? A.size(1) : A.size(0);
