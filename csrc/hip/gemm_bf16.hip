@@ -48,14 +48,15 @@ constexpr int BKP = BK + 8; // padded row length (16 B) vs bank conflicts
 // blockIdx.x = tile (XCD-swizzled), blockIdx.y = K slice (split-K).
 // If Cpart != nullptr: write fp32 partials at Cpart[slice*M*N + ...] and
 // skip bias/relu (applied by the reduce kernel). kslice = K per slice.
-template <int BM, int BN, int WR, int WC, bool TA, bool TB>
+template <int BM, int BN, int WR, int WC, bool TA, bool TB,
+          bool CONV = false>
 __launch_bounds__(WR * WC * 64)
 __global__ void gemm_kernel(const bf16* __restrict__ A,
                             const bf16* __restrict__ B,
                             bf16* __restrict__ C, float* __restrict__ Cpart,
                             const bf16* __restrict__ bias, long M, long N,
                             long K, long kslice, int relu, int store_mode,
-                            long ohw, int vecA, int vecB) {
+                            long ohw, int vecA, int vecB, ConvShape csh) {
   constexpr int THREADS = WR * WC * 64;
   constexpr int FM = BM / WR / 16;
   constexpr int FN = BN / WC / 16;
@@ -90,7 +91,39 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
   const bf16 zero = f2b(0.f);
   for (long k0 = k_begin; k0 < k_end; k0 += BK) {
     // ---- stage A tile (16-byte vector path when layout permits) ----
-    if (vecA) {
+    if (CONV) {  // implicit NHWC im2col gather (C % 8 == 0)
+      constexpr int GROUPS = (BM * BK) / 8;
+#pragma unroll
+      for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
+        const int g = tid + i * THREADS;
+        if (GROUPS % THREADS != 0 && g >= GROUPS) break;
+        const int m = g / (BK / 8), k8 = (g % (BK / 8)) * 8;
+        const long gm = tile_m + m;
+        const long gk = k0 + k8;
+        bool ok = gm < M && gk < k_end;
+        long src = 0;
+        if (ok) {
+          const int rs = (int)(gk / csh.C);
+          const int c8 = (int)(gk - (long)rs * csh.C);
+          const int rr = rs / csh.S, ss = rs - rr * csh.S;
+          const int ow = (int)(gm % csh.OW);
+          const int oh = (int)((gm / csh.OW) % csh.OH);
+          const int nn = (int)(gm / ((long)csh.OW * csh.OH));
+          const int ih = oh * csh.stride - csh.pad + rr;
+          const int iw = ow * csh.stride - csh.pad + ss;
+          ok = ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W;
+          src = (((long)nn * csh.H + ih) * csh.W + iw) * csh.C + c8;
+        }
+        if (ok) {
+          *reinterpret_cast<bf16x8_t*>(&As[m][k8]) =
+              *reinterpret_cast<const bf16x8_t*>(&A[src]);
+        } else {
+          u16x8_t z = {};
+          *reinterpret_cast<bf16x8_t*>(&As[m][k8]) =
+              *reinterpret_cast<const bf16x8_t*>(&z);
+        }
+      }
+    } else if (vecA) {
       constexpr int GROUPS = (BM * BK) / 8;
 #pragma unroll
       for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
@@ -1083,19 +1116,23 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
     if (!ta && !tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, false, false>),
                          grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
-                         M, N, K, kslice, relu, sm, ohw, vecA, vecB);
+                         M, N, K, kslice, relu, sm, ohw, vecA, vecB,
+                         ConvShape{});
     else if (!ta && tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, false, true>),
                          grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
-                         M, N, K, kslice, relu, sm, ohw, vecA, vecB);
+                         M, N, K, kslice, relu, sm, ohw, vecA, vecB,
+                         ConvShape{});
     else if (ta && !tb)
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, true, false>),
                          grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
-                         M, N, K, kslice, relu, sm, ohw, vecA, vecB);
+                         M, N, K, kslice, relu, sm, ohw, vecA, vecB,
+                         ConvShape{});
     else
       hipLaunchKernelGGL((gemm_kernel<BMv, BNv, WRv, WCv, true, true>),
                          grid, block, 0, cur_stream(), a, b, c, part_ptr, bs,
-                         M, N, K, kslice, relu, sm, ohw, vecA, vecB);
+                         M, N, K, kslice, relu, sm, ohw, vecA, vecB,
+                         ConvShape{});
   };
 
   using c32 = std::integral_constant<int, 32>;
@@ -1176,7 +1213,66 @@ bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
                        torch::Tensor& y, const ConvShape& sh,
                        const torch::Tensor* bias, bool relu) {
   const long M = sh.M(), N = sh.Kout, K = sh.RSC();
-  if (sh.C % 8 != 0 || N % 64 != 0 || N < 64 || M < 48) return false;
+  if (sh.C % 8 != 0 || M < 48) return false;
+  if (N % 64 != 0 || N < 64) {
+    // narrow-N convs (ResNet-20's Kout 16/32): implicit gather in the
+    // synchronous small-tile kernel — the col read it replaces is the
+    // traffic bound there (col is R*S times the activation)
+    const bf16* a = (const bf16*)x.data_ptr();
+    const bf16* b = (const bf16*)w2.data_ptr();
+    bf16* c = (bf16*)y.data_ptr();
+    const bf16* bs = bias ? (const bf16*)bias->data_ptr() : nullptr;
+    const TileCfg t = pick_tile(M, N);
+    const long tiles = ((M + t.bm - 1) / t.bm) * ((N + t.bn - 1) / t.bn);
+    const long ksteps = (K + BK - 1) / BK;
+    long S = pick_splitk(M, N, K, tiles, ksteps, 512, 150.0e12);
+    const long kslice = ((ksteps + S - 1) / S) * BK;
+    S = (K + kslice - 1) / kslice;
+    torch::Tensor part;
+    float* part_ptr = nullptr;
+    if (S > 1) {
+      part = torch::empty({S, M, N}, x.options().dtype(at::kFloat));
+      part_ptr = part.data_ptr<float>();
+    }
+    dim3 grid((unsigned)tiles, (unsigned)S);
+    auto launchcc = [&](auto bm, auto bn, auto wr, auto wc) {
+      constexpr int BMv = decltype(bm)::value, BNv = decltype(bn)::value;
+      constexpr int WRv = decltype(wr)::value, WCv = decltype(wc)::value;
+      dim3 block(WRv * WCv * 64);
+      hipLaunchKernelGGL(
+          (gemm_kernel<BMv, BNv, WRv, WCv, false, true, true>), grid, block,
+          0, cur_stream(), a, b, c, part_ptr, bs, M, N, K, kslice,
+          relu ? 1 : 0, (int)EpStore::kPlain, 0, 0, 1, sh);
+    };
+    using c32 = std::integral_constant<int, 32>;
+    using c64 = std::integral_constant<int, 64>;
+    using c128 = std::integral_constant<int, 128>;
+    using c1 = std::integral_constant<int, 1>;
+    using c2 = std::integral_constant<int, 2>;
+    using c4 = std::integral_constant<int, 4>;
+    switch (t.bm * 1000 + t.bn) {
+      case 128128: launchcc(c128{}, c128{}, c2{}, c2{}); break;
+      case 128064: launchcc(c128{}, c64{}, c2{}, c2{}); break;
+      case 128032: launchcc(c128{}, c32{}, c2{}, c2{}); break;
+      case  64128: launchcc(c64{}, c128{}, c2{}, c2{}); break;
+      case  64064: launchcc(c64{}, c64{}, c2{}, c2{}); break;
+      case  64032: launchcc(c64{}, c32{}, c2{}, c2{}); break;
+      case  32128: launchcc(c32{}, c128{}, c1{}, c4{}); break;
+      case  32064: launchcc(c32{}, c64{}, c1{}, c4{}); break;
+      case  32032: launchcc(c32{}, c32{}, c2{}, c1{}); break;
+      default: return false;
+    }
+    HIP_CHECK(hipGetLastError());
+    if (S > 1) {
+      const long total = M * N;
+      int blocks = (int)std::min<long>((total + 31) / 32, 16384);
+      hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
+                         cur_stream(), part_ptr, (int)S, M, N, c, bs,
+                         relu ? 1 : 0, (int)EpStore::kPlain, 0);
+      HIP_CHECK(hipGetLastError());
+    }
+    return true;
+  }
   const long ksteps64 = (K + BK2 - 1) / BK2;
   const int bn2 = (N % 256 == 0) ? 256 : (N % 128 == 0 ? 128 : 64);
   const long ntn2 = (N + bn2 - 1) / bn2;
