@@ -374,17 +374,27 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   auto w2 = w.view({(long)sh.Kout, sh.RSC()});
   auto dy2 = dy.view({sh.M(), (long)sh.Kout});  // NHWC: free view
 
-  // dgrad: dcol[M, RSC] = dy2[M, Kout] @ W[Kout, RSC]. Pre-transposing
-  // the (small) weight puts B in the vector-staging [N][K] layout for
-  // every tile path.
-  auto wT = transpose_bf16(w2);  // [RSC, Kout]
+  // dgrad
   torch::Tensor dx;
+  bool dgrad_done = false;
   if (is_1x1_s1(sh)) {
+    auto wT = transpose_bf16(w2);  // [C, Kout]
     dx = torch::empty_like(x);
     auto dxv = dx.view({sh.M(), (long)sh.C});
     gemm_bf16_raw(dy2, wT, dxv, sh.M(), sh.RSC(), sh.Kout, false, true,
                   nullptr, false, EpStore::kPlain, 0);
-  } else {
+    dgrad_done = true;
+  } else if (sh.Kout % 8 == 0) {
+    // implicit: dx[M, C] = dy-gather @ w.permute(3,1,2,0) — no dcol
+    // matrix, no col2im pass
+    auto wrot2 = w.permute({3, 1, 2, 0}).contiguous()
+                     .view({(long)sh.C, (long)sh.R * sh.S * sh.Kout});
+    dx = torch::empty_like(x);
+    auto dxv = dx.view({(long)sh.N * sh.H * sh.W, (long)sh.C});
+    dgrad_done = gemm_conv_dgrad_raw(dy2, wrot2, dxv, sh);
+  }
+  if (!dgrad_done) {
+    auto wT = transpose_bf16(w2);  // [RSC, Kout]
     auto dcol = torch::empty({sh.M(), sh.RSC()}, x.options());
     gemm_bf16_raw(dy2, wT, dcol, sh.M(), sh.RSC(), sh.Kout, false, true,
                   nullptr, false, EpStore::kPlain, 0);

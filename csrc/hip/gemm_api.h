@@ -44,6 +44,14 @@ bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
                        torch::Tensor& y, const ConvShape& sh,
                        const torch::Tensor* bias, bool relu);
 
+// Implicit-GEMM NHWC conv data-gradient: dx[M, C] = dy-gather @ wrot2
+// (wrot2[c][(r,s,kout)] = w[kout][r][s][c], i.e. w.permute(3,1,2,0)
+// viewed [C, R*S*Kout]; r,s indices walk the SAME orientation as the
+// forward because the gather uses oh = (ih + pad - r)/stride).
+// Requires sh.Kout % 8 == 0; false => caller uses dcol + col2im.
+bool gemm_conv_dgrad_raw(const torch::Tensor& dy, const torch::Tensor& wrot2,
+                         torch::Tensor& dx, const ConvShape& sh);
+
 // colsum: out[n] = sum_m X[m,n]  (bias gradient)
 torch::Tensor colsum_bf16(const torch::Tensor& X);
 

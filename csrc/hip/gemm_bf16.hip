@@ -49,7 +49,7 @@ constexpr int BKP = BK + 8; // padded row length (16 B) vs bank conflicts
 // If Cpart != nullptr: write fp32 partials at Cpart[slice*M*N + ...] and
 // skip bias/relu (applied by the reduce kernel). kslice = K per slice.
 template <int BM, int BN, int WR, int WC, bool TA, bool TB,
-          bool CONV = false>
+          int CMODE = 0>
 __launch_bounds__(WR * WC * 64)
 __global__ void gemm_kernel(const bf16* __restrict__ A,
                             const bf16* __restrict__ B,
@@ -91,7 +91,7 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
   const bf16 zero = f2b(0.f);
   for (long k0 = k_begin; k0 < k_end; k0 += BK) {
     // ---- stage A tile (16-byte vector path when layout permits) ----
-    if (CONV) {  // implicit NHWC im2col gather (C % 8 == 0)
+    if (CMODE != 0) {  // implicit NHWC conv gather (granule % 8)
       constexpr int GROUPS = (BM * BK) / 8;
 #pragma unroll
       for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
@@ -102,7 +102,7 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         const long gk = k0 + k8;
         bool ok = gm < M && gk < k_end;
         long src = 0;
-        if (ok) {
+        if (ok && CMODE == 1) {  // fwd: x gather
           const int rs = (int)(gk / csh.C);
           const int c8 = (int)(gk - (long)rs * csh.C);
           const int rr = rs / csh.S, ss = rs - rr * csh.S;
@@ -113,6 +113,21 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
           const int iw = ow * csh.stride - csh.pad + ss;
           ok = ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W;
           src = (((long)nn * csh.H + ih) * csh.W + iw) * csh.C + c8;
+        } else if (ok) {  // dgrad: dy gather, flipped correlation
+          const int rs = (int)(gk / csh.Kout);
+          const int kk8 = (int)(gk - (long)rs * csh.Kout);
+          const int rr = rs / csh.S, ss = rs - rr * csh.S;
+          const int iw = (int)(gm % csh.W);
+          const int ih = (int)((gm / csh.W) % csh.H);
+          const int nn = (int)(gm / ((long)csh.W * csh.H));
+          const int oh_num = ih + csh.pad - rr;
+          const int ow_num = iw + csh.pad - ss;
+          const int oh = oh_num / csh.stride;
+          const int ow = ow_num / csh.stride;
+          ok = oh_num >= 0 && ow_num >= 0 &&
+               oh_num % csh.stride == 0 && ow_num % csh.stride == 0 &&
+               oh < csh.OH && ow < csh.OW;
+          src = (((long)nn * csh.OH + oh) * csh.OW + ow) * csh.Kout + kk8;
         }
         if (ok) {
           *reinterpret_cast<bf16x8_t*>(&As[m][k8]) =
@@ -296,11 +311,15 @@ constexpr int BKP2 = BK2 + 16;
 // Both operands in the vector-staging layout (A [M][K], B [N][K]; the
 // host pre-transposes anything else). M and K edges are bounds-guarded,
 // so only N % BN == 0 and K % 8 == 0 gate this path.
-// CONV = implicit-GEMM NHWC convolution forward: A is x [N,H,W,C] and
-// the im2col gather (k = (r*S+s)*C + c, C % 8 == 0 so a 16-B granule
-// stays inside one (r,s) window) happens in the A staging — the col
-// matrix is never materialized.
-template <int BM, int BN, bool CONV = false>
+// CMODE 1 = implicit-GEMM NHWC conv forward: A is x [N,H,W,C] and the
+// im2col gather (k = (r*S+s)*C + c, C % 8 == 0 so a 16-B granule stays
+// inside one (r,s) window) happens in the A staging — no col matrix.
+// CMODE 2 = implicit dgrad: A is dy [N,OH,OW,Kout], k = (r*S+s)*Kout +
+// kout (Kout % 8 == 0), gathered with the flipped-correlation geometry
+// oh = (ih + pad - r) / stride (zero when misaligned/out of range) —
+// no dcol matrix and no col2im pass. B is the pre-permuted
+// wrot[c][(r,s,kout)] = w[kout][r][s][c].
+template <int BM, int BN, int CMODE = 0>
 __launch_bounds__(512, 1)
 __global__ void gemm256_kernel(const bf16* __restrict__ A,
                                const bf16* __restrict__ B,
@@ -345,10 +364,10 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
   // (zeros are MFMA-neutral), so edge tiles need no special kernel.
   bf16x8_t ra[GA], rb[GB];
   // implicit-conv per-granule row geometry (fixed across K-steps)
-  long cv_rowbase[CONV ? GA : 1];
-  int cv_ih0[CONV ? GA : 1], cv_iw0[CONV ? GA : 1];
-  bool cv_mok[CONV ? GA : 1];
-  if (CONV) {
+  long cv_rowbase[CMODE ? GA : 1];
+  int cv_ih0[CMODE ? GA : 1], cv_iw0[CMODE ? GA : 1];
+  bool cv_mok[CMODE ? GA : 1];
+  if (CMODE == 1) {
 #pragma unroll
     for (int i = 0; i < GA; ++i) {
       const int g = tid + i * 512;
@@ -362,13 +381,27 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
       cv_ih0[i] = oh * csh.stride - csh.pad;
       cv_iw0[i] = ow * csh.stride - csh.pad;
     }
+  } else if (CMODE == 2) {
+#pragma unroll
+    for (int i = 0; i < GA; ++i) {
+      const int g = tid + i * 512;
+      const long gm = tile_m + (g >> 3);  // row of dx: (n, ih, iw)
+      cv_mok[i] = gm < M;
+      const long m = cv_mok[i] ? gm : 0;
+      const int iw = (int)(m % csh.W);
+      const int ih = (int)((m / csh.W) % csh.H);
+      const int n = (int)(m / ((long)csh.W * csh.H));
+      cv_rowbase[i] = (long)n * csh.OH * csh.OW * csh.Kout;
+      cv_ih0[i] = ih + csh.pad;  // oh_num = ih + pad - r
+      cv_iw0[i] = iw + csh.pad;
+    }
   }
   auto load_tiles = [&](long k0) {
 #pragma unroll
     for (int i = 0; i < GA; ++i) {
       const int g = tid + i * 512;
       const long gk = k0 + (g & 7) * 8;
-      if (CONV) {
+      if (CMODE == 1) {
         bool ok = cv_mok[i] && gk < k_end;
         int ih = 0, iw = 0, c8 = 0;
         if (ok) {
@@ -382,6 +415,30 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
         if (ok) {
           ra[i] = *reinterpret_cast<const bf16x8_t*>(
               &A[cv_rowbase[i] + ((long)ih * csh.W + iw) * csh.C + c8]);
+        } else {
+          u16x8_t z = {};
+          ra[i] = *reinterpret_cast<const bf16x8_t*>(&z);
+        }
+        continue;
+      }
+      if (CMODE == 2) {
+        bool ok = cv_mok[i] && gk < k_end;
+        int oh = 0, ow = 0, k8 = 0;
+        if (ok) {
+          const int rs = (int)(gk / csh.Kout);
+          k8 = (int)(gk - (long)rs * csh.Kout);
+          const int rr = rs / csh.S, ss = rs - rr * csh.S;
+          const int oh_num = cv_ih0[i] - rr;
+          const int ow_num = cv_iw0[i] - ss;
+          oh = oh_num / csh.stride;
+          ow = ow_num / csh.stride;
+          ok = oh_num >= 0 && ow_num >= 0 &&
+               oh_num % csh.stride == 0 && ow_num % csh.stride == 0 &&
+               oh < csh.OH && ow < csh.OW;
+        }
+        if (ok) {
+          ra[i] = *reinterpret_cast<const bf16x8_t*>(
+              &A[cv_rowbase[i] + ((long)oh * csh.OW + ow) * csh.Kout + k8]);
         } else {
           u16x8_t z = {};
           ra[i] = *reinterpret_cast<const bf16x8_t*>(&z);
@@ -1209,11 +1266,17 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
 }
 
 // Raw GEMM entry (benchmark/ablation): C[M,N] = op(A) @ op(B).
-bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
-                       torch::Tensor& y, const ConvShape& sh,
-                       const torch::Tensor* bias, bool relu) {
-  const long M = sh.M(), N = sh.Kout, K = sh.RSC();
-  if (sh.C % 8 != 0 || M < 48) return false;
+template <int CMODE>
+bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
+                        torch::Tensor& y, const ConvShape& sh,
+                        const torch::Tensor* bias, bool relu) {
+  // CMODE 1 (fwd):   y[M=N*OH*OW][Kout], K = R*S*C, A = x gather
+  // CMODE 2 (dgrad): y[M=N*H*W][C],      K = R*S*Kout, A = dy gather
+  const long M = CMODE == 1 ? sh.M() : (long)sh.N * sh.H * sh.W;
+  const long N = CMODE == 1 ? sh.Kout : sh.C;
+  const long K = CMODE == 1 ? sh.RSC() : (long)sh.R * sh.S * sh.Kout;
+  const int inner = CMODE == 1 ? sh.C : sh.Kout;  // granule dimension
+  if (inner % 8 != 0 || M < 48) return false;
   if (N % 64 != 0 || N < 64) {
     // narrow-N convs (ResNet-20's Kout 16/32): implicit gather in the
     // synchronous small-tile kernel — the col read it replaces is the
@@ -1240,8 +1303,8 @@ bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
       constexpr int WRv = decltype(wr)::value, WCv = decltype(wc)::value;
       dim3 block(WRv * WCv * 64);
       hipLaunchKernelGGL(
-          (gemm_kernel<BMv, BNv, WRv, WCv, false, true, true>), grid, block,
-          0, cur_stream(), a, b, c, part_ptr, bs, M, N, K, kslice,
+          (gemm_kernel<BMv, BNv, WRv, WCv, false, true, CMODE>), grid,
+          block, 0, cur_stream(), a, b, c, part_ptr, bs, M, N, K, kslice,
           relu ? 1 : 0, (int)EpStore::kPlain, 0, 0, 1, sh);
     };
     using c32 = std::integral_constant<int, 32>;
@@ -1308,7 +1371,7 @@ bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
   dim3 block(512);
   auto launchc = [&](auto bmv, auto bnv) {
     hipLaunchKernelGGL(
-        (gemm256_kernel<decltype(bmv)::value, decltype(bnv)::value, true>),
+        (gemm256_kernel<decltype(bmv)::value, decltype(bnv)::value, CMODE>),
         grid, block, 0, cur_stream(), a, b, c, part_ptr, bs, M, N, K,
         kslice, relu ? 1 : 0, (int)EpStore::kPlain, 0, sh);
   };
@@ -1336,6 +1399,17 @@ bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
     HIP_CHECK(hipGetLastError());
   }
   return true;
+}
+
+bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
+                       torch::Tensor& y, const ConvShape& sh,
+                       const torch::Tensor* bias, bool relu) {
+  return conv_implicit_gemm<1>(x, w2, y, sh, bias, relu);
+}
+
+bool gemm_conv_dgrad_raw(const torch::Tensor& dy, const torch::Tensor& wrot2,
+                         torch::Tensor& dx, const ConvShape& sh) {
+  return conv_implicit_gemm<2>(dy, wrot2, dx, sh, nullptr, false);
 }
 
 torch::Tensor gemm_raw(torch::Tensor A, torch::Tensor B, bool ta, bool tb) {
