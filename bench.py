@@ -29,6 +29,9 @@ def main() -> None:
                              "mlp", "logreg"])
     ap.add_argument("--samples-per-client", type=int, default=None)
     ap.add_argument("--batch-size", type=int, default=None)
+    ap.add_argument("--byzantine", type=int, default=0,
+                    help="label-flip attacker clients (BASELINE config 4; "
+                         "committee scoring is the defense)")
     ap.add_argument("--eval", action="store_true", help="eval every round")
     ap.add_argument("--phases", action="store_true",
                     help="print per-phase timings of each timed round")
@@ -60,7 +63,7 @@ def main() -> None:
         samples_per_client=spc,
         batch_size=bs, partition="dirichlet",
         dirichlet_alpha=0.3, eval_samples=min(4096, 2 * spc),
-        learning_rate=0.01)
+        learning_rate=0.01, byzantine_clients=args.byzantine)
 
     t = Transport()
     shards, test = make_federated(cfg)
@@ -133,6 +136,7 @@ def main() -> None:
                 "samples_per_client": cfg.samples_per_client,
                 "batch_size": cfg.batch_size,
                 "partition": cfg.partition,
+                "byzantine_clients": cfg.byzantine_clients,
             },
         }
         print(json.dumps(out), flush=True)
