@@ -29,6 +29,9 @@ def main() -> None:
                              "mlp", "logreg"])
     ap.add_argument("--samples-per-client", type=int, default=None)
     ap.add_argument("--batch-size", type=int, default=None)
+    ap.add_argument("--clients", type=int, default=None,
+                    help="FL clients (default: one per rank; more than "
+                         "ranks = multiple clients per GPU, reference-style)")
     ap.add_argument("--byzantine", type=int, default=0,
                     help="label-flip attacker clients (BASELINE config 4; "
                          "committee scoring is the defense)")
@@ -58,8 +61,9 @@ def main() -> None:
     bs = args.batch_size or md["bs"]
 
     n = int(os.environ.get("WORLD_SIZE", args.gpus))
+    n_clients = args.clients or n
     cfg = FLConfig.for_world(
-        n, model=args.model, n_class=md["n_class"],
+        n_clients, model=args.model, n_class=md["n_class"],
         samples_per_client=spc,
         batch_size=bs, partition="dirichlet",
         dirichlet_alpha=0.3, eval_samples=min(4096, 2 * spc),
