@@ -421,14 +421,21 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     HIP_CHECK(hipGetLastError());
   }
 
-  // wgrad: dW[Kout, RSC] = dy2^T @ col (col reused from fwd when given)
-  auto col = col_cache.has_value()
-                 ? *col_cache
-                 : (is_1x1_s1(sh) ? x.view({sh.M(), (long)sh.C})
-                                  : im2col(x, sh));
+  // wgrad: dW[Kout, RSC] = dy2^T @ col. Implicit (col gathered from x
+  // inside the GEMM staging) when C % 8 == 0; else a materialized col
+  // (reused from the fwd when it produced one).
   auto dw = torch::empty_like(w2);
-  gemm_bf16_raw(dy2, col, dw, sh.Kout, sh.RSC(), sh.M(), true, false,
-                nullptr, false, EpStore::kPlain, 0);
+  bool wgrad_done = false;
+  if (!is_1x1_s1(sh) && !(col_cache.has_value() && col_cache->numel() > 0))
+    wgrad_done = gemm_conv_wgrad_raw(dy2, x, dw, sh);
+  if (!wgrad_done) {
+    auto col = (col_cache.has_value() && col_cache->numel() > 0)
+                   ? *col_cache
+                   : (is_1x1_s1(sh) ? x.view({sh.M(), (long)sh.C})
+                                    : im2col(x, sh));
+    gemm_bf16_raw(dy2, col, dw, sh.Kout, sh.RSC(), sh.M(), true, false,
+                  nullptr, false, EpStore::kPlain, 0);
+  }
 
   auto db = colsum_bf16(dy2);
   return {dx, dw.view(w.sizes()), db};

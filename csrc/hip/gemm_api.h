@@ -52,6 +52,11 @@ bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
 bool gemm_conv_dgrad_raw(const torch::Tensor& dy, const torch::Tensor& wrot2,
                          torch::Tensor& dx, const ConvShape& sh);
 
+// Implicit-GEMM NHWC conv weight-gradient: dW[Kout, RSC] = dy2^T @
+// implicit-col(x) — no col matrix. Requires sh.C % 8 == 0.
+bool gemm_conv_wgrad_raw(const torch::Tensor& dy2, const torch::Tensor& x,
+                         torch::Tensor& dw, const ConvShape& sh);
+
 // colsum: out[n] = sum_m X[m,n]  (bias gradient)
 torch::Tensor colsum_bf16(const torch::Tensor& X);
 

@@ -184,7 +184,38 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
       }
     }
     // ---- stage B tile ----
-    if (vecB) {
+    if (CMODE == 3) {
+      // wgrad implicit col: B[k'=m][n'=(r,s,c)]; a granule is 8
+      // consecutive c at fixed (m, r, s) -> one contiguous x load,
+      // scatter-written like the vecB !TB path (col never exists)
+      constexpr int GROUPS = (BN * BK) / 8;
+#pragma unroll
+      for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
+        const int g = tid + i * THREADS;
+        if (GROUPS % THREADS != 0 && g >= GROUPS) break;
+        const int k = g / (BN / 8), n8 = (g % (BN / 8)) * 8;
+        const long gn = tile_n + n8;   // (r, s, c8) flat index
+        const long gm = k0 + k;        // m = (n, oh, ow)
+        bool ok = gm < k_end && gn + 8 <= N;
+        u16x8_t v = {};
+        if (ok) {
+          const int rs = (int)(gn / csh.C);
+          const int c8 = (int)(gn - (long)rs * csh.C);
+          const int rr = rs / csh.S, ss = rs - rr * csh.S;
+          const int ow = (int)(gm % csh.OW);
+          const int oh = (int)((gm / csh.OW) % csh.OH);
+          const int nn = (int)(gm / ((long)csh.OW * csh.OH));
+          const int ih = oh * csh.stride - csh.pad + rr;
+          const int iw = ow * csh.stride - csh.pad + ss;
+          if (ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W)
+            v = *reinterpret_cast<const u16x8_t*>(
+                &B[(((long)nn * csh.H + ih) * csh.W + iw) * csh.C + c8]);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *reinterpret_cast<unsigned short*>(&Bs[n8 + j][k]) = v[j];
+      }
+    } else if (vecB) {
       constexpr int GROUPS = (BN * BK) / 8;
 #pragma unroll
       for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
@@ -1410,6 +1441,69 @@ bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
 bool gemm_conv_dgrad_raw(const torch::Tensor& dy, const torch::Tensor& wrot2,
                          torch::Tensor& dx, const ConvShape& sh) {
   return conv_implicit_gemm<2>(dy, wrot2, dx, sh, nullptr, false);
+}
+
+// wgrad: dW[Kout, RSC] = dy2^T @ implicit-col(x). A = dy2 [M, Kout]
+// through the TA scatter staging, B = x through the CMODE-3 gather —
+// the col matrix never exists. Always the small-tile split-K shape
+// (M' = Kout is small, K' = M is huge).
+bool gemm_conv_wgrad_raw(const torch::Tensor& dy2, const torch::Tensor& x,
+                         torch::Tensor& dw, const ConvShape& sh) {
+  if (sh.C % 8 != 0) return false;
+  const long M = sh.Kout, N = sh.RSC(), K = sh.M();
+  const bf16* a = (const bf16*)dy2.data_ptr();
+  const bf16* b = (const bf16*)x.data_ptr();
+  bf16* c = (bf16*)dw.data_ptr();
+  const TileCfg t = pick_tile(M, N);
+  const long tiles = ((M + t.bm - 1) / t.bm) * ((N + t.bn - 1) / t.bn);
+  const long ksteps = (K + BK - 1) / BK;
+  long S = pick_splitk(M, N, K, tiles, ksteps, 512, 150.0e12);
+  const long kslice = ((ksteps + S - 1) / S) * BK;
+  S = (K + kslice - 1) / kslice;
+  torch::Tensor part;
+  float* part_ptr = nullptr;
+  if (S > 1) {
+    part = torch::empty({S, M, N}, x.options().dtype(at::kFloat));
+    part_ptr = part.data_ptr<float>();
+  }
+  dim3 grid((unsigned)tiles, (unsigned)S);
+  auto launchw = [&](auto bm, auto bn, auto wr, auto wc) {
+    constexpr int BMv = decltype(bm)::value, BNv = decltype(bn)::value;
+    constexpr int WRv = decltype(wr)::value, WCv = decltype(wc)::value;
+    dim3 block(WRv * WCv * 64);
+    hipLaunchKernelGGL(
+        (gemm_kernel<BMv, BNv, WRv, WCv, true, false, 3>), grid, block, 0,
+        cur_stream(), a, b, c, part_ptr, nullptr, M, N, K, kslice, 0,
+        (int)EpStore::kPlain, 0, 1, 0, sh);
+  };
+  using c32 = std::integral_constant<int, 32>;
+  using c64 = std::integral_constant<int, 64>;
+  using c128 = std::integral_constant<int, 128>;
+  using c1 = std::integral_constant<int, 1>;
+  using c2 = std::integral_constant<int, 2>;
+  using c4 = std::integral_constant<int, 4>;
+  switch (t.bm * 1000 + t.bn) {
+    case 128128: launchw(c128{}, c128{}, c2{}, c2{}); break;
+    case 128064: launchw(c128{}, c64{}, c2{}, c2{}); break;
+    case 128032: launchw(c128{}, c32{}, c2{}, c2{}); break;
+    case  64128: launchw(c64{}, c128{}, c2{}, c2{}); break;
+    case  64064: launchw(c64{}, c64{}, c2{}, c2{}); break;
+    case  64032: launchw(c64{}, c32{}, c2{}, c2{}); break;
+    case  32128: launchw(c32{}, c128{}, c1{}, c4{}); break;
+    case  32064: launchw(c32{}, c64{}, c1{}, c4{}); break;
+    case  32032: launchw(c32{}, c32{}, c2{}, c1{}); break;
+    default: return false;
+  }
+  HIP_CHECK(hipGetLastError());
+  if (S > 1) {
+    const long total = M * N;
+    int blocks = (int)std::min<long>((total + 31) / 32, 16384);
+    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
+                       cur_stream(), part_ptr, (int)S, M, N, c, nullptr, 0,
+                       (int)EpStore::kPlain, 0);
+    HIP_CHECK(hipGetLastError());
+  }
+  return true;
 }
 
 torch::Tensor gemm_raw(torch::Tensor A, torch::Tensor B, bool ta, bool tb) {
