@@ -91,7 +91,7 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
   const bf16 zero = f2b(0.f);
   for (long k0 = k_begin; k0 < k_end; k0 += BK) {
     // ---- stage A tile (16-byte vector path when layout permits) ----
-    if (CMODE != 0) {  // implicit NHWC conv gather (granule % 8)
+    if (CMODE == 1 || CMODE == 2) {  // implicit A gather (x / dy)
       constexpr int GROUPS = (BM * BK) / 8;
 #pragma unroll
       for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
