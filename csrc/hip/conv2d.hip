@@ -426,7 +426,12 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   // (reused from the fwd when it produced one).
   auto dw = torch::empty_like(w2);
   bool wgrad_done = false;
-  if (!is_1x1_s1(sh) && !(col_cache.has_value() && col_cache->numel() > 0))
+  // implicit only when the col round trip would dominate: the wgrad
+  // GEMM does Kout flops per col element, so small Kout => col-bound
+  // (measured: implicit at Kout>=128 ran 150us/call, slower than the
+  // materialized dbuf path it replaced)
+  if (!is_1x1_s1(sh) && sh.Kout <= 64 &&
+      !(col_cache.has_value() && col_cache->numel() > 0))
     wgrad_done = gemm_conv_wgrad_raw(dy2, x, dw, sh);
   if (!wgrad_done) {
     auto col = (col_cache.has_value() && col_cache->numel() > 0)
