@@ -316,12 +316,19 @@ __global__ void bn_norm_vec_kernel(const bf16* __restrict__ x,
     const int c8 = (int)(g % c8g) * 8;
     const long i = (g / c8g) * C + c8;
     const bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(&x[i]);
+    // vector param loads: 6 wide loads instead of 32 scalar ones
+    const float4 m0 = *reinterpret_cast<const float4*>(&mean[c8]);
+    const float4 m1 = *reinterpret_cast<const float4*>(&mean[c8 + 4]);
+    const float4 i0 = *reinterpret_cast<const float4*>(&invstd[c8]);
+    const float4 i1 = *reinterpret_cast<const float4*>(&invstd[c8 + 4]);
+    const bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(&gamma[c8]);
+    const bf16x8_t bv = *reinterpret_cast<const bf16x8_t*>(&beta[c8]);
+    const float mj[8] = {m0.x, m0.y, m0.z, m0.w, m1.x, m1.y, m1.z, m1.w};
+    const float ij[8] = {i0.x, i0.y, i0.z, i0.w, i1.x, i1.y, i1.z, i1.w};
     bf16x8_t out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int c = c8 + j;
-      float f = (b2f(v[j]) - mean[c]) * invstd[c] * b2f(gamma[c]) +
-                b2f(beta[c]);
+      float f = (b2f(v[j]) - mj[j]) * ij[j] * b2f(gv[j]) + b2f(bv[j]);
       if (relu) f = fmaxf(f, 0.f);
       out[j] = f2b(f);
     }
@@ -445,13 +452,25 @@ __global__ void bn_bwd_dx_vec_kernel(const bf16* __restrict__ x,
       for (int j = 0; j < 8; ++j)
         if (!(b2f(yv[j]) > 0.f)) gv[j] = (__bf16)0.f;
     }
+    const float4 m0 = *reinterpret_cast<const float4*>(&mean[c8]);
+    const float4 m1 = *reinterpret_cast<const float4*>(&mean[c8 + 4]);
+    const float4 i0 = *reinterpret_cast<const float4*>(&invstd[c8]);
+    const float4 i1 = *reinterpret_cast<const float4*>(&invstd[c8 + 4]);
+    const float4 s0 = *reinterpret_cast<const float4*>(&sdy[c8]);
+    const float4 s1 = *reinterpret_cast<const float4*>(&sdy[c8 + 4]);
+    const float4 q0 = *reinterpret_cast<const float4*>(&sdyx[c8]);
+    const float4 q1 = *reinterpret_cast<const float4*>(&sdyx[c8 + 4]);
+    const bf16x8_t ga = *reinterpret_cast<const bf16x8_t*>(&gamma[c8]);
+    const float mj[8] = {m0.x, m0.y, m0.z, m0.w, m1.x, m1.y, m1.z, m1.w};
+    const float ij[8] = {i0.x, i0.y, i0.z, i0.w, i1.x, i1.y, i1.z, i1.w};
+    const float sj[8] = {s0.x, s0.y, s0.z, s0.w, s1.x, s1.y, s1.z, s1.w};
+    const float qj[8] = {q0.x, q0.y, q0.z, q0.w, q1.x, q1.y, q1.z, q1.w};
     bf16x8_t out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int c = c8 + j;
-      const float xhat = (b2f(xv[j]) - mean[c]) * invstd[c];
-      out[j] = f2b(b2f(gamma[c]) * invstd[c] *
-                   (b2f(gv[j]) - sdy[c] / count - xhat * sdyx[c] / count));
+      const float xhat = (b2f(xv[j]) - mj[j]) * ij[j];
+      out[j] = f2b(b2f(ga[j]) * ij[j] *
+                   (b2f(gv[j]) - sj[j] / count - xhat * qj[j] / count));
     }
     *reinterpret_cast<bf16x8_t*>(&dx[i]) = out;
   }
@@ -584,7 +603,7 @@ inline long pick_chunk_rows(long M, int cblocks) {
   // <= 256 chunks keeps the final pass small; >= 768 total blocks
   // fills the chip when the channel dimension alone cannot
   const long target =
-      std::max<long>(1, std::min<long>(768 / std::max(cblocks, 1), 256));
+      std::max<long>(1, std::min<long>(2048 / std::max(cblocks, 1), 512));
   return std::max<long>(64, (M + target - 1) / target);
 }
 
