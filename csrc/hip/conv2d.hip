@@ -14,10 +14,18 @@
 // ResNet-20/50) demand Conv2D fwd/bwd as hand-written CDNA4 kernels.
 //
 // Layouts: x [N, H, W, C], w [Kout, R, S, C], y [N, OH, OW, Kout].
-// GEMM views: fwd   y2[M,Kout] = col[M,RSC] @ w2[Kout,RSC]^T
-//             dgrad dcol[M,RSC] = dy2[M,Kout] @ w2, dx = col2im(dcol)
-//             wgrad dw[Kout,RSC] = dy2^T @ col   (col cached from fwd)
-// where M = N*OH*OW, RSC = R*S*C, k = (r*S + s)*C + c.
+// GEMM views (M = N*OH*OW, RSC = R*S*C, k = (r*S + s)*C + c):
+//   fwd   y2[M,Kout] = col[M,RSC] @ w2[Kout,RSC]^T — implicit GEMM
+//         (the col gather runs inside the GEMM's A staging) whenever
+//         C % 8 == 0; 1x1 stride-1 convs feed x [M, C] directly.
+//   dgrad dx[M',C] = dy-gather @ w.permute(3,1,2,0) implicit when
+//         stride == 1 and 8 <= C <= 64 (small C = col-traffic-bound);
+//         else dcol = dy2 @ w2^T then vectorized col2im.
+//   wgrad dw[Kout,RSC] = dy2^T @ implicit-col(x) when Kout <= 64;
+//         else a materialized col (reused from fwd when available).
+// The gates are measured crossovers: implicit wins where the col
+// matrix round trip (R*S times the activation bytes) dominates, the
+// materialized dbuf/8-phase GEMM wins where MFMA work dominates.
 
 #include "common.h"
 #include "gemm_api.h"

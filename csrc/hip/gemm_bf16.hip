@@ -1,28 +1,38 @@
-// MFMA-tiled bf16 GEMM for gfx950 (CDNA4) + linear fwd/bwd entry points.
+// MFMA bf16 GEMM family for gfx950 (CDNA4) + linear fwd/bwd entries.
 //
 // Owns the reference matmul ops (main.py:120 forward, 127-130 backward
 // via compute_gradients) for every Dense/FC layer, and is the GEMM core
-// behind im2col convolution (conv_im2col.hip).
+// behind the NHWC convolutions (conv2d.hip), including their
+// implicit-GEMM forms.
 //
-// Design (cdna_hip_programming.md §5 canonical CDNA GEMM):
-//   - __builtin_amdgcn_mfma_f32_16x16x32_bf16: per-wave 16x16 tile,
-//     K=32 per instruction, fp32 accumulate in AGPRs.
-//   - LDS staging: A tile [BM][BK], B tile [BN][BK] both K-contiguous so
-//     every fragment load is one 16-byte ds_read (b128); rows padded
-//     +8 bf16 (16 B) against bank conflicts.
-//   - 64-wide wavefronts; template wave grid WRxWC, each wave computes a
-//     (BM/WR)x(BN/WC) sub-tile as 16x16 fragments.
-//   - Transposed A/B operands are handled at the staging gather, so the
-//     MFMA inner loop is layout-independent.
-//   - Shape-aware tile selection (conv GEMMs are tall/skinny: M up to
-//     N*OH*OW ~ 1M rows with N as small as 9..64 columns).
-//   - Split-K with a FIXED-ORDER fp32 reduce for small-tile/huge-K
-//     shapes (conv wgrad: M=Kout, N=CRS, K=N*OH*OW): K slices land in a
-//     partial buffer [S, M, N] and a deterministic second kernel sums
-//     ascending s — bitwise identical on every rank, no atomics.
-//   - Epilogue (bias + optional ReLU + optional NCHW scatter) fused
-//     into the GEMM (or into the split-K reduce).
-//   - 1-D tile grid with an XCD-bijective swizzle (guide T1).
+// Three kernel tiers, all __builtin_amdgcn_mfma_f32_16x16x32_bf16 with
+// fp32 accumulation, picked per shape by a cost model:
+//   1. gemm8p_kernel — the 8-phase global_load_lds schedule
+//      (cdna_hip_programming.md §5 "8-phase template"): counted vmcnt
+//      keeps prefetched half-tiles in flight across raw barriers, one
+//      barrier per phase, setprio around each 16-MFMA cluster, and a
+//      conflict-free XOR source/read swizzle (bits 8,10 -> 5,6).
+//      980 TF @4096^3 / 1114 @8192^3 on random data. Fully aligned
+//      shapes (M,N % 256, K % 64) only — glds cannot zero-fill edges.
+//   2. gemm256_kernel<BM, BN, CMODE> — double-buffered register-staged
+//      BMxBN tile family (guide T14: write tile t+1 after the barrier,
+//      re-issue t+2 loads immediately, one barrier per K-step), 8
+//      waves, M/K edges zero-guarded. CMODE 1/2 fuse the NHWC conv
+//      forward / data-grad gathers into the A staging.
+//   3. gemm_kernel<..., CMODE> — synchronous small-tile kernel for
+//      skinny shapes; CMODE 1 = implicit conv fwd (narrow Kout),
+//      CMODE 3 = implicit wgrad (x gathered into the B staging).
+// Shared machinery:
+//   - operands not in the vector-staging layout (A [M][K] / B [N][K])
+//     are pre-transposed by the LDS-tiled transpose (scatter staging
+//     measured 3.4x slower than all-vector at 4096^3);
+//   - split-K with a FIXED-ORDER fp32 reduce: K slices land in a
+//     partial buffer [S, M, N], a deterministic 8-lane-tree kernel sums
+//     ascending s — bitwise identical on every rank, no atomics; the
+//     slice count comes from a GEMM-fill vs reduce-traffic cost model;
+//   - fused bias/ReLU epilogues; 1-D tile grid with an XCD-bijective
+//     swizzle (guide T1); hierarchical fixed-order colsum for bias
+//     gradients.
 //
 // Fragment layout (gfx950 mfma_f32_16x16x32_bf16, cdna4_isa.md §10),
 // verified on MI355X hardware by tests/test_ops_gpu.py (asymmetric
