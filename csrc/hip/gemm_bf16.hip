@@ -1099,7 +1099,8 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
       // rough per-tile efficiency from the measured ladder; narrow
       // tiles are staging-heavier
       const double bn_pen = bn2 >= 128 ? 1.0 : (bn2 == 64 ? 0.75 : 0.5);
-      for (int bm : {256, 128, 64}) {
+      for (int bm : (bn2 == 64 ? std::initializer_list<int>{128, 64}
+                               : std::initializer_list<int>{256, 128, 64})) {
         const double tf = (bm == 256 ? 800.0e12
                                      : (bm == 128 ? 600.0e12 : 320.0e12)) *
                           bn_pen;
@@ -1385,7 +1386,8 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
   {
     double best_t = 1e30;
     const double bn_pen = bn2 >= 128 ? 1.0 : 0.75;
-    for (int bm : {256, 128, 64}) {
+    for (int bm : (bn2 == 64 ? std::initializer_list<int>{128, 64}
+                               : std::initializer_list<int>{256, 128, 64})) {
       const double tf = (bm == 256 ? 800.0e12
                                    : (bm == 128 ? 600.0e12 : 320.0e12)) *
                         bn_pen;
