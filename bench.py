@@ -35,6 +35,10 @@ def main() -> None:
     ap.add_argument("--byzantine", type=int, default=0,
                     help="label-flip attacker clients (BASELINE config 4; "
                          "committee scoring is the defense)")
+    ap.add_argument("--optimizer", default="sgd", choices=["sgd", "adam"],
+                    help="local optimizer (reference main.py:126-130)")
+    ap.add_argument("--metrics", default=None,
+                    help="write per-round JSONL records to this path")
     ap.add_argument("--eval", action="store_true", help="eval every round")
     ap.add_argument("--phases", action="store_true",
                     help="print per-phase timings of each timed round")
@@ -67,11 +71,12 @@ def main() -> None:
         samples_per_client=spc,
         batch_size=bs, partition="dirichlet",
         dirichlet_alpha=0.3, eval_samples=min(4096, 2 * spc),
-        learning_rate=0.01, byzantine_clients=args.byzantine)
+        learning_rate=0.01, byzantine_clients=args.byzantine,
+        optimizer=args.optimizer)
 
     t = Transport()
     shards, test = make_federated(cfg)
-    eng = FLEngine(cfg, t, shards, test)
+    eng = FLEngine(cfg, t, shards, test, metrics_path=args.metrics)
 
     use_cuda = t.device.type == "cuda"
 
