@@ -1,0 +1,44 @@
+"""Engine-level GPU tests: long-run stability + checkpoint/resume
+(the chain's persistence role) and accuracy progress on
+the flagship config."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = torch.device("cuda", 0)
+
+
+def _engine(cfg):
+    from bflc_amd.comm import Transport
+    from bflc_amd.data import make_federated
+    from bflc_amd.fl import FLEngine
+    shards, test = make_federated(cfg)
+    return FLEngine(cfg, Transport(device=DEV), shards, test)
+
+
+def test_long_run_checkpoint_resume(tmp_path):
+    from bflc_amd.config import FLConfig
+    cfg = FLConfig.for_world(1, model="femnist_cnn", n_class=62,
+                             samples_per_client=1024, batch_size=512,
+                             eval_samples=1024, partition="dirichlet",
+                             learning_rate=0.01)
+    eng = _engine(cfg)
+    eng.run(200)
+    acc_mid = eng.evaluate_global()
+    p = str(tmp_path / "ck.pt")
+    eng.save(p)
+    flat_at_save = eng.global_flat.clone()
+    eng.run(25)
+
+    # fresh engine, restore, must pick up bitwise where the save was
+    eng2 = _engine(cfg)
+    eng2.load(p)
+    assert torch.equal(eng2.global_flat, flat_at_save)
+    assert eng2.ledger.epoch == 200
+    assert abs(eng2.evaluate_global() - acc_mid) < 1e-6
+    eng2.run(25)
+    assert eng2.ledger.epoch == 225
+    # the flagship config LEARNS: synthetic FEMNIST at 200+ rounds is
+    # far above the 62-class chance rate (~0.016)
+    assert acc_mid > 0.10
