@@ -94,26 +94,28 @@ class ResNet(FlatModel):
         g, b = self.p(f"{bnname}.g"), self.p(f"{bnname}.b")
         return O.batchnorm2d(h, g, b, relu=relu)  # fused BN(+relu)
 
-    def _basic_block(self, x, p, stride):
-        h = self._cbr(x, f"{p}.c1", f"{p}.bn1", stride, 1, relu=True)
-        h = self._cbr(h, f"{p}.c2", f"{p}.bn2", 1, 1, relu=False)
-        sc = x
+    def _shortcut(self, x, p, stride):
         if f"{p}.sc.w" in self._offsets:
             sc = O.conv2d(x, self.p(f"{p}.sc.w"), None, stride, 0)
-            sc = O.batchnorm2d(sc, self.p(f"{p}.scbn.g"),
-                               self.p(f"{p}.scbn.b"))
-        return O.add_relu(h, sc)
+            return O.batchnorm2d(sc, self.p(f"{p}.scbn.g"),
+                                 self.p(f"{p}.scbn.b"))
+        return x
+
+    def _basic_block(self, x, p, stride):
+        h = self._cbr(x, f"{p}.c1", f"{p}.bn1", stride, 1, relu=True)
+        h = O.conv2d(h, self.p(f"{p}.c2.w"), None, 1, 1)
+        sc = self._shortcut(x, p, stride)
+        # final BN fuses the residual add + relu (one pass)
+        return O.batchnorm2d(h, self.p(f"{p}.bn2.g"), self.p(f"{p}.bn2.b"),
+                             relu=True, residual=sc)
 
     def _bottleneck_block(self, x, p, stride):
         h = self._cbr(x, f"{p}.c1", f"{p}.bn1", 1, 0, relu=True)
         h = self._cbr(h, f"{p}.c2", f"{p}.bn2", stride, 1, relu=True)
-        h = self._cbr(h, f"{p}.c3", f"{p}.bn3", 1, 0, relu=False)
-        sc = x
-        if f"{p}.sc.w" in self._offsets:
-            sc = O.conv2d(x, self.p(f"{p}.sc.w"), None, stride, 0)
-            sc = O.batchnorm2d(sc, self.p(f"{p}.scbn.g"),
-                               self.p(f"{p}.scbn.b"))
-        return O.add_relu(h, sc)
+        h = O.conv2d(h, self.p(f"{p}.c3.w"), None, 1, 0)
+        sc = self._shortcut(x, p, stride)
+        return O.batchnorm2d(h, self.p(f"{p}.bn3.g"), self.p(f"{p}.bn3.b"),
+                             relu=True, residual=sc)
 
     def forward(self, x):
         if self.bottleneck:

@@ -238,20 +238,23 @@ class _BatchNormFn(torch.autograd.Function):
     x viewed [N*H*W, C] (csrc/hip/batchnorm.hip); CPU: fp32 oracle."""
 
     @staticmethod
-    def forward(ctx, x, gamma, beta, eps: float, relu: bool):
+    def forward(ctx, x, gamma, beta, eps: float, relu: bool, residual):
         if x.is_cuda:
             y, mean, invstd = hip_ops().batchnorm_fwd(x, gamma, beta, eps,
-                                                      relu)
+                                                      relu, residual)
         else:
             xf = x.float()
             mean = xf.mean(dim=(0, 1, 2))
             var = xf.var(dim=(0, 1, 2), unbiased=False)
             invstd = (var + eps).rsqrt()
-            y = ((xf - mean) * invstd * gamma.float()
-                 + beta.float()).to(x.dtype)
+            y = (xf - mean) * invstd * gamma.float() + beta.float()
+            if residual is not None:
+                y = y + residual.float()
             if relu:
                 y = torch.relu(y)
+            y = y.to(x.dtype)
         ctx.relu = relu
+        ctx.has_res = residual is not None
         if relu:
             ctx.save_for_backward(x, gamma, mean, invstd, y)
         else:
@@ -263,13 +266,20 @@ class _BatchNormFn(torch.autograd.Function):
         x, gamma, mean, invstd = ctx.saved_tensors[:4]
         yr = ctx.saved_tensors[4] if ctx.relu else None
         dy = dy.contiguous()
+        dres = None
         if x.is_cuda:
+            if ctx.has_res:
+                # residual grad = relu-masked dy (same mask BN bwd uses)
+                dres = (hip_ops().add_relu_bwd(yr, dy) if yr is not None
+                        else dy)
             dx, dgamma, dbeta = hip_ops().batchnorm_bwd(x, dy, mean, invstd,
                                                         gamma, yr)
         else:
             xf, dyf = x.float(), dy.float()
             if yr is not None:
                 dyf = dyf * (yr.float() > 0)
+            if ctx.has_res:
+                dres = dyf.to(x.dtype)
             n = x.numel() / x.shape[-1]
             xhat = (xf - mean) * invstd
             sdy = dyf.sum(dim=(0, 1, 2))
@@ -278,14 +288,15 @@ class _BatchNormFn(torch.autograd.Function):
             dx = dx.to(x.dtype)
             dgamma = sdyx.to(x.dtype)
             dbeta = sdy.to(x.dtype)
-        return dx, dgamma, dbeta, None, None
+        return dx, dgamma, dbeta, None, None, dres
 
 
-def batchnorm2d(x, gamma, beta, eps: float = 1e-5,
-                relu: bool = False) -> torch.Tensor:
-    """Batch-stats BN with optionally FUSED relu (one kernel fwd, the
-    relu mask folded into the backward reduction kernels)."""
-    return _BatchNormFn.apply(x, gamma, beta, eps, relu)
+def batchnorm2d(x, gamma, beta, eps: float = 1e-5, relu: bool = False,
+                residual=None) -> torch.Tensor:
+    """Batch-stats BN with optionally FUSED residual add + relu (one
+    kernel fwd; the relu mask folds into the backward reductions and
+    the residual grad is the masked dy)."""
+    return _BatchNormFn.apply(x, gamma, beta, eps, relu, residual)
 
 
 class _GlobalAvgPoolFn(torch.autograd.Function):

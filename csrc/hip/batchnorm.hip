@@ -306,7 +306,8 @@ __global__ void bn_norm_vec_kernel(const bf16* __restrict__ x,
                                    const float* __restrict__ mean,
                                    const float* __restrict__ invstd,
                                    const bf16* __restrict__ gamma,
-                                   const bf16* __restrict__ beta, long M,
+                                   const bf16* __restrict__ beta,
+                                   const bf16* __restrict__ res, long M,
                                    int C, int relu, bf16* __restrict__ y) {
   const int c8g = C / 8;
   const long total_g = M * c8g;
@@ -325,10 +326,13 @@ __global__ void bn_norm_vec_kernel(const bf16* __restrict__ x,
     const bf16x8_t bv = *reinterpret_cast<const bf16x8_t*>(&beta[c8]);
     const float mj[8] = {m0.x, m0.y, m0.z, m0.w, m1.x, m1.y, m1.z, m1.w};
     const float ij[8] = {i0.x, i0.y, i0.z, i0.w, i1.x, i1.y, i1.z, i1.w};
+    bf16x8_t rv;
+    if (res) rv = *reinterpret_cast<const bf16x8_t*>(&res[i]);
     bf16x8_t out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float f = (b2f(v[j]) - mj[j]) * ij[j] * b2f(gv[j]) + b2f(bv[j]);
+      if (res) f += b2f(rv[j]);
       if (relu) f = fmaxf(f, 0.f);
       out[j] = f2b(f);
     }
@@ -340,7 +344,8 @@ __global__ void bn_norm_kernel(const bf16* __restrict__ x,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
                                const bf16* __restrict__ gamma,
-                               const bf16* __restrict__ beta, long M, int C,
+                               const bf16* __restrict__ beta,
+                               const bf16* __restrict__ res, long M, int C,
                                int relu, bf16* __restrict__ y) {
   const long total = M * C;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -349,6 +354,7 @@ __global__ void bn_norm_kernel(const bf16* __restrict__ x,
     const int c = (int)(i % C);
     float f = (b2f(x[i]) - mean[c]) * invstd[c] * b2f(gamma[c]) +
               b2f(beta[c]);
+    if (res) f += b2f(res[i]);
     if (relu) f = fmaxf(f, 0.f);
     y[i] = f2b(f);
   }
@@ -611,8 +617,10 @@ inline long pick_chunk_rows(long M, int cblocks) {
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     torch::Tensor x, torch::Tensor gamma, torch::Tensor beta, double eps,
-    bool relu) {
+    bool relu, c10::optional<torch::Tensor> residual) {
   CHECK_GPU(x); CHECK_CONTIG(x);
+  const bf16* resp =
+      residual.has_value() ? (const bf16*)residual->data_ptr() : nullptr;
   const int C = (int)x.size(-1);
   const long M = x.numel() / C;
   const int G = (C % 8 == 0) ? granule_lanes(C) : 0;
@@ -660,14 +668,14 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
                        dim3(1024), 0, cur_stream(),
                        (const bf16*)x.data_ptr(), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), (const bf16*)gc.data_ptr(),
-                       (const bf16*)bc.data_ptr(), M, C, relu ? 1 : 0,
+                       (const bf16*)bc.data_ptr(), resp, M, C, relu ? 1 : 0,
                        (bf16*)y.data_ptr());
   else
     hipLaunchKernelGGL(bn_norm_kernel, dim3(ew_grid(x.numel())), dim3(1024),
                        0, cur_stream(), (const bf16*)x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        (const bf16*)gc.data_ptr(), (const bf16*)bc.data_ptr(),
-                       M, C, relu ? 1 : 0, (bf16*)y.data_ptr());
+                       resp, M, C, relu ? 1 : 0, (bf16*)y.data_ptr());
   HIP_CHECK(hipGetLastError());
   return {y, mean, invstd};
 }

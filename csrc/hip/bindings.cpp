@@ -55,7 +55,7 @@ torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx,
 // batchnorm.hip
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     torch::Tensor x, torch::Tensor gamma, torch::Tensor beta, double eps,
-    bool relu);
+    bool relu, c10::optional<torch::Tensor> residual);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
     torch::Tensor invstd, torch::Tensor gamma,
@@ -95,6 +95,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2d_fwd", &bflc::maxpool2d_fwd);
   m.def("maxpool2d_bwd", &bflc::maxpool2d_bwd);
   m.def("batchnorm_fwd", &bflc::batchnorm_fwd,
+        py::arg("x"), py::arg("gamma"), py::arg("beta"), py::arg("eps"),
+        py::arg("relu"), py::arg("residual") = py::none(),
         "(y, mean, invstd) — batch-stats BN, optional fused relu");
   m.def("batchnorm_bwd", &bflc::batchnorm_bwd, "(dx, dgamma, dbeta)",
         py::arg("x"), py::arg("dy"), py::arg("mean"), py::arg("invstd"),

@@ -187,10 +187,27 @@ class CommitteeLedger {
     update_count_ = 0;
     score_count_ = 0;
     // Role rotation (.cpp:443-455): all comm -> trainer, then the top
-    // comm_count scored trainers -> comm.
+    // comm_count scored trainers -> comm. Divergence from the
+    // reference: if fewer trainers were scored than comm_count, the
+    // reference's committee SHRINKS while the score_count==COMM_COUNT
+    // aggregation trigger stays — a permanent deadlock (latent, masked
+    // there by 16 trainers racing for 10 slots). Here the remaining
+    // seats are refilled deterministically in registration order, so
+    // the protocol is live for any admitted-update count >= 1.
     for (auto& kv : roles_)
       if (kv.second == "comm") kv.second = "trainer";
-    for (const auto& id : pending_->next_committee) roles_[id] = "comm";
+    int seated = 0;
+    for (const auto& id : pending_->next_committee) {
+      roles_[id] = "comm";
+      ++seated;
+    }
+    for (const auto& id : reg_order_) {
+      if (seated >= cfg_.comm_count) break;
+      if (roles_[id] != "comm") {
+        roles_[id] = "comm";
+        ++seated;
+      }
+    }
     pending_.reset();
   }
 

@@ -85,10 +85,17 @@ class PyLedgerModel:
                                                   + vals[k // 2])
             meds[tr] = m
         ranked = sorted(meds.items(), key=lambda kv: (-kv[1], kv[0]))
-        new_comm = {o for o, _ in ranked[: self.cfg.comm_count]}
+        new_comm = [o for o, _ in ranked[: self.cfg.comm_count]]
+        # refill short committees in registration order (ledger
+        # anti-deadlock divergence, csrc/ledger.cpp commit_aggregate)
+        for o in self.registered:
+            if len(new_comm) >= self.cfg.comm_count:
+                break
+            if o not in new_comm:
+                new_comm.append(o)
         if self.cfg.client_num > 1:
             for o in self.roles:
-                self.roles[o] = "comm" if o in new_comm else "trainer"
+                self.roles[o] = ("comm" if o in new_comm else "trainer")
         self.updates.clear()
         self.scores.clear()
         self.epoch += 1
