@@ -1,3 +1,5 @@
+"""Per-round wall clock + allocator stats (used to isolate a ~35 ms
+CPython gen-2 GC pause; see bench.py's gc pinning)."""
 import sys, torch
 sys.path.insert(0, "/root/repo")
 from bflc_amd.config import FLConfig
