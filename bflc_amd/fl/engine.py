@@ -163,13 +163,15 @@ class FLEngine:
         candidate = W0 - lr*delta, scored by accuracy on the scorer's own
         local shard."""
         shard = self.shards[scorer]
-        out: Dict[str, float] = {}
+        accs: List[Tuple[str, torch.Tensor]] = []
         for origin, delta in updates:
             cand = self.global_flat.clone()
             O.axpy_(cand, -self.cfg.learning_rate, delta)
             self.model.set_flat(cand)
-            out[origin] = self.model.accuracy(shard.x, shard.y)
-        return out
+            # device-resident: ONE host sync per scorer (below), not
+            # one per candidate
+            accs.append((origin, self.model.accuracy_t(shard.x, shard.y)))
+        return {origin: float(a) for origin, a in accs}
 
     # ------------------------------------------------------------------
     def run_round(self, eval_global: bool = False) -> RoundStats:

@@ -180,15 +180,23 @@ class FlatModel:
         return x.to(device=self.device, dtype=self.compute_dtype)
 
     @torch.no_grad()
-    def accuracy(self, x: torch.Tensor, y: torch.Tensor,
-                 batch: int = 4096) -> float:
-        """mean(argmax(pred)==y) over the set, batched (reference
-        main.py:172-193 local_testing / 285-306 global_testing)."""
-        total, correct = 0, 0.0
+    def accuracy_t(self, x: torch.Tensor, y: torch.Tensor,
+                   batch: int = 4096) -> torch.Tensor:
+        """Device-resident accuracy (no host sync): callers batching
+        several evaluations sync once at the end."""
+        total = 0
+        correct = torch.zeros((), device=self.device)
         for i in range(0, x.shape[0], batch):
             xb = self._cast(x[i:i + batch])
             yb = y[i:i + batch].to(self.device)
             logits = self.forward(xb)
-            correct += O.accuracy(logits, yb) * xb.shape[0]
+            correct += O.accuracy_t(logits, yb) * xb.shape[0]
             total += xb.shape[0]
         return correct / max(total, 1)
+
+    @torch.no_grad()
+    def accuracy(self, x: torch.Tensor, y: torch.Tensor,
+                 batch: int = 4096) -> float:
+        """mean(argmax(pred)==y) over the set, batched (reference
+        main.py:172-193 local_testing / 285-306 global_testing)."""
+        return float(self.accuracy_t(x, y, batch))

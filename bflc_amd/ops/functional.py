@@ -352,12 +352,17 @@ def add_relu(a, b) -> torch.Tensor:
 # non-autograd FL math (flat-tensor ops)
 # ---------------------------------------------------------------------------
 
-def accuracy(logits: torch.Tensor, target: torch.Tensor) -> float:
-    """mean(argmax(pred) == label) — reference main.py:182-183. Fused
-    argmax-compare-reduce kernel on GPU."""
+def accuracy_t(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """mean(argmax(pred) == label) as a DEVICE tensor (no host sync) —
+    fused argmax-compare-reduce kernel on GPU."""
     if logits.is_cuda:
-        return float(hip_ops().accuracy(logits, target))
-    return float((logits.argmax(dim=1) == target).float().mean())
+        return hip_ops().accuracy_t(logits, target)
+    return (logits.argmax(dim=1) == target).float().mean()
+
+
+def accuracy(logits: torch.Tensor, target: torch.Tensor) -> float:
+    """mean(argmax(pred) == label) — reference main.py:182-183."""
+    return float(accuracy_t(logits, target))
 
 
 def axpy_(y: torch.Tensor, alpha: float, x: torch.Tensor) -> torch.Tensor:

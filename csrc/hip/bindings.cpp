@@ -30,6 +30,7 @@ torch::Tensor softmax_ce_bwd(torch::Tensor probs, torch::Tensor target,
 
 // reduce.hip
 double accuracy(torch::Tensor logits, torch::Tensor target);
+torch::Tensor accuracy_t(torch::Tensor logits, torch::Tensor target);
 
 // gemm_bf16.hip
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b);
@@ -84,6 +85,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused softmax cross-entropy fwd (loss, probs)");
   m.def("softmax_ce_bwd", &bflc::softmax_ce_bwd);
   m.def("accuracy", &bflc::accuracy, "fused argmax-compare-reduce");
+  m.def("accuracy_t", &bflc::accuracy_t,
+        "accuracy as a device tensor (no host sync)");
   m.def("linear_fwd", &bflc::linear_fwd, "MFMA bf16 GEMM + bias");
   m.def("gemm_raw", &bflc::gemm_raw, "raw GEMM (bench/ablation)");
   m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");

@@ -43,7 +43,10 @@ __global__ void accuracy_kernel(const T* __restrict__ logits,
 
 }  // namespace
 
-double accuracy(torch::Tensor logits, torch::Tensor target) {
+// Device-resident variant: returns a 0-dim fp32 tensor (mean accuracy)
+// with NO host sync — batched evaluations (committee scoring) sync once
+// at the end.
+torch::Tensor accuracy_t(torch::Tensor logits, torch::Tensor target) {
   CHECK_GPU(logits); CHECK_CONTIG(logits);
   TORCH_CHECK(logits.dim() == 2);
   int M = (int)logits.size(0), C = (int)logits.size(1);
@@ -61,7 +64,11 @@ double accuracy(torch::Tensor logits, torch::Tensor target) {
                        correct.data_ptr<int>());
   }
   HIP_CHECK(hipGetLastError());
-  return correct.item<int>() / (double)M;
+  return correct.squeeze().to(at::kFloat) / (double)M;
+}
+
+double accuracy(torch::Tensor logits, torch::Tensor target) {
+  return accuracy_t(logits, target).item<float>();
 }
 
 }  // namespace bflc
