@@ -90,32 +90,33 @@ class ResNet(FlatModel):
 
     # ------------------------------------------------------------------
     def _cbr(self, x, conv, bnname, stride=1, pad=1, relu=True):
-        h = O.conv2d(x, self.p(f"{conv}.w"), None, stride, pad)
-        g, b = self.p(f"{bnname}.g"), self.p(f"{bnname}.b")
-        return O.batchnorm2d(h, g, b, relu=relu)  # fused BN(+relu)
+        # fused conv+BN(+relu): epilogue stats, single norm pass
+        return O.conv2d_bn(x, self.p(f"{conv}.w"), self.p(f"{bnname}.g"),
+                           self.p(f"{bnname}.b"), stride, pad, relu=relu)
 
     def _shortcut(self, x, p, stride):
         if f"{p}.sc.w" in self._offsets:
-            sc = O.conv2d(x, self.p(f"{p}.sc.w"), None, stride, 0)
-            return O.batchnorm2d(sc, self.p(f"{p}.scbn.g"),
-                                 self.p(f"{p}.scbn.b"))
+            return O.conv2d_bn(x, self.p(f"{p}.sc.w"),
+                               self.p(f"{p}.scbn.g"),
+                               self.p(f"{p}.scbn.b"), stride, 0,
+                               relu=False)
         return x
 
     def _basic_block(self, x, p, stride):
         h = self._cbr(x, f"{p}.c1", f"{p}.bn1", stride, 1, relu=True)
-        h = O.conv2d(h, self.p(f"{p}.c2.w"), None, 1, 1)
         sc = self._shortcut(x, p, stride)
-        # final BN fuses the residual add + relu (one pass)
-        return O.batchnorm2d(h, self.p(f"{p}.bn2.g"), self.p(f"{p}.bn2.b"),
-                             relu=True, residual=sc)
+        # final conv+BN fuses the residual add + relu as well
+        return O.conv2d_bn(h, self.p(f"{p}.c2.w"), self.p(f"{p}.bn2.g"),
+                           self.p(f"{p}.bn2.b"), 1, 1, relu=True,
+                           residual=sc)
 
     def _bottleneck_block(self, x, p, stride):
         h = self._cbr(x, f"{p}.c1", f"{p}.bn1", 1, 0, relu=True)
         h = self._cbr(h, f"{p}.c2", f"{p}.bn2", stride, 1, relu=True)
-        h = O.conv2d(h, self.p(f"{p}.c3.w"), None, 1, 0)
         sc = self._shortcut(x, p, stride)
-        return O.batchnorm2d(h, self.p(f"{p}.bn3.g"), self.p(f"{p}.bn3.b"),
-                             relu=True, residual=sc)
+        return O.conv2d_bn(h, self.p(f"{p}.c3.w"), self.p(f"{p}.bn3.g"),
+                           self.p(f"{p}.bn3.b"), 1, 0, relu=True,
+                           residual=sc)
 
     def forward(self, x):
         if self.bottleneck:

@@ -291,6 +291,62 @@ class _BatchNormFn(torch.autograd.Function):
         return dx, dgamma, dbeta, None, None, dres
 
 
+class _ConvBNFn(torch.autograd.Function):
+    """Fused conv + batch-stats BN (+relu, +residual) for the GPU path:
+    the conv GEMM's epilogue emits per-tile channel partials, one tiny
+    fixed-tree kernel finalizes mean/invstd, and a single normalization
+    pass produces the block output — the standalone BN stats sweep over
+    the conv activation disappears. Backward = BN backward (relu mask
+    folded into its reductions) then conv backward."""
+
+    @staticmethod
+    def forward(ctx, x, w, gamma, beta, stride, padding, eps, relu,
+                residual):
+        h = hip_ops()
+        ctx.stride, ctx.padding, ctx.relu = stride, padding, relu
+        ctx.has_res = residual is not None
+        yc, col, psum, psq = h.conv2d_fwd_bn(x, w, stride, padding)
+        M = yc.numel() // yc.shape[-1]
+        if psum.numel() > 0:
+            mean, invstd = h.bn_stats_finalize(psum, psq, float(M), eps)
+        else:  # split-K / unsupported shape: standalone stats pass
+            mean, invstd = h.bn_stats(yc, eps)
+        y = h.batchnorm_norm(yc, gamma, beta, mean, invstd, relu, residual)
+        saved = [x, w, col, yc, gamma, mean, invstd]
+        if relu:
+            saved.append(y)
+        ctx.save_for_backward(*saved)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        h = hip_ops()
+        x, w, col, yc, gamma, mean, invstd = ctx.saved_tensors[:7]
+        yr = ctx.saved_tensors[7] if ctx.relu else None
+        dy = dy.contiguous()
+        dres = None
+        if ctx.has_res:
+            dres = h.add_relu_bwd(yr, dy) if yr is not None else dy
+        dyc, dgamma, dbeta = h.batchnorm_bwd(yc, dy, mean, invstd, gamma,
+                                             yr)
+        dx, dw, _db = h.conv2d_bwd(x, w, dyc, ctx.stride, ctx.padding,
+                                   col if col.numel() > 0 else None)
+        return (dx, dw, dgamma, dbeta, None, None, None, None, dres)
+
+
+def conv2d_bn(x, w, gamma, beta, stride: int = 1, padding: int = 0,
+              eps: float = 1e-5, relu: bool = True,
+              residual=None) -> torch.Tensor:
+    """conv2d (no bias) -> batch-stats BN -> optional residual+relu,
+    fused on GPU (epilogue stats + single norm pass); on CPU the
+    composition of the fp32 oracle ops."""
+    if not x.is_cuda:
+        return batchnorm2d(conv2d(x, w, None, stride, padding), gamma,
+                           beta, eps, relu, residual)
+    return _ConvBNFn.apply(x, w, gamma, beta, stride, padding, eps, relu,
+                           residual)
+
+
 def batchnorm2d(x, gamma, beta, eps: float = 1e-5, relu: bool = False,
                 residual=None) -> torch.Tensor:
     """Batch-stats BN with optionally FUSED residual add + relu (one

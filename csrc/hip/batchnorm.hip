@@ -755,6 +755,116 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
   return {dx, dgamma, dbeta};
 }
 
+// Stats-only pass: (mean, invstd) of x viewed [M, C] (the standalone
+// path when the producing GEMM could not fuse the partials).
+std::tuple<torch::Tensor, torch::Tensor> bn_stats(torch::Tensor x,
+                                                  double eps) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  const int C = (int)x.size(-1);
+  const long M = x.numel() / C;
+  const int G = (C % 8 == 0) ? granule_lanes(C) : 0;
+  const int cblocks = (C % 8 == 0) ? (int)ceil_div(C / 8, G)
+                                   : (int)ceil_div(C, 32);
+  const long rows = pick_chunk_rows(M, cblocks);
+  const int chunks = (int)((M + rows - 1) / rows);
+  auto opts = x.options().dtype(at::kFloat);
+  auto psum = torch::empty({chunks, C}, opts);
+  auto psq = torch::empty({chunks, C}, opts);
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  if (C % 8 == 0) {
+    hipLaunchKernelGGL(bn_stats_part_vec_kernel, dim3(cblocks, chunks),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), M, C, G, rows,
+                       psum.data_ptr<float>(), psq.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(bn_stats_part_kernel, dim3(cblocks, chunks),
+                       dim3(256), 0, cur_stream(), (const bf16*)x.data_ptr(),
+                       M, C, rows, psum.data_ptr<float>(),
+                       psq.data_ptr<float>());
+  }
+  HIP_CHECK(hipGetLastError());
+  if (C % 4 == 0) {
+    int G4 = 1;
+    while (G4 * 2 <= std::min(C / 4, 16)) G4 *= 2;
+    hipLaunchKernelGGL(bn_stats_final_vec_kernel,
+                       dim3(ceil_div(C / 4, G4)), dim3(256), 0,
+                       cur_stream(), psum.data_ptr<float>(),
+                       psq.data_ptr<float>(), chunks, C, G4, (float)M,
+                       (float)eps, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(bn_stats_final_kernel, dim3(ceil_div(C, 32)),
+                       dim3(256), 0, cur_stream(), psum.data_ptr<float>(),
+                       psq.data_ptr<float>(), chunks, C, (float)M,
+                       (float)eps, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>());
+  }
+  HIP_CHECK(hipGetLastError());
+  return {mean, invstd};
+}
+
+// Finalize per-tile channel partials (psum, psq — each [chunks][C],
+// e.g. produced by the GEMM epilogues) into (mean, invstd). Same
+// fixed-tree kernel the standalone stats path uses.
+std::tuple<torch::Tensor, torch::Tensor> bn_stats_finalize(
+    torch::Tensor psum, torch::Tensor psq, double count, double eps) {
+  CHECK_GPU(psum); CHECK_CONTIG(psum); CHECK_CONTIG(psq);
+  const int chunks = (int)psum.size(0);
+  const int C = (int)psum.size(1);
+  auto opts = psum.options();
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  if (C % 4 == 0) {
+    int G4 = 1;
+    while (G4 * 2 <= std::min(C / 4, 16)) G4 *= 2;
+    hipLaunchKernelGGL(bn_stats_final_vec_kernel,
+                       dim3(ceil_div(C / 4, G4)), dim3(256), 0,
+                       cur_stream(), psum.data_ptr<float>(),
+                       psq.data_ptr<float>(), chunks, C, G4, (float)count,
+                       (float)eps, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(bn_stats_final_kernel, dim3(ceil_div(C, 32)),
+                       dim3(256), 0, cur_stream(), psum.data_ptr<float>(),
+                       psq.data_ptr<float>(), chunks, C, (float)count,
+                       (float)eps, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>());
+  }
+  HIP_CHECK(hipGetLastError());
+  return {mean, invstd};
+}
+
+// Normalization pass only (stats already known).
+torch::Tensor batchnorm_norm(torch::Tensor x, torch::Tensor gamma,
+                             torch::Tensor beta, torch::Tensor mean,
+                             torch::Tensor invstd, bool relu,
+                             c10::optional<torch::Tensor> residual) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  const int C = (int)x.size(-1);
+  const long M = x.numel() / C;
+  const bf16* resp =
+      residual.has_value() ? (const bf16*)residual->data_ptr() : nullptr;
+  auto y = torch::empty_like(x);
+  auto gc = gamma.contiguous();
+  auto bc = beta.contiguous();
+  if (C % 8 == 0)
+    hipLaunchKernelGGL(bn_norm_vec_kernel, dim3(ew_grid(x.numel() / 8)),
+                       dim3(1024), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), (const bf16*)gc.data_ptr(),
+                       (const bf16*)bc.data_ptr(), resp, M, C, relu ? 1 : 0,
+                       (bf16*)y.data_ptr());
+  else
+    hipLaunchKernelGGL(bn_norm_kernel, dim3(ew_grid(x.numel())), dim3(1024),
+                       0, cur_stream(), (const bf16*)x.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       (const bf16*)gc.data_ptr(), (const bf16*)bc.data_ptr(),
+                       resp, M, C, relu ? 1 : 0, (bf16*)y.data_ptr());
+  HIP_CHECK(hipGetLastError());
+  return y;
+}
+
 torch::Tensor global_avgpool_fwd(torch::Tensor x) {
   CHECK_GPU(x); CHECK_CONTIG(x);
   const int N = (int)x.size(0), C = (int)x.size(-1);

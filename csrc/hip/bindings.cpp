@@ -41,6 +41,16 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
 // conv_im2col.hip
 std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
     torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride, long pad);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+conv2d_fwd_bn(torch::Tensor x, torch::Tensor w, long stride, long pad);
+std::tuple<torch::Tensor, torch::Tensor> bn_stats(torch::Tensor x,
+                                                  double eps);
+std::tuple<torch::Tensor, torch::Tensor> bn_stats_finalize(
+    torch::Tensor psum, torch::Tensor psq, double count, double eps);
+torch::Tensor batchnorm_norm(torch::Tensor x, torch::Tensor gamma,
+                             torch::Tensor beta, torch::Tensor mean,
+                             torch::Tensor invstd, bool relu,
+                             c10::optional<torch::Tensor> residual);
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          long stride, long pad);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
@@ -92,6 +102,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");
   m.def("conv2d_fwd", &bflc::conv2d_fwd, "im2col + MFMA GEMM, NCHW");
   m.def("conv2d_fwd_col", &bflc::conv2d_fwd_col, "(y, col) - col for bwd");
+  m.def("conv2d_fwd_bn", &bflc::conv2d_fwd_bn,
+        "(y, col, psum, psq) - conv with fused epilogue BN stats");
+  m.def("bn_stats", &bflc::bn_stats, "(mean, invstd) of x [.., C]");
+  m.def("bn_stats_finalize", &bflc::bn_stats_finalize,
+        "(mean, invstd) from [chunks][C] partials");
+  m.def("batchnorm_norm", &bflc::batchnorm_norm,
+        py::arg("x"), py::arg("gamma"), py::arg("beta"), py::arg("mean"),
+        py::arg("invstd"), py::arg("relu"),
+        py::arg("residual") = py::none(),
+        "normalization pass with known stats");
   m.def("conv2d_bwd", &bflc::conv2d_bwd, "(dx, dw, db)",
         py::arg("x"), py::arg("w"), py::arg("dy"), py::arg("stride"),
         py::arg("pad"), py::arg("col_cache") = py::none());

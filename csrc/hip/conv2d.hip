@@ -374,6 +374,38 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
   return std::get<0>(conv2d_fwd_col(x, w, b, stride, pad));
 }
 
+// Conv forward WITH fused per-tile BN stats from the GEMM epilogue:
+// (y, col, psum, psq). col / psum / psq may be 0-size when that piece
+// was not produced (materialized-col fallback keeps col; split-K or
+// unsupported shapes leave the stats empty and the caller runs the
+// standalone stats pass).
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+conv2d_fwd_bn(torch::Tensor x, torch::Tensor w, long stride, long pad) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
+  auto sh = make_shape(x, w, stride, pad);
+  auto w2 = w.view({(long)sh.Kout, sh.RSC()});
+  auto zb = torch::zeros({(long)sh.Kout}, x.options());
+  auto y = torch::empty({(long)sh.N, (long)sh.OH, (long)sh.OW,
+                         (long)sh.Kout}, x.options());
+  std::pair<torch::Tensor, torch::Tensor> stats;
+  auto none = torch::empty({0}, x.options());
+  if (is_1x1_s1(sh)) {
+    auto col = x.view({sh.M(), (long)sh.C});
+    gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &zb,
+                  false, EpStore::kPlain, 0, &stats);
+    return {y, col, stats.first.defined() ? stats.first : none,
+            stats.second.defined() ? stats.second : none};
+  }
+  if (gemm_conv_fwd_raw(x, w2, y, sh, &zb, false, &stats))
+    return {y, none, stats.first.defined() ? stats.first : none,
+            stats.second.defined() ? stats.second : none};
+  auto col = im2col(x, sh);
+  gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &zb,
+                false, EpStore::kPlain, 0, &stats);
+  return {y, col, stats.first.defined() ? stats.first : none,
+          stats.second.defined() ? stats.second : none};
+}
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
     long pad, c10::optional<torch::Tensor> col_cache) {

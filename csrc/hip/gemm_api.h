@@ -31,10 +31,16 @@ enum class EpStore : int {
 //   tb: B accessed as B[n*K + k] (stored [N,K]), else B[k*N + n]
 //   bias: optional [N] bf16 added per column; relu: fused max(0,.)
 //   ohw: only for EpStore::kConvNCHW
+// bn_stats (optional out): when non-null and the launch runs as a
+// single K slice, receives fused per-tile-row channel partials
+// (psum, psq), each [chunks][N] fp32, finalized by bn_stats_finalize.
+// Left empty when the shape took a split-K or unsupported path.
 void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
                    torch::Tensor& C, long M, long N, long K, bool ta, bool tb,
                    const torch::Tensor* bias, bool relu, EpStore store,
-                   long ohw);
+                   long ohw,
+                   std::pair<torch::Tensor, torch::Tensor>* bn_stats
+                   = nullptr);
 
 // Implicit-GEMM NHWC conv forward: y[M, Kout] = im2col(x) @ w2^T with
 // the im2col gather fused into the GEMM's A staging (no col matrix).
@@ -42,7 +48,9 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
 // shape cannot take this path (caller materializes col instead).
 bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
                        torch::Tensor& y, const ConvShape& sh,
-                       const torch::Tensor* bias, bool relu);
+                       const torch::Tensor* bias, bool relu,
+                       std::pair<torch::Tensor, torch::Tensor>* bn_stats
+                       = nullptr);
 
 // Implicit-GEMM NHWC conv data-gradient: dx[M, C] = dy-gather @ wrot2
 // (wrot2[c][(r,s,kout)] = w[kout][r][s][c], i.e. w.permute(3,1,2,0)

@@ -66,7 +66,9 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
                             bf16* __restrict__ C, float* __restrict__ Cpart,
                             const bf16* __restrict__ bias, long M, long N,
                             long K, long kslice, int relu, int store_mode,
-                            long ohw, int vecA, int vecB, ConvShape csh) {
+                            long ohw, int vecA, int vecB, ConvShape csh,
+                            float* __restrict__ bn_psum = nullptr,
+                            float* __restrict__ bn_psq = nullptr) {
   constexpr int THREADS = WR * WC * 64;
   constexpr int FM = BM / WR / 16;
   constexpr int FN = BN / WC / 16;
@@ -307,6 +309,7 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
     return;
   }
 
+  float ssum[FN] = {}, ssq[FN] = {};
 #pragma unroll
   for (int fm = 0; fm < FM; ++fm) {
 #pragma unroll
@@ -320,13 +323,48 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         if (row >= M) continue;
         float v = acc[fm][fn][r] + bv;
         if (relu) v = fmaxf(v, 0.f);
+        const bf16 stored = f2b(v);
+        if (bn_psum) {  // per-channel stats of the ROUNDED output
+          const float sv = b2f(stored);
+          ssum[fn] += sv;
+          ssq[fn] += sv * sv;
+        }
         if (store_mode == (int)EpStore::kConvNCHW) {
           const long img = row / ohw, sp = row % ohw;
-          C[(img * N + col) * ohw + sp] = f2b(v);
+          C[(img * N + col) * ohw + sp] = stored;
         } else {
-          C[row * N + col] = f2b(v);
+          C[row * N + col] = stored;
         }
       }
+    }
+  }
+  if (bn_psum) {
+    // fixed-structure block reduce: each (column, slot) has exactly
+    // one writing thread; slot = (m-wave, l4) -> deterministic sums,
+    // partial row = this block's tile_m index (bitwise-stable layout).
+    constexpr int SLOTS = WR * 4;
+    __shared__ float eps_sum[BN * SLOTS], eps_sq[BN * SLOTS];
+    __syncthreads();
+    const int slot = (wave / WC) * 4 + l4;
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      const int cl = wn0 + fn * 16 + l15;
+      eps_sum[cl * SLOTS + slot] = ssum[fn];
+      eps_sq[cl * SLOTS + slot] = ssq[fn];
+    }
+    __syncthreads();
+    const long mrow = tile_m / BM;
+    for (int cl = tid; cl < BN; cl += THREADS) {
+      const long col = tile_n + cl;
+      if (col >= N) continue;
+      float a = 0.f, b2 = 0.f;
+#pragma unroll
+      for (int sl = 0; sl < SLOTS; ++sl) {
+        a += eps_sum[cl * SLOTS + sl];
+        b2 += eps_sq[cl * SLOTS + sl];
+      }
+      bn_psum[mrow * N + col] = a;
+      bn_psq[mrow * N + col] = b2;
     }
   }
 }
@@ -367,7 +405,9 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
                                bf16* __restrict__ C, float* __restrict__ Cpart,
                                const bf16* __restrict__ bias, long M, long N,
                                long K, long kslice, int relu, int store_mode,
-                               long ohw, ConvShape csh) {
+                               long ohw, ConvShape csh,
+                               float* __restrict__ bn_psum = nullptr,
+                               float* __restrict__ bn_psq = nullptr) {
   // wave grid: 2(M) x 4(N) for BN >= 64; 4(M) x 2(N) for BN == 32
   constexpr int WR = BN >= 64 ? 2 : 4, WC = BN >= 64 ? 4 : 2;
   constexpr int FM = BM / WR / 16, FN = BN / WC / 16;
@@ -586,6 +626,7 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
     return;
   }
 
+  float ssum[FN] = {}, ssq[FN] = {};
 #pragma unroll
   for (int fm = 0; fm < FM; ++fm) {
 #pragma unroll
@@ -599,13 +640,51 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
         if (row >= M) continue;
         float v = acc[fm][fn][r] + bv;
         if (relu) v = fmaxf(v, 0.f);
+        const bf16 stored = f2b(v);
+        if (bn_psum) {  // per-channel stats of the ROUNDED output
+          const float sv = b2f(stored);
+          ssum[fn] += sv;
+          ssq[fn] += sv * sv;
+        }
         if (store_mode == (int)EpStore::kConvNCHW) {
           const long img = row / ohw, sp = row % ohw;
-          C[(img * N + col) * ohw + sp] = f2b(v);
+          C[(img * N + col) * ohw + sp] = stored;
         } else {
-          C[row * N + col] = f2b(v);
+          C[row * N + col] = stored;
         }
       }
+    }
+  }
+  if (bn_psum) {
+    // fixed-structure block reduce: each (column, slot) has exactly
+    // one writing thread; slot = (m-wave, l4) -> deterministic sums,
+    // partial row = this block's tile_m index (bitwise-stable layout).
+    // Scratch ALIASES the staging LDS (full at 160 KiB; dead after the
+    // K loop, whose trailing barrier ordered all reads).
+    constexpr int SLOTS = WR * 4;
+    float* eps_sum = reinterpret_cast<float*>(&As[0][0][0]);
+    float* eps_sq = eps_sum + BN * SLOTS;
+    __syncthreads();
+    const int slot = (wave / WC) * 4 + l4;
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      const int cl = wn0 + fn * 16 + l15;
+      eps_sum[cl * SLOTS + slot] = ssum[fn];
+      eps_sq[cl * SLOTS + slot] = ssq[fn];
+    }
+    __syncthreads();
+    const long mrow = tile_m / BM;
+    for (int cl = tid; cl < BN; cl += 512) {
+      const long col = tile_n + cl;
+      if (col >= N) continue;
+      float a = 0.f, b2 = 0.f;
+#pragma unroll
+      for (int sl = 0; sl < SLOTS; ++sl) {
+        a += eps_sum[cl * SLOTS + sl];
+        b2 += eps_sq[cl * SLOTS + sl];
+      }
+      bn_psum[mrow * N + col] = a;
+      bn_psq[mrow * N + col] = b2;
     }
   }
 }
@@ -637,7 +716,9 @@ __global__ void gemm8p_kernel(const bf16* __restrict__ A,
                               const bf16* __restrict__ B,
                               bf16* __restrict__ C, float* __restrict__ Cpart,
                               const bf16* __restrict__ bias, long M, long N,
-                              long K, long kslice, int relu) {
+                              long K, long kslice, int relu,
+                              float* __restrict__ bn_psum = nullptr,
+                              float* __restrict__ bn_psq = nullptr) {
   constexpr int FM = 8, FN = 4;
   // ONE shared object (a second one makes hipcc drain vmcnt before
   // every ds_read of a glds pipeline — guide §5 trap 4a):
@@ -776,6 +857,7 @@ __global__ void gemm8p_kernel(const bf16* __restrict__ A,
       }
     return;
   }
+  float ssum[FN] = {}, ssq[FN] = {};
 #pragma unroll
   for (int fm = 0; fm < FM; ++fm) {
 #pragma unroll
@@ -787,8 +869,44 @@ __global__ void gemm8p_kernel(const bf16* __restrict__ A,
         const long row = tile_m + wm0 + fm * 16 + l4 * 4 + r;
         float v = acc[fm][fn][r] + bv;
         if (relu) v = fmaxf(v, 0.f);
-        C[row * N + col] = f2b(v);
+        const bf16 stored = f2b(v);
+        if (bn_psum) {
+          const float sv = b2f(stored);
+          ssum[fn] += sv;
+          ssq[fn] += sv * sv;
+        }
+        C[row * N + col] = stored;
       }
+    }
+  }
+  if (bn_psum) {
+    // same fixed-structure per-tile stats as gemm256_kernel; the 128 KB
+    // glds LDS image is dead after the K loop (loop barriers ordered
+    // all reads) and is reused as scratch.
+    constexpr int SLOTS = 8;  // 2 m-waves x l4
+    float* eps_sum = reinterpret_cast<float*>(&lds[0][0][0][0]);
+    float* eps_sq = eps_sum + 256 * SLOTS;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // drain any glds
+    __builtin_amdgcn_s_barrier();
+    const int slot = (wave >> 2) * 4 + l4;
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      const int cl = wn0 + fn * 16 + l15;
+      eps_sum[cl * SLOTS + slot] = ssum[fn];
+      eps_sq[cl * SLOTS + slot] = ssq[fn];
+    }
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const long mrow = tile_m / 256;
+    for (int cl = tid; cl < 256; cl += 512) {
+      float a = 0.f, b2 = 0.f;
+#pragma unroll
+      for (int sl = 0; sl < SLOTS; ++sl) {
+        a += eps_sum[cl * SLOTS + sl];
+        b2 += eps_sq[cl * SLOTS + sl];
+      }
+      bn_psum[mrow * N + tile_n + cl] = a;
+      bn_psq[mrow * N + tile_n + cl] = b2;
     }
   }
 }
@@ -1068,7 +1186,8 @@ torch::Tensor transpose_bf16(const torch::Tensor& X) {
 void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
                    torch::Tensor& C, long M, long N, long K, bool ta, bool tb,
                    const torch::Tensor* bias, bool relu, EpStore store,
-                   long ohw) {
+                   long ohw,
+                   std::pair<torch::Tensor, torch::Tensor>* bn_stats) {
   TORCH_CHECK(A.scalar_type() == at::kBFloat16 &&
               B.scalar_type() == at::kBFloat16, "gemm: bf16 only");
   CHECK_GPU(A); CHECK_GPU(B); CHECK_CONTIG(A); CHECK_CONTIG(B);
@@ -1133,12 +1252,23 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
     dim3 grid((unsigned)tiles, (unsigned)S);
     dim3 block(512);
     const int sm = (int)store;
+    float* stp = nullptr;
+    float* stq = nullptr;
+    if (bn_stats && S == 1 && store == EpStore::kPlain) {
+      const long chunks = (M + bm2 - 1) / bm2;
+      bn_stats->first = torch::empty({chunks, N},
+                                     A.options().dtype(at::kFloat));
+      bn_stats->second = torch::empty({chunks, N},
+                                      A.options().dtype(at::kFloat));
+      stp = bn_stats->first.data_ptr<float>();
+      stq = bn_stats->second.data_ptr<float>();
+    }
     // fully aligned large shapes take the 8-phase glds schedule
     if (bm2 == 256 && bn2 == 256 && M % 256 == 0 && N % 256 == 0 &&
         K % 64 == 0 && store == EpStore::kPlain) {
       hipLaunchKernelGGL(gemm8p_kernel, grid, block, 0, cur_stream(), a2,
                          b2, c, part_ptr, bs, M, N, K, kslice,
-                         relu ? 1 : 0);
+                         relu ? 1 : 0, stp, stq);
       HIP_CHECK(hipGetLastError());
       if (S > 1) {
         const long total = M * N;
@@ -1154,7 +1284,7 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
       hipLaunchKernelGGL(
           (gemm256_kernel<decltype(bmv)::value, decltype(bnv)::value>),
           grid, block, 0, cur_stream(), a2, b2, c, part_ptr, bs, M, N, K,
-          kslice, relu, sm, ohw, ConvShape{});
+          kslice, relu, sm, ohw, ConvShape{}, stp, stq);
     };
     using c64i = std::integral_constant<int, 64>;
     using c128i = std::integral_constant<int, 128>;
@@ -1311,7 +1441,9 @@ torch::Tensor colsum_bf16(const torch::Tensor& X) {
 template <int CMODE>
 bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
                         torch::Tensor& y, const ConvShape& sh,
-                        const torch::Tensor* bias, bool relu) {
+                        const torch::Tensor* bias, bool relu,
+                        std::pair<torch::Tensor, torch::Tensor>* bn_stats
+                        = nullptr) {
   // CMODE 1 (fwd):   y[M=N*OH*OW][Kout], K = R*S*C, A = x gather
   // CMODE 2 (dgrad): y[M=N*H*W][C],      K = R*S*Kout, A = dy gather
   const long M = CMODE == 1 ? sh.M() : (long)sh.N * sh.H * sh.W;
@@ -1340,6 +1472,17 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
       part_ptr = part.data_ptr<float>();
     }
     dim3 grid((unsigned)tiles, (unsigned)S);
+    float* stp = nullptr;
+    float* stq = nullptr;
+    if (bn_stats && CMODE == 1 && S == 1) {
+      const long chunks = (M + t.bm - 1) / t.bm;
+      bn_stats->first = torch::empty({chunks, N},
+                                     x.options().dtype(at::kFloat));
+      bn_stats->second = torch::empty({chunks, N},
+                                      x.options().dtype(at::kFloat));
+      stp = bn_stats->first.data_ptr<float>();
+      stq = bn_stats->second.data_ptr<float>();
+    }
     auto launchcc = [&](auto bm, auto bn, auto wr, auto wc) {
       constexpr int BMv = decltype(bm)::value, BNv = decltype(bn)::value;
       constexpr int WRv = decltype(wr)::value, WCv = decltype(wc)::value;
@@ -1347,7 +1490,7 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
       hipLaunchKernelGGL(
           (gemm_kernel<BMv, BNv, WRv, WCv, false, true, CMODE>), grid,
           block, 0, cur_stream(), a, b, c, part_ptr, bs, M, N, K, kslice,
-          relu ? 1 : 0, (int)EpStore::kPlain, 0, 0, 1, sh);
+          relu ? 1 : 0, (int)EpStore::kPlain, 0, 0, 1, sh, stp, stq);
     };
     using c32 = std::integral_constant<int, 32>;
     using c64 = std::integral_constant<int, 64>;
@@ -1412,11 +1555,22 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
   const bf16* bs = bias ? (const bf16*)bias->data_ptr() : nullptr;
   dim3 grid((unsigned)tiles, (unsigned)S);
   dim3 block(512);
+  float* stp = nullptr;
+  float* stq = nullptr;
+  if (bn_stats && CMODE == 1 && S == 1) {
+    const long chunks = (M + bm2 - 1) / bm2;
+    bn_stats->first = torch::empty({chunks, N},
+                                   x.options().dtype(at::kFloat));
+    bn_stats->second = torch::empty({chunks, N},
+                                    x.options().dtype(at::kFloat));
+    stp = bn_stats->first.data_ptr<float>();
+    stq = bn_stats->second.data_ptr<float>();
+  }
   auto launchc = [&](auto bmv, auto bnv) {
     hipLaunchKernelGGL(
         (gemm256_kernel<decltype(bmv)::value, decltype(bnv)::value, CMODE>),
         grid, block, 0, cur_stream(), a, b, c, part_ptr, bs, M, N, K,
-        kslice, relu ? 1 : 0, (int)EpStore::kPlain, 0, sh);
+        kslice, relu ? 1 : 0, (int)EpStore::kPlain, 0, sh, stp, stq);
   };
   using c64i = std::integral_constant<int, 64>;
   using c128i = std::integral_constant<int, 128>;
@@ -1446,8 +1600,9 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
 
 bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
                        torch::Tensor& y, const ConvShape& sh,
-                       const torch::Tensor* bias, bool relu) {
-  return conv_implicit_gemm<1>(x, w2, y, sh, bias, relu);
+                       const torch::Tensor* bias, bool relu,
+                       std::pair<torch::Tensor, torch::Tensor>* bn_stats) {
+  return conv_implicit_gemm<1>(x, w2, y, sh, bias, relu, bn_stats);
 }
 
 bool gemm_conv_dgrad_raw(const torch::Tensor& dy, const torch::Tensor& wrot2,

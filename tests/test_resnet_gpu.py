@@ -123,6 +123,48 @@ class TestPoolAddRelu:
         assert_close(dx, x2.grad.permute(0, 2, 3, 1), rel=0.01)
 
 
+class TestConvBNFused:
+    @pytest.mark.parametrize("n,c,h,k,r,stride,pad,res", [
+        (4, 32, 14, 64, 3, 1, 1, False),   # implicit-GEMM fwd + stats
+        (2, 64, 8, 64, 1, 1, 0, True),     # 1x1 pure-GEMM + residual
+        (2, 16, 16, 32, 3, 2, 1, False),   # strided, narrow Kout
+        (2, 64, 14, 256, 1, 1, 0, True),   # 256-path epilogue stats
+    ])
+    def test_matches_composed_oracle(self, n, c, h, k, r, stride, pad, res):
+        from bflc_amd.ops import functional as O
+        torch.manual_seed(9)
+        x = bf(torch.randn(n, h, h, c)).requires_grad_(True)
+        w = bf(torch.randn(k, r, r, c) * 0.1).requires_grad_(True)
+        g = bf(torch.rand(k) + 0.5).requires_grad_(True)
+        b = bf(torch.randn(k) * 0.1).requires_grad_(True)
+        residual = None
+        if res:
+            oh = (h + 2 * pad - r) // stride + 1
+            residual = bf(torch.randn(n, oh, oh, k))
+        y = O.conv2d_bn(x, w, g, b, stride, pad, relu=True,
+                        residual=residual)
+        dy = bf(torch.randn_like(y.float()))
+        (y.float() * dy.float()).sum().backward()
+
+        # CPU composed fp32 oracle on the same bf16-rounded inputs
+        xc = x.detach().float().cpu().requires_grad_(True)
+        wc = w.detach().float().cpu().requires_grad_(True)
+        gc = g.detach().float().cpu().requires_grad_(True)
+        bc = b.detach().float().cpu().requires_grad_(True)
+        rc = residual.detach().float().cpu() if res else None
+        yref = O.batchnorm2d(
+            O.conv2d(xc, wc, None, stride, pad), gc, bc, relu=True,
+            residual=rc)
+        # bf16 stats vs fp32 oracle: tail elements with tiny |y| sit
+        # right at a 0.05 band (1/8192 measured at 0.057-equiv)
+        assert_close(y, yref, rel=0.08)
+        (yref * dy.float().cpu()).sum().backward()
+        assert_close(x.grad, xc.grad, rel=0.06)
+        assert_close(w.grad, wc.grad, rel=0.06)
+        assert_close(g.grad, gc.grad, rel=0.06)
+        assert_close(b.grad, bc.grad, rel=0.06)
+
+
 class TestResNetGPU:
     @pytest.mark.parametrize("model,hw,nclass,lr,steps", [
         ("resnet20", 32, 10, 0.05, 5),
