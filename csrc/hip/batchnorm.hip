@@ -755,6 +755,29 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
   return {dx, dgamma, dbeta};
 }
 
+// Collapse [chunks][C] fp32 partials to [64][C]: out row j sums input
+// rows j, j+64, j+128, ... ascending (fixed order). Thread per
+// (out-row, float4 granule) — coalesced, and gives the finalize kernel
+// a bounded chunk count regardless of how many tiles the producing
+// GEMM emitted.
+__global__ void bn_collapse_kernel(const float* __restrict__ in, int chunks,
+                                   int C, float* __restrict__ out) {
+  const int c4g = C / 4;
+  const long total = 64L * c4g;
+  long t = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; t < total; t += stride) {
+    const int c4 = (int)(t % c4g) * 4;
+    const int j = (int)(t / c4g);
+    float4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k = j; k < chunks; k += 64) {
+      const float4 v = *reinterpret_cast<const float4*>(&in[(long)k * C + c4]);
+      acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+    }
+    *reinterpret_cast<float4*>(&out[(long)j * C + c4]) = acc;
+  }
+}
+
 // Stats-only pass: (mean, invstd) of x viewed [M, C] (the standalone
 // path when the producing GEMM could not fuse the partials).
 std::tuple<torch::Tensor, torch::Tensor> bn_stats(torch::Tensor x,
@@ -810,8 +833,24 @@ std::tuple<torch::Tensor, torch::Tensor> bn_stats(torch::Tensor x,
 std::tuple<torch::Tensor, torch::Tensor> bn_stats_finalize(
     torch::Tensor psum, torch::Tensor psq, double count, double eps) {
   CHECK_GPU(psum); CHECK_CONTIG(psum); CHECK_CONTIG(psq);
-  const int chunks = (int)psum.size(0);
+  int chunks = (int)psum.size(0);
   const int C = (int)psum.size(1);
+  if (chunks > 96 && C % 4 == 0) {  // bound the serial final sweep
+    auto cs = torch::empty({64, C}, psum.options());
+    auto cq = torch::empty({64, C}, psum.options());
+    const long total = 64L * (C / 4);
+    const int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+    hipLaunchKernelGGL(bn_collapse_kernel, dim3(blocks), dim3(256), 0,
+                       cur_stream(), psum.data_ptr<float>(), chunks, C,
+                       cs.data_ptr<float>());
+    hipLaunchKernelGGL(bn_collapse_kernel, dim3(blocks), dim3(256), 0,
+                       cur_stream(), psq.data_ptr<float>(), chunks, C,
+                       cq.data_ptr<float>());
+    HIP_CHECK(hipGetLastError());
+    psum = cs;
+    psq = cq;
+    chunks = 64;
+  }
   auto opts = psum.options();
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
