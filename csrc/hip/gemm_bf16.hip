@@ -1610,67 +1610,6 @@ bool gemm_conv_dgrad_raw(const torch::Tensor& dy, const torch::Tensor& wrot2,
   return conv_implicit_gemm<2>(dy, wrot2, dx, sh, nullptr, false);
 }
 
-// Small-M' wgrad: dW[Kout, RSC] with Kout <= 64 is a "fat colsum" —
-// per k-row an outer product dy2[k][:] x col[k][n0..n0+32), col
-// gathered implicitly from x with the (n, oh, ow) raster walked
-// INCREMENTALLY (no per-k divisions). Each thread owns (col c,
-// m-lane) and accumulates Kout/8 fp32 values; K splits into slices
-// whose fp32 partials the fixed-order splitk_reduce sums. The MFMA
-// tile path measured ~20x off the traffic bound here (tiles of 16-64
-// rows cannot amortize staging latency).
-__global__ void wgrad_small_kernel(const bf16* __restrict__ dy2,
-                                   const bf16* __restrict__ x,
-                                   float* __restrict__ part, long Mw,
-                                   long Nw, long K, long kslice,
-                                   ConvShape sh) {
-  const int MLANES = 8;
-  const int c = blockIdx.x * 32 + (threadIdx.x & 31);
-  const int mlane = (threadIdx.x >> 5);  // 0..7
-  const long k0 = (long)blockIdx.y * kslice;
-  const long k1 = min(K, k0 + kslice);
-  // decompose the flat RSC column once per thread
-  int rr = 0, ss = 0, cc = 0;
-  if (c < Nw) {
-    const int rs = (int)(c / sh.C);
-    cc = (int)(c - (long)rs * sh.C);
-    rr = rs / sh.S;
-    ss = rs - rr * sh.S;
-  }
-  // raster coordinates of k0, walked incrementally
-  int ow = (int)(k0 % sh.OW);
-  int oh = (int)((k0 / sh.OW) % sh.OH);
-  int nn = (int)(k0 / ((long)sh.OW * sh.OH));
-  float acc[8] = {};  // m = mlane + j*8; static 8-iteration loops so
-                      // the accumulators stay in registers (rule 20)
-  for (long k = k0; k < k1; ++k) {
-    if (c < Nw) {
-      const int ih = oh * sh.stride - sh.pad + rr;
-      const int iw = ow * sh.stride - sh.pad + ss;
-      if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W) {
-        const float cv =
-            b2f(x[(((long)nn * sh.H + ih) * sh.W + iw) * sh.C + cc]);
-        const bf16* dr = &dy2[k * Mw];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int m = mlane + j * MLANES;
-          if (m < Mw) acc[j] += cv * b2f(dr[m]);
-        }
-      }
-    }
-    if (++ow == sh.OW) {
-      ow = 0;
-      if (++oh == sh.OH) { oh = 0; ++nn; }
-    }
-  }
-  if (c >= Nw) return;
-  float* out = part + (long)blockIdx.y * Mw * Nw;
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    const int m = mlane + j * MLANES;
-    if (m < Mw) out[(long)m * Nw + c] = acc[j];
-  }
-}
-
 // wgrad: dW[Kout, RSC] = dy2^T @ implicit-col(x). A = dy2 [M, Kout]
 // through the TA scatter staging, B = x through the CMODE-3 gather —
 // the col matrix never exists. Always the small-tile split-K shape
@@ -1679,29 +1618,6 @@ bool gemm_conv_wgrad_raw(const torch::Tensor& dy2, const torch::Tensor& x,
                          torch::Tensor& dw, const ConvShape& sh) {
   if (sh.C % 8 != 0) return false;
   const long M = sh.Kout, N = sh.RSC(), K = sh.M();
-  if (M <= 64) {  // fat-colsum path (see wgrad_small_kernel)
-    const long nblocks = (N + 31) / 32;
-    // slices sized so nblocks * S ~ 1024 blocks and >= ~512 rows each
-    long S = std::min<long>(std::max<long>(1024 / std::max(nblocks, 1L), 1),
-                            std::max<long>(K / 512, 1));
-    const long kslice = (K + S - 1) / S;
-    S = (K + kslice - 1) / kslice;
-    auto part = torch::empty({S, M, N}, x.options().dtype(at::kFloat));
-    hipLaunchKernelGGL(wgrad_small_kernel,
-                       dim3((unsigned)nblocks, (unsigned)S), dim3(256), 0,
-                       cur_stream(), (const bf16*)dy2.data_ptr(),
-                       (const bf16*)x.data_ptr(), part.data_ptr<float>(),
-                       M, N, K, kslice, sh);
-    HIP_CHECK(hipGetLastError());
-    const long total = M * N;
-    int blocks = (int)std::min<long>((total + 31) / 32, 16384);
-    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
-                       cur_stream(), part.data_ptr<float>(), (int)S, M, N,
-                       (bf16*)dw.data_ptr(), nullptr, 0,
-                       (int)EpStore::kPlain, 0);
-    HIP_CHECK(hipGetLastError());
-    return true;
-  }
   const bf16* a = (const bf16*)dy2.data_ptr();
   const bf16* b = (const bf16*)x.data_ptr();
   bf16* c = (bf16*)dw.data_ptr();
