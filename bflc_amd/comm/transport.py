@@ -16,7 +16,7 @@ from __future__ import annotations
 import datetime
 import os
 import pickle
-from typing import Any, List, Optional, Sequence
+from typing import Any, List, Optional
 
 import torch
 import torch.distributed as dist
