@@ -1,0 +1,36 @@
+"""Driver-contract guard: bench.py must print ONE JSON line with the
+agreed fields (BASELINE.json metric, whole-job value, weak scaling)."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_json_contract():
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"), "--steps", "1",
+         "--warmup", "0", "--samples-per-client", "64",
+         "--batch-size", "32"],
+        capture_output=True, text=True, timeout=600, cwd=ROOT)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["metric"] == "fl_round_wall_clock_ms"
+    assert d["unit"] == "ms/round"
+    assert d["higher_is_better"] is False
+    assert d["scaling"] == "weak"
+    assert d["n_gpus"] == 1 and d["steps"] == 1 and d["warmup"] == 0
+    assert d["value"] == d["ms_per_step"] > 0
+    assert d["dtype"] in ("bf16", "fp32")
+    assert d["data"] == "synthetic"
+    assert 0.0 <= d["test_acc"] <= 1.0
+    cfg = d["config"]
+    for key in ("model", "global_batch", "parallelism", "clients",
+                "committee", "update_quota", "aggregate_top_k",
+                "batch_size", "partition"):
+        assert key in cfg, key
