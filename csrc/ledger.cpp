@@ -20,7 +20,12 @@
 //  - ties in top-k broken by (score desc, id asc) (reference: unstable
 //    std::sort with cmp_by_value, .cpp:118-120, 364-366);
 //  - duplicate score upload overwrites without double-counting
-//    (reference increments score_count unconditionally, .cpp:279-289).
+//    (reference increments score_count unconditionally, .cpp:279-289);
+//  - a committee rotation that found fewer scored trainers than
+//    comm_count refills the remaining seats in registration order
+//    (the reference's committee shrinks while its aggregation trigger
+//    stays at COMM_COUNT — a permanent deadlock; found by the
+//    hypothesis fuzz in tests/test_ledger_property.py).
 
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
