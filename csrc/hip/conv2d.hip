@@ -424,16 +424,17 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     gemm_bf16_raw(dy2, wT, dxv, sh.M(), sh.RSC(), sh.Kout, false, true,
                   nullptr, false, EpStore::kPlain, 0);
     dgrad_done = true;
-  } else if (sh.Kout % 8 == 0 && sh.C <= 64 && sh.C >= 8 &&
+  } else if (sh.Kout % 8 == 0 && sh.C <= 128 && sh.C >= 8 &&
              sh.stride == 1) {
     // (stride > 1 wastes stride^2 of the implicit MFMA work on
     // misaligned taps — the 7x7/2 stem forced implicit cost +17% of a
     // ResNet-50 round; tiny C additionally wastes the N tile)
     // implicit: dx[M, C] = dy-gather @ w.permute(3,1,2,0) — no dcol
-    // matrix, no col2im pass. Only when C is small: there the dcol
-    // round trip dominates (R*S*C columns vs C outputs); at large C
-    // the materialized dcol GEMM runs on the faster 8-phase/dbuf path
-    // and wins (measured +20% on ResNet-50 when forced implicit).
+    // matrix, no col2im pass. Only while C is small-to-mid: there the
+    // dcol round trip dominates (R*S*C columns vs C outputs); at
+    // large C the materialized dcol GEMM runs on the faster
+    // 8-phase/dbuf path and wins (A/B on ResNet-50: C<=128 86.9
+    // ms/round, C<=256 87.9, forced always +20%).
     auto wrot2 = w.permute({3, 1, 2, 0}).contiguous()
                      .view({(long)sh.C, (long)sh.R * sh.S * sh.Kout});
     dx = torch::empty_like(x);
