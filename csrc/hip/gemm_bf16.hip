@@ -1264,6 +1264,8 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
       stq = bn_stats->second.data_ptr<float>();
     }
     // fully aligned large shapes take the 8-phase glds schedule
+    // (also at shallow K: an A/B routing K-slices < 8 tiles to the
+    // double-buffered kernel measured +9% on a ResNet-50 round)
     if (bm2 == 256 && bn2 == 256 && M % 256 == 0 && N % 256 == 0 &&
         K % 64 == 0 && store == EpStore::kPlain) {
       hipLaunchKernelGGL(gemm8p_kernel, grid, block, 0, cur_stream(), a2,
