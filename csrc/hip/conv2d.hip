@@ -73,7 +73,7 @@ __global__ void im2col_nhwc_vec_kernel(const bf16* __restrict__ x,
 // thread per (m, r, s) copies the C contiguous channels.
 __global__ void im2col_nhwc_kernel(const bf16* __restrict__ x,
                                    bf16* __restrict__ col, ConvShape sh,
-                                   long total_mrs) {
+                                   long ldc, long total_mrs) {
   const int rs = sh.R * sh.S;
   long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
@@ -86,7 +86,7 @@ __global__ void im2col_nhwc_kernel(const bf16* __restrict__ x,
     const int n = (int)(m / ((long)sh.OW * sh.OH));
     const int ih = oh * sh.stride - sh.pad + r;
     const int iw = ow * sh.stride - sh.pad + s;
-    bf16* out = &col[m * sh.RSC() + (long)k * sh.C];
+    bf16* out = &col[m * ldc + (long)k * sh.C];
     if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W) {
       const bf16* src = &x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C];
       for (int c = 0; c < sh.C; ++c) out[c] = src[c];
@@ -315,23 +315,41 @@ ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
   return sh;
 }
 
+// When C % 8 != 0 (FEMNIST C=1, ResNet stems C=3) the col matrix's K
+// is zero-padded to a multiple of 8 so the consuming GEMM takes the
+// 16-B vector staging paths (the K=9 FEMNIST conv1 GEMM ran scalar
+// staging at ~6 TF); the padded columns contribute zeros and the
+// weight is padded to match by the callers.
 torch::Tensor im2col(const torch::Tensor& x, const ConvShape& sh) {
-  auto col = torch::empty({sh.M(), sh.RSC()}, x.options());
   if (sh.C % 8 == 0) {
+    auto col = torch::empty({sh.M(), sh.RSC()}, x.options());
     const long total_g = sh.M() * sh.R * sh.S * (sh.C / 8);
     hipLaunchKernelGGL(im2col_nhwc_vec_kernel, dim3(ew_grid(total_g)),
                        dim3(256), 0, cur_stream(),
                        (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
                        total_g);
-  } else {
-    const long total_mrs = sh.M() * sh.R * sh.S;
-    hipLaunchKernelGGL(im2col_nhwc_kernel, dim3(ew_grid(total_mrs)),
-                       dim3(256), 0, cur_stream(),
-                       (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
-                       total_mrs);
+    HIP_CHECK(hipGetLastError());
+    return col;
   }
+  const long rscp = (sh.RSC() + 7) / 8 * 8;
+  auto col = rscp == sh.RSC()
+                 ? torch::empty({sh.M(), rscp}, x.options())
+                 : torch::zeros({sh.M(), rscp}, x.options());
+  const long total_mrs = sh.M() * sh.R * sh.S;
+  hipLaunchKernelGGL(im2col_nhwc_kernel, dim3(ew_grid(total_mrs)),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
+                     rscp, total_mrs);
   HIP_CHECK(hipGetLastError());
   return col;
+}
+
+// Zero-pad the [Kout, RSC] weight view to the (padded) col K.
+torch::Tensor pad_w2(const torch::Tensor& w2, long kp) {
+  if (kp == w2.size(1)) return w2;
+  auto out = torch::zeros({w2.size(0), kp}, w2.options());
+  out.slice(1, 0, w2.size(1)).copy_(w2);
+  return out;
 }
 
 bool is_1x1_s1(const ConvShape& sh) {
@@ -400,8 +418,9 @@ conv2d_fwd_bn(torch::Tensor x, torch::Tensor w, long stride, long pad) {
     return {y, none, stats.first.defined() ? stats.first : none,
             stats.second.defined() ? stats.second : none};
   auto col = im2col(x, sh);
-  gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &zb,
-                false, EpStore::kPlain, 0, &stats);
+  const long kp = col.size(1);
+  gemm_bf16_raw(col, pad_w2(w2, kp), y, sh.M(), sh.Kout, kp, false, true,
+                &zb, false, EpStore::kPlain, 0, &stats);
   return {y, col, stats.first.defined() ? stats.first : none,
           stats.second.defined() ? stats.second : none};
 }
@@ -479,8 +498,16 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
                    ? *col_cache
                    : (is_1x1_s1(sh) ? x.view({sh.M(), (long)sh.C})
                                     : im2col(x, sh));
-    gemm_bf16_raw(dy2, col, dw, sh.Kout, sh.RSC(), sh.M(), true, false,
-                  nullptr, false, EpStore::kPlain, 0);
+    const long kp = col.size(1);
+    if (kp == sh.RSC()) {
+      gemm_bf16_raw(dy2, col, dw, sh.Kout, sh.RSC(), sh.M(), true, false,
+                    nullptr, false, EpStore::kPlain, 0);
+    } else {  // K-padded col: compute into [Kout, kp], narrow back
+      auto dwp = torch::empty({(long)sh.Kout, kp}, dw.options());
+      gemm_bf16_raw(dy2, col, dwp, sh.Kout, kp, sh.M(), true, false,
+                    nullptr, false, EpStore::kPlain, 0);
+      dw.copy_(dwp.slice(1, 0, sh.RSC()));
+    }
   }
 
   auto db = colsum_bf16(dy2);
