@@ -382,8 +382,9 @@ std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
   if (gemm_conv_fwd_raw(x, w2, y, sh, &bc, false))
     return {y, torch::empty({0}, x.options())};
   auto col = im2col(x, sh);
-  gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &bc,
-                false, EpStore::kPlain, 0);
+  const long kp = col.size(1);  // may be K-padded to %8 (C<8 convs)
+  gemm_bf16_raw(col, pad_w2(w2, kp), y, sh.M(), sh.Kout, kp, false, true,
+                &bc, false, EpStore::kPlain, 0);
   return {y, col};
 }
 
