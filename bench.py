@@ -43,6 +43,8 @@ def main() -> None:
     ap.add_argument("--eval", action="store_true", help="eval every round")
     ap.add_argument("--phases", action="store_true",
                     help="print per-phase timings of each timed round")
+    ap.add_argument("--no-graphs", action="store_true",
+                    help="disable the hipGraph-captured train step")
     args = ap.parse_args()
 
     from bflc_amd.config import FLConfig
@@ -73,7 +75,7 @@ def main() -> None:
         batch_size=bs, partition="dirichlet",
         dirichlet_alpha=0.3, eval_samples=min(4096, 2 * spc),
         learning_rate=0.01, byzantine_clients=args.byzantine,
-        optimizer=args.optimizer)
+        optimizer=args.optimizer, use_graphs=not args.no_graphs)
 
     t = Transport()
     shards, test = make_federated(cfg)

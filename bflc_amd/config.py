@@ -46,6 +46,10 @@ class FLConfig:
     dirichlet_alpha: float = 0.3
     byzantine_clients: int = 0     # label-flip attackers (BASELINE config 4)
 
+    # --- execution ---
+    use_graphs: bool = True        # hipGraph-captured train step (GPU+SGD;
+                                   # eager fallback elsewhere, fl/graphs.py)
+
     def __post_init__(self) -> None:
         if self.comm_count < 1 or self.client_num < 1:
             raise ValueError("comm_count and client_num must be >= 1")

@@ -103,6 +103,16 @@ class FLEngine:
         self._round = 0
         self.metrics = JsonlLogger(metrics_path, rank=self.rank)
 
+        # hipGraph-captured train step (fl/graphs.py): replay one
+        # captured kernel DAG per minibatch instead of relaunching
+        # ~dozens of kernels. SGD-only; falls back to eager on CPU, on
+        # Adam, or if capture fails on this ROCm build.
+        import os
+        self._stepper = None
+        self._use_graphs = (cfg.use_graphs and self.device.type == "cuda"
+                            and cfg.optimizer == "sgd"
+                            and os.environ.get("BFLC_GRAPHS", "1") != "0")
+
     # ------------------------------------------------------------------
     def _planned_submitters(self) -> List[int]:
         """The deterministic prefix of trainers whose updates will be
@@ -119,35 +129,69 @@ class FLEngine:
         return order[: self.cfg.needed_update_count]
 
     # ------------------------------------------------------------------
+    def _graphed_stepper(self, shard: Shard, bs: int):
+        """Lazily capture the train-step hipGraph (fl/graphs.py); one
+        stepper serves every local client (equal batch shapes). Returns
+        None — and stops trying — if capture is unavailable."""
+        if self._stepper is not None:
+            xb = shard.x[:bs]
+            return self._stepper if self._stepper.matches(xb, shard.y[:bs]) \
+                else None
+        try:
+            from bflc_amd.fl.graphs import GraphedTrainStep
+            self._stepper = GraphedTrainStep(
+                self.model, self.cfg.learning_rate, shard.x[:bs],
+                shard.y[:bs])
+        except Exception as e:  # capture unsupported: eager fallback
+            import warnings
+            warnings.warn(f"hipGraph capture failed, running eager: {e}")
+            self._use_graphs = False
+            return None
+        return self._stepper
+
+    # ------------------------------------------------------------------
     def _local_train(self, client: int) -> Tuple[torch.Tensor, int, float]:
         """Local train step (reference local_training, main.py:103-158):
         start from the global model, run local_epochs passes of
         minibatch SGD/Adam, return pseudo-gradient delta=(W0-W)/lr."""
         cfg = self.cfg
         shard = self.shards[client]
-        self.model.set_flat(self.global_flat)
         n = shard.n
         bs = min(cfg.batch_size, n)
         total_batches = max(n // bs, 1)
-        # accumulate the loss on-device; ONE host sync per client per round
-        cost_accum = torch.zeros((), device=self.device)
+        # capture BEFORE loading the round's weights: graph warmup and
+        # capture execute real SGD steps and mutate flat/cflat
+        stepper = self._graphed_stepper(shard, bs) if self._use_graphs \
+            else None
+        self.model.set_flat(self.global_flat)
         if cfg.optimizer == "adam":
             m = torch.zeros_like(self.global_flat)
             v = torch.zeros_like(self.global_flat)
             step = 0
-        for _ in range(cfg.local_epochs):
-            for bi in range(total_batches):
-                xb = shard.x[bi * bs:(bi + 1) * bs]
-                yb = shard.y[bi * bs:(bi + 1) * bs]
-                self.model.zero_grad()
-                loss = self.model.loss(xb, yb)
-                loss.backward()
-                if cfg.optimizer == "adam":
-                    step += 1
-                    self.model.adam_step(m, v, step, cfg.learning_rate)
-                else:
-                    self.model.sgd_step(cfg.learning_rate)
-                cost_accum += loss.detach()
+        if stepper is not None:
+            stepper.cost.zero_()
+            for _ in range(cfg.local_epochs):
+                for bi in range(total_batches):
+                    stepper.step(shard.x[bi * bs:(bi + 1) * bs],
+                                 shard.y[bi * bs:(bi + 1) * bs])
+            cost_accum = stepper.cost
+        else:
+            # accumulate the loss on-device; ONE host sync per client
+            # per round
+            cost_accum = torch.zeros((), device=self.device)
+            for _ in range(cfg.local_epochs):
+                for bi in range(total_batches):
+                    xb = shard.x[bi * bs:(bi + 1) * bs]
+                    yb = shard.y[bi * bs:(bi + 1) * bs]
+                    self.model.zero_grad()
+                    loss = self.model.loss(xb, yb)
+                    loss.backward()
+                    if cfg.optimizer == "adam":
+                        step += 1
+                        self.model.adam_step(m, v, step, cfg.learning_rate)
+                    else:
+                        self.model.sgd_step(cfg.learning_rate)
+                    cost_accum += loss.detach()
         avg_cost = float(cost_accum) / (total_batches * cfg.local_epochs)
         # delta = (W0 - W)/lr  (reference main.py:153-154)
         delta = self.global_flat.clone()

@@ -43,3 +43,33 @@ def test_long_run_checkpoint_resume(tmp_path):
     # well above the 62-class chance rate (~0.016; measured ~0.05 at
     # this reduced shard size, ~0.43 at the full bench config)
     assert acc_mid > 0.04
+
+
+def test_graphed_step_matches_eager():
+    """The hipGraph-captured train step (fl/graphs.py) replays exactly
+    the eager kernel sequence: after the same rounds from the same seed
+    the global models must agree."""
+    from bflc_amd.config import FLConfig
+    base = dict(model="femnist_cnn", n_class=62, samples_per_client=512,
+                batch_size=256, eval_samples=512, partition="dirichlet",
+                learning_rate=0.01)
+    runs = {}
+    for graphs in (True, False):
+        cfg = FLConfig.for_world(1, use_graphs=graphs, **base)
+        eng = _engine(cfg)
+        eng.run(10)
+        runs[graphs] = eng.global_flat.clone()
+        if graphs:  # the graph path must actually have been taken
+            assert eng._stepper is not None, \
+                "hipGraph capture fell back to eager on the GPU"
+    assert torch.allclose(runs[True], runs[False], atol=1e-6, rtol=1e-5)
+
+
+def test_graphed_step_off_for_adam():
+    from bflc_amd.config import FLConfig
+    cfg = FLConfig.for_world(1, model="mlp", n_features=16, n_class=4,
+                             samples_per_client=256, batch_size=128,
+                             eval_samples=128, optimizer="adam")
+    eng = _engine(cfg)
+    eng.run(3)
+    assert eng._stepper is None
