@@ -69,30 +69,49 @@ __global__ void im2col_nhwc_vec_kernel(const bf16* __restrict__ x,
   }
 }
 
-// Scalar variant for C % 8 != 0 (FEMNIST C=1, ResNet stems C=3): one
-// thread per (m, r, s) copies the C contiguous channels.
-__global__ void im2col_nhwc_kernel(const bf16* __restrict__ x,
-                                   bf16* __restrict__ col, ConvShape sh,
-                                   long ldc, long total_mrs) {
-  const int rs = sh.R * sh.S;
+// Variant for C % 8 != 0 (FEMNIST C=1, ResNet stems C=3): one thread
+// per 8-element output GRANULE gathers its (r,s,c) taps and emits one
+// 16-B store; the K padding to %8 is written inline (the host
+// allocates with torch::empty, no separate fill pass) and parallelism
+// scales with the row length, so both the 9-wide FEMNIST rows and the
+// 147-wide 7x7 stems fill the chip. (A thread-per-(m,r,s) scalar
+// kernel paid 2-B scattered stores + a zero-fill pass — 49 us on the
+// FEMNIST conv1 shape vs ~6 us of traffic; a thread-per-row register
+// window fixed FEMNIST (25 us) but serialized the 147-tap stem rows,
+// 397 us vs 344.)
+__global__ void im2col_nhwc_gran_kernel(const bf16* __restrict__ x,
+                                        bf16* __restrict__ col,
+                                        ConvShape sh, long ldc,
+                                        long total_g) {
+  const int gran = (int)(ldc / 8);
+  const int rsc = (int)sh.RSC();
   long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   const bf16 zero = f2b(0.f);
-  for (; g < total_mrs; g += stride) {
-    const int k = (int)(g % rs);
-    const long m = g / rs;
-    const int s = k % sh.S, r = k / sh.S;
+  for (; g < total_g; g += stride) {
+    const int gi = (int)(g % gran);
+    const long m = g / gran;
     const int ow = (int)(m % sh.OW), oh = (int)((m / sh.OW) % sh.OH);
     const int n = (int)(m / ((long)sh.OW * sh.OH));
-    const int ih = oh * sh.stride - sh.pad + r;
-    const int iw = ow * sh.stride - sh.pad + s;
-    bf16* out = &col[m * ldc + (long)k * sh.C];
-    if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W) {
-      const bf16* src = &x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C];
-      for (int c = 0; c < sh.C; ++c) out[c] = src[c];
-    } else {
-      for (int c = 0; c < sh.C; ++c) out[c] = zero;
+    const int ih0 = oh * sh.stride - sh.pad;
+    const int iw0 = ow * sh.stride - sh.pad;
+    const int k0 = gi * 8;
+    bf16x8_t v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = k0 + j;
+      bf16 val = zero;
+      if (k < rsc) {
+        const int c = k % sh.C;
+        const int rs = k / sh.C;
+        const int s = rs % sh.S, r = rs / sh.S;
+        const int ih = ih0 + r, iw = iw0 + s;
+        if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)
+          val = x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C + c];
+      }
+      v[j] = val;
     }
+    *reinterpret_cast<bf16x8_t*>(&col[m * ldc + k0]) = v;
   }
 }
 
@@ -165,10 +184,16 @@ __global__ void col2im_nhwc_kernel(const bf16* __restrict__ dcol,
   }
 }
 
-// ---- maxpool (NHWC, C-vectorized, int32 argmax = ih*W+iw) ----
+// ---- maxpool (NHWC, C-vectorized) ----
+// The argmax is stored as the WINDOW code r*S+s in a uint8 (windows are
+// tiny; R*S <= 255), not a global int32 offset: the index plane is the
+// same element count as y, so int32 indices cost 2x the y bytes
+// themselves — uint8 makes the fwd+bwd index traffic 1/4.
+typedef uint8_t u8x8_t __attribute__((ext_vector_type(8)));
+
 __global__ void maxpool_nhwc_vec_fwd(const bf16* __restrict__ x,
                                      bf16* __restrict__ y,
-                                     int* __restrict__ idx, ConvShape sh,
+                                     uint8_t* __restrict__ idx, ConvShape sh,
                                      long total_g) {
   const int c8g = sh.C / 8;
   long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -190,7 +215,7 @@ __global__ void maxpool_nhwc_vec_fwd(const bf16* __restrict__ x,
         if (iw >= sh.W) break;
         const bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(
             &x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C + c8]);
-        const int code = ih * sh.W + iw;
+        const int code = r * sh.S + s;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const float f = b2f(v[j]);
@@ -199,16 +224,19 @@ __global__ void maxpool_nhwc_vec_fwd(const bf16* __restrict__ x,
       }
     }
     bf16x8_t out;
-    i32x8_t oidx;
+    u8x8_t oidx;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) { out[j] = f2b(best[j]); oidx[j] = bi[j]; }
+    for (int j = 0; j < 8; ++j) {
+      out[j] = f2b(best[j]);
+      oidx[j] = (uint8_t)bi[j];
+    }
     *reinterpret_cast<bf16x8_t*>(&y[m * sh.C + c8]) = out;
-    *reinterpret_cast<i32x8_t*>(&idx[m * sh.C + c8]) = oidx;
+    *reinterpret_cast<u8x8_t*>(&idx[m * sh.C + c8]) = oidx;
   }
 }
 
 __global__ void maxpool_nhwc_vec_bwd(const bf16* __restrict__ dy,
-                                     const int* __restrict__ idx,
+                                     const uint8_t* __restrict__ idx,
                                      bf16* __restrict__ dx, ConvShape sh,
                                      long total_g) {
   const int c8g = sh.C / 8;
@@ -219,7 +247,6 @@ __global__ void maxpool_nhwc_vec_bwd(const bf16* __restrict__ dy,
     const long i = g / c8g;  // (n, ih, iw)
     const int iw = (int)(i % sh.W), ih = (int)((i / sh.W) % sh.H);
     const int n = (int)(i / ((long)sh.W * sh.H));
-    const int code = ih * sh.W + iw;
     const int oh_lo = max(0, (ih - sh.R + sh.stride) / sh.stride);
     const int oh_hi = min(sh.OH - 1, ih / sh.stride);
     const int ow_lo = max(0, (iw - sh.S + sh.stride) / sh.stride);
@@ -227,14 +254,17 @@ __global__ void maxpool_nhwc_vec_bwd(const bf16* __restrict__ dy,
     float acc[8] = {};
     for (int oh = oh_lo; oh <= oh_hi; ++oh)
       for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+        // this input pixel sits at window position (r,s) of (oh,ow)
+        const int code = (ih - oh * sh.stride) * sh.S
+                         + (iw - ow * sh.stride);
         const long m = ((long)n * sh.OH + oh) * sh.OW + ow;
-        const i32x8_t iv = *reinterpret_cast<const i32x8_t*>(
+        const u8x8_t iv = *reinterpret_cast<const u8x8_t*>(
             &idx[m * sh.C + c8]);
         const bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(
             &dy[m * sh.C + c8]);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          if (iv[j] == code) acc[j] += b2f(dv[j]);
+          if ((int)iv[j] == code) acc[j] += b2f(dv[j]);
       }
     bf16x8_t out;
 #pragma unroll
@@ -245,7 +275,8 @@ __global__ void maxpool_nhwc_vec_bwd(const bf16* __restrict__ dy,
 
 // Scalar maxpool for C % 8 != 0.
 __global__ void maxpool_nhwc_fwd(const bf16* __restrict__ x,
-                                 bf16* __restrict__ y, int* __restrict__ idx,
+                                 bf16* __restrict__ y,
+                                 uint8_t* __restrict__ idx,
                                  ConvShape sh, long total) {
   long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
@@ -264,16 +295,16 @@ __global__ void maxpool_nhwc_fwd(const bf16* __restrict__ x,
         if (iw >= sh.W) break;
         const float f =
             b2f(x[(((long)n * sh.H + ih) * sh.W + iw) * sh.C + c]);
-        if (f > best) { best = f; bi = ih * sh.W + iw; }
+        if (f > best) { best = f; bi = r * sh.S + s; }
       }
     }
     y[g] = f2b(best);
-    idx[g] = bi;
+    idx[g] = (uint8_t)bi;
   }
 }
 
 __global__ void maxpool_nhwc_bwd(const bf16* __restrict__ dy,
-                                 const int* __restrict__ idx,
+                                 const uint8_t* __restrict__ idx,
                                  bf16* __restrict__ dx, ConvShape sh,
                                  long total) {
   long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -283,7 +314,6 @@ __global__ void maxpool_nhwc_bwd(const bf16* __restrict__ dy,
     const long i = g / sh.C;
     const int iw = (int)(i % sh.W), ih = (int)((i / sh.W) % sh.H);
     const int n = (int)(i / ((long)sh.W * sh.H));
-    const int code = ih * sh.W + iw;
     const int oh_lo = max(0, (ih - sh.R + sh.stride) / sh.stride);
     const int oh_hi = min(sh.OH - 1, ih / sh.stride);
     const int ow_lo = max(0, (iw - sh.S + sh.stride) / sh.stride);
@@ -291,8 +321,10 @@ __global__ void maxpool_nhwc_bwd(const bf16* __restrict__ dy,
     float acc = 0.f;
     for (int oh = oh_lo; oh <= oh_hi; ++oh)
       for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+        const int code = (ih - oh * sh.stride) * sh.S
+                         + (iw - ow * sh.stride);
         const long m = ((long)n * sh.OH + oh) * sh.OW + ow;
-        if (idx[m * sh.C + c] == code) acc += b2f(dy[m * sh.C + c]);
+        if ((int)idx[m * sh.C + c] == code) acc += b2f(dy[m * sh.C + c]);
       }
     dx[g] = f2b(acc);
   }
@@ -332,14 +364,12 @@ torch::Tensor im2col(const torch::Tensor& x, const ConvShape& sh) {
     return col;
   }
   const long rscp = (sh.RSC() + 7) / 8 * 8;
-  auto col = rscp == sh.RSC()
-                 ? torch::empty({sh.M(), rscp}, x.options())
-                 : torch::zeros({sh.M(), rscp}, x.options());
-  const long total_mrs = sh.M() * sh.R * sh.S;
-  hipLaunchKernelGGL(im2col_nhwc_kernel, dim3(ew_grid(total_mrs)),
+  auto col = torch::empty({sh.M(), rscp}, x.options());
+  const long total_g = sh.M() * (rscp / 8);
+  hipLaunchKernelGGL(im2col_nhwc_gran_kernel, dim3(ew_grid(total_g)),
                      dim3(256), 0, cur_stream(),
                      (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
-                     rscp, total_mrs);
+                     rscp, total_g);
   HIP_CHECK(hipGetLastError());
   return col;
 }
@@ -528,18 +558,20 @@ std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
   sh.OW = (sh.W - sh.S) / sh.stride + 1;
   auto y = torch::empty({(long)sh.N, (long)sh.OH, (long)sh.OW, (long)sh.C},
                         x.options());
-  auto idx = torch::empty_like(y, y.options().dtype(at::kInt));
+  TORCH_CHECK(sh.R * sh.S <= 255, "maxpool window too large for u8 idx");
+  auto idx = torch::empty_like(y, y.options().dtype(at::kByte));
   if (sh.C % 8 == 0) {
     const long total_g = sh.M() * (sh.C / 8);
     hipLaunchKernelGGL(maxpool_nhwc_vec_fwd, dim3(ew_grid(total_g)),
                        dim3(256), 0, cur_stream(),
                        (const bf16*)x.data_ptr(), (bf16*)y.data_ptr(),
-                       idx.data_ptr<int>(), sh, total_g);
+                       idx.data_ptr<uint8_t>(), sh, total_g);
   } else {
     const long total = y.numel();
     hipLaunchKernelGGL(maxpool_nhwc_fwd, dim3(ew_grid(total)), dim3(256), 0,
                        cur_stream(), (const bf16*)x.data_ptr(),
-                       (bf16*)y.data_ptr(), idx.data_ptr<int>(), sh, total);
+                       (bf16*)y.data_ptr(), idx.data_ptr<uint8_t>(), sh,
+                       total);
   }
   HIP_CHECK(hipGetLastError());
   return {y, idx};
@@ -561,13 +593,14 @@ torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx,
     const long total_g = (long)sh.N * sh.H * sh.W * (sh.C / 8);
     hipLaunchKernelGGL(maxpool_nhwc_vec_bwd, dim3(ew_grid(total_g)),
                        dim3(256), 0, cur_stream(),
-                       (const bf16*)dy.data_ptr(), idx.data_ptr<int>(),
+                       (const bf16*)dy.data_ptr(), idx.data_ptr<uint8_t>(),
                        (bf16*)dx.data_ptr(), sh, total_g);
   } else {
     const long total = dx.numel();
     hipLaunchKernelGGL(maxpool_nhwc_bwd, dim3(ew_grid(total)), dim3(256), 0,
                        cur_stream(), (const bf16*)dy.data_ptr(),
-                       idx.data_ptr<int>(), (bf16*)dx.data_ptr(), sh, total);
+                       idx.data_ptr<uint8_t>(), (bf16*)dx.data_ptr(), sh,
+                       total);
   }
   HIP_CHECK(hipGetLastError());
   return dx;

@@ -11,6 +11,8 @@ try:
 except ImportError:  # pragma: no cover
     pytest.skip("hypothesis not installed", allow_module_level=True)
 
+import numpy as np
+
 from bflc_amd._ledger import Admit, CommitteeLedger
 from bflc_amd.config import FLConfig
 
@@ -60,16 +62,19 @@ class PyLedgerModel:
         need = (1 if self.cfg.client_num == 1 else self.cfg.comm_count)
         if len(self.scores) < need:
             return None
-        # aggregate: median per trainer, top aggregate_count
+        # aggregate: median per trainer (in fp32 — the ledger keeps
+        # the reference's float rounding, csrc/ledger.cpp:81),
+        # top aggregate_count
         meds = {}
         for tr in self.updates:
-            vals = sorted(s.get(tr, 0.0) for s in self.scores.values())
+            vals = sorted(np.float32(s.get(tr, 0.0))
+                          for s in self.scores.values())
             k = len(vals)
             if k % 2 == 1:
                 m = vals[k // 2]
             else:  # reference GetMid averages the two middles
-                m = 0.5 * (vals[k // 2 - 1] + vals[k // 2])
-            meds[tr] = m
+                m = np.float32(0.5) * (vals[k // 2 - 1] + vals[k // 2])
+            meds[tr] = float(m)
         ranked = sorted(meds.items(), key=lambda kv: (-kv[1], kv[0]))
         sel = [o for o, _ in ranked[: self.cfg.aggregate_count]]
         return sel
@@ -79,11 +84,12 @@ class PyLedgerModel:
         # trainers -> next committee
         meds = {}
         for tr in self.updates:
-            vals = sorted(s.get(tr, 0.0) for s in self.scores.values())
+            vals = sorted(np.float32(s.get(tr, 0.0))
+                          for s in self.scores.values())
             k = len(vals)
-            m = vals[k // 2] if k % 2 else 0.5 * (vals[k // 2 - 1]
-                                                  + vals[k // 2])
-            meds[tr] = m
+            m = vals[k // 2] if k % 2 else \
+                np.float32(0.5) * (vals[k // 2 - 1] + vals[k // 2])
+            meds[tr] = float(m)
         ranked = sorted(meds.items(), key=lambda kv: (-kv[1], kv[0]))
         new_comm = [o for o, _ in ranked[: self.cfg.comm_count]]
         # refill short committees in registration order (ledger

@@ -51,6 +51,19 @@ class TestLinear:
         ref = bf(x).float().cpu() @ bf(w).float().cpu() + bf(b).float().cpu()
         assert_close(y, ref)
 
+    @pytest.mark.parametrize("m,k,n", [
+        # thin VALU path (K<=32, N<=64, M>=64k, B [N][K]):
+        (131072, 16, 32),   # FEMNIST conv1 GEMM shape
+        (131072, 32, 16),   # CIFAR stem shape
+    ])
+    def test_thin_gemm(self, m, k, n):
+        torch.manual_seed(3)
+        a = torch.randn(m, k)
+        bn = torch.randn(n, k)  # B in [N][K] layout (tb=True)
+        y = hip().gemm_raw(bf(a), bf(bn), False, True)
+        ref = bf(a).float().cpu() @ bf(bn).float().cpu().t()
+        assert_close(y, ref)
+
     @pytest.mark.parametrize("m,k,n", [(100, 5, 2), (64, 3136, 128),
                                        (37, 100, 62), (128, 256, 192),
                                        # 256-path TB (dx) / TA (dw) layouts
