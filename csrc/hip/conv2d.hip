@@ -79,6 +79,51 @@ __global__ void im2col_nhwc_vec_kernel(const bf16* __restrict__ x,
 // FEMNIST conv1 shape vs ~6 us of traffic; a thread-per-row register
 // window fixed FEMNIST (25 us) but serialized the 147-tap stem rows,
 // 397 us vs 344.)
+// Short padded rows (rscp <= 32: FEMNIST conv1, CIFAR stems): one
+// thread per row stages the whole row through a bf16x8 register window
+// — fewer index divmods than the granule kernel and the row fits in
+// 2-4 stores (FEMNIST: 25 us rowvec vs 40 us granule vs 49 us scalar).
+__global__ void im2col_nhwc_rowvec_kernel(const bf16* __restrict__ x,
+                                          bf16* __restrict__ col,
+                                          ConvShape sh, long ldc,
+                                          long total_m) {
+  long m = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const bf16 zero = f2b(0.f);
+  for (; m < total_m; m += stride) {
+    const int ow = (int)(m % sh.OW), oh = (int)((m / sh.OW) % sh.OH);
+    const int n = (int)(m / ((long)sh.OW * sh.OH));
+    const int ih0 = oh * sh.stride - sh.pad;
+    const int iw0 = ow * sh.stride - sh.pad;
+    bf16x8_t buf;
+    int nb = 0;
+    long w = m * ldc;
+    for (int r = 0; r < sh.R; ++r) {
+      const int ih = ih0 + r;
+      const bool okh = ih >= 0 && ih < sh.H;
+      const bf16* row = &x[((long)n * sh.H + ih) * sh.W * sh.C];
+      for (int s = 0; s < sh.S; ++s) {
+        const int iw = iw0 + s;
+        const bool ok = okh && iw >= 0 && iw < sh.W;
+        for (int c = 0; c < sh.C; ++c) {
+          buf[nb++] = ok ? row[(long)iw * sh.C + c] : zero;
+          if (nb == 8) {
+            *reinterpret_cast<bf16x8_t*>(&col[w]) = buf;
+            w += 8;
+            nb = 0;
+          }
+        }
+      }
+    }
+    if (nb) {  // K padding: zero the tail granule
+      for (; nb < 8; ++nb) buf[nb] = zero;
+      *reinterpret_cast<bf16x8_t*>(&col[w]) = buf;
+    }
+  }
+}
+
+// Long rows (7x7 stems, RSC 147): granule-per-thread keeps the
+// parallelism proportional to the row length.
 __global__ void im2col_nhwc_gran_kernel(const bf16* __restrict__ x,
                                         bf16* __restrict__ col,
                                         ConvShape sh, long ldc,
@@ -365,11 +410,19 @@ torch::Tensor im2col(const torch::Tensor& x, const ConvShape& sh) {
   }
   const long rscp = (sh.RSC() + 7) / 8 * 8;
   auto col = torch::empty({sh.M(), rscp}, x.options());
-  const long total_g = sh.M() * (rscp / 8);
-  hipLaunchKernelGGL(im2col_nhwc_gran_kernel, dim3(ew_grid(total_g)),
-                     dim3(256), 0, cur_stream(),
-                     (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
-                     rscp, total_g);
+  if (rscp <= 32) {
+    const long total_m = sh.M();
+    hipLaunchKernelGGL(im2col_nhwc_rowvec_kernel, dim3(ew_grid(total_m)),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
+                       rscp, total_m);
+  } else {
+    const long total_g = sh.M() * (rscp / 8);
+    hipLaunchKernelGGL(im2col_nhwc_gran_kernel, dim3(ew_grid(total_g)),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), (bf16*)col.data_ptr(), sh,
+                       rscp, total_g);
+  }
   HIP_CHECK(hipGetLastError());
   return col;
 }

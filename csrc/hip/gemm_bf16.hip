@@ -1192,6 +1192,10 @@ __global__ __launch_bounds__(256) void gemm_thin_kernel(
   for (int i = threadIdx.x; i < N; i += blockDim.x)
     bl[i] = bias ? b2f(bias[i]) : 0.f;
   __syncthreads();
+  // One row per thread. (A 4-rows-per-thread ILP variant — 1 LDS read
+  // amortized over 4 FMAs — measured 69 us vs 46 on the FEMNIST conv1
+  // shape: the fatter register file cut occupancy more than the issue
+  // mix helped.)
   long m = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (; m < M; m += stride) {
