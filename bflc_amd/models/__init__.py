@@ -41,7 +41,7 @@ class MLP(FlatModel):
                 ("b2", (c.n_class,), "zeros")]
 
     def forward(self, x):
-        h = O.relu(O.linear(x, self.p("w1"), self.p("b1")))
+        h = O.linear(x, self.p("w1"), self.p("b1"), relu=True)
         return O.linear(h, self.p("w2"), self.p("b2"))
 
 
@@ -59,12 +59,12 @@ class FemnistCNN(FlatModel):
                 ("f2w", (128, c.n_class), "xavier"), ("f2b", (c.n_class,), "zeros")]
 
     def forward(self, x):
-        h = O.relu(O.conv2d(x, self.p("c1w"), self.p("c1b"), 1, 1))
+        h = O.conv2d(x, self.p("c1w"), self.p("c1b"), 1, 1, relu=True)
         h = O.maxpool2d(h, 2)
-        h = O.relu(O.conv2d(h, self.p("c2w"), self.p("c2b"), 1, 1))
+        h = O.conv2d(h, self.p("c2w"), self.p("c2b"), 1, 1, relu=True)
         h = O.maxpool2d(h, 2)
         h = h.reshape(h.shape[0], -1)
-        h = O.relu(O.linear(h, self.p("f1w"), self.p("f1b")))
+        h = O.linear(h, self.p("f1w"), self.p("f1b"), relu=True)
         return O.linear(h, self.p("f2w"), self.p("f2b"))
 
 

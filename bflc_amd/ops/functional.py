@@ -62,27 +62,37 @@ class _LinearFn(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor):
-        ctx.save_for_backward(x, w)
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
+                relu: bool):
+        ctx.relu = relu
         if x.is_cuda:
-            return hip_ops().linear_fwd(x, w, b)
-        return torch.addmm(b, x, w)
+            y = hip_ops().linear_fwd(x, w, b, relu)
+        else:
+            y = torch.addmm(b, x, w)
+            if relu:
+                y = torch.relu(y)
+        ctx.save_for_backward(x, w, y if relu else None)
+        return y
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
-        x, w = ctx.saved_tensors
+        x, w, y = ctx.saved_tensors
         dy = dy.contiguous()
+        if ctx.relu:  # fused-relu epilogue: mask dy by the saved output
+            dy = hip_ops().relu_bwd(y, dy) if x.is_cuda \
+                else dy * (y > 0).to(dy.dtype)
         if x.is_cuda:
             dx, dw, db = hip_ops().linear_bwd(x, w, dy)
         else:
             dx = dy @ w.t()
             dw = x.t() @ dy
             db = dy.sum(0)
-        return dx, dw, db
+        return dx, dw, db, None
 
 
-def linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    return _LinearFn.apply(x, w, b)
+def linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
+           relu: bool = False) -> torch.Tensor:
+    return _LinearFn.apply(x, w, b, relu)
 
 
 class _SoftmaxCEFn(torch.autograd.Function):
@@ -152,28 +162,36 @@ class _Conv2dFn(torch.autograd.Function):
     permute to NCHW, torch fp32 conv, permute back."""
 
     @staticmethod
-    def forward(ctx, x, w, b, stride: int, padding: int):
+    def forward(ctx, x, w, b, stride: int, padding: int, relu: bool):
         ctx.stride, ctx.padding = stride, padding
         ctx.has_bias = b is not None
+        ctx.relu = relu
         if x.is_cuda:
             y, col = hip_ops().conv2d_fwd_col(
                 x, w, b if b is not None else
                 torch.zeros(w.shape[0], device=x.device, dtype=x.dtype),
-                stride, padding)
+                stride, padding, relu)
             # keep col for wgrad when the fwd materialized one (the
             # implicit-GEMM path returns an empty marker; bwd then
             # builds its own)
-            ctx.save_for_backward(x, w, col)
+            ctx.save_for_backward(x, w, col, y if relu else None)
             return y
-        ctx.save_for_backward(x, w)
         y = F.conv2d(x.permute(0, 3, 1, 2), w.permute(0, 3, 1, 2), b,
                      stride=stride, padding=padding)
-        return y.permute(0, 2, 3, 1).contiguous()
+        y = y.permute(0, 2, 3, 1).contiguous()
+        if relu:
+            y = torch.relu(y)
+        ctx.save_for_backward(x, w, None, y if relu else None)
+        return y
 
     @staticmethod
     def backward(ctx, dy):
         x, w = ctx.saved_tensors[:2]
         dy = dy.contiguous()
+        if ctx.relu:  # fused-relu epilogue: mask dy by the saved output
+            y = ctx.saved_tensors[3]
+            dy = hip_ops().relu_bwd(y, dy) if x.is_cuda \
+                else dy * (y > 0).to(dy.dtype)
         if x.is_cuda:
             col = ctx.saved_tensors[2]
             if col.numel() == 0:
@@ -191,11 +209,12 @@ class _Conv2dFn(torch.autograd.Function):
             dx = dx.permute(0, 2, 3, 1).contiguous()
             dw = dw.permute(0, 2, 3, 1).contiguous()
             db = dy.sum(dim=(0, 1, 2))
-        return dx, dw, (db if ctx.has_bias else None), None, None
+        return dx, dw, (db if ctx.has_bias else None), None, None, None
 
 
-def conv2d(x, w, b=None, stride: int = 1, padding: int = 0) -> torch.Tensor:
-    return _Conv2dFn.apply(x, w, b, stride, padding)
+def conv2d(x, w, b=None, stride: int = 1, padding: int = 0,
+           relu: bool = False) -> torch.Tensor:
+    return _Conv2dFn.apply(x, w, b, stride, padding, relu)
 
 
 class _MaxPool2dFn(torch.autograd.Function):

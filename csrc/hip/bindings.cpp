@@ -33,14 +33,16 @@ double accuracy(torch::Tensor logits, torch::Tensor target);
 torch::Tensor accuracy_t(torch::Tensor logits, torch::Tensor target);
 
 // gemm_bf16.hip
-torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b);
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         bool relu);
 torch::Tensor gemm_raw(torch::Tensor A, torch::Tensor B, bool ta, bool tb);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy);
 
 // conv_im2col.hip
 std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
-    torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride, long pad);
+    torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride, long pad,
+    bool relu);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 conv2d_fwd_bn(torch::Tensor x, torch::Tensor w, long stride, long pad);
 std::tuple<torch::Tensor, torch::Tensor> bn_stats(torch::Tensor x,
@@ -97,11 +99,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("accuracy", &bflc::accuracy, "fused argmax-compare-reduce");
   m.def("accuracy_t", &bflc::accuracy_t,
         "accuracy as a device tensor (no host sync)");
-  m.def("linear_fwd", &bflc::linear_fwd, "MFMA bf16 GEMM + bias");
+  m.def("linear_fwd", &bflc::linear_fwd, "MFMA bf16 GEMM + bias (+relu)",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("relu") = false);
   m.def("gemm_raw", &bflc::gemm_raw, "raw GEMM (bench/ablation)");
   m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");
   m.def("conv2d_fwd", &bflc::conv2d_fwd, "im2col + MFMA GEMM, NCHW");
-  m.def("conv2d_fwd_col", &bflc::conv2d_fwd_col, "(y, col) - col for bwd");
+  m.def("conv2d_fwd_col", &bflc::conv2d_fwd_col, "(y, col) - col for bwd",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("stride"),
+        py::arg("pad"), py::arg("relu") = false);
   m.def("conv2d_fwd_bn", &bflc::conv2d_fwd_bn,
         "(y, col, psum, psq) - conv with fused epilogue BN stats");
   m.def("bn_stats", &bflc::bn_stats, "(mean, invstd) of x [.., C]");

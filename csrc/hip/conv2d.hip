@@ -69,16 +69,12 @@ __global__ void im2col_nhwc_vec_kernel(const bf16* __restrict__ x,
   }
 }
 
-// Variant for C % 8 != 0 (FEMNIST C=1, ResNet stems C=3): one thread
-// per 8-element output GRANULE gathers its (r,s,c) taps and emits one
-// 16-B store; the K padding to %8 is written inline (the host
-// allocates with torch::empty, no separate fill pass) and parallelism
-// scales with the row length, so both the 9-wide FEMNIST rows and the
-// 147-wide 7x7 stems fill the chip. (A thread-per-(m,r,s) scalar
-// kernel paid 2-B scattered stores + a zero-fill pass — 49 us on the
-// FEMNIST conv1 shape vs ~6 us of traffic; a thread-per-row register
-// window fixed FEMNIST (25 us) but serialized the 147-tap stem rows,
-// 397 us vs 344.)
+// Variants for C % 8 != 0 (FEMNIST C=1, ResNet stems C=3). Both write
+// the K padding to %8 inline (the host allocates with torch::empty, no
+// separate fill pass) and emit only 16-B stores; a thread-per-(m,r,s)
+// scalar kernel paid 2-B scattered stores + a zero-fill pass (49 us on
+// the FEMNIST conv1 shape vs ~6 us of traffic).
+//
 // Short padded rows (rscp <= 32: FEMNIST conv1, CIFAR stems): one
 // thread per row stages the whole row through a bf16x8 register window
 // — fewer index divmods than the granule kernel and the row fits in
@@ -446,7 +442,7 @@ bool is_1x1_s1(const ConvShape& sh) {
 // cache free). For 1x1 stride-1 convs col is just a VIEW of x.
 std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
     torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride,
-    long pad) {
+    long pad, bool relu) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv: bf16 only");
   auto sh = make_shape(x, w, stride, pad);
@@ -457,23 +453,23 @@ std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
   if (is_1x1_s1(sh)) {
     auto col = x.view({sh.M(), (long)sh.C});
     gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &bc,
-                  false, EpStore::kPlain, 0);
+                  relu, EpStore::kPlain, 0);
     return {y, col};
   }
   // implicit-GEMM: the im2col gather runs inside the GEMM's A staging;
   // no col matrix exists (wgrad materializes its own in the backward).
-  if (gemm_conv_fwd_raw(x, w2, y, sh, &bc, false))
+  if (gemm_conv_fwd_raw(x, w2, y, sh, &bc, relu))
     return {y, torch::empty({0}, x.options())};
   auto col = im2col(x, sh);
   const long kp = col.size(1);  // may be K-padded to %8 (C<8 convs)
   gemm_bf16_raw(col, pad_w2(w2, kp), y, sh.M(), sh.Kout, kp, false, true,
-                &bc, false, EpStore::kPlain, 0);
+                &bc, relu, EpStore::kPlain, 0);
   return {y, col};
 }
 
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          long stride, long pad) {
-  return std::get<0>(conv2d_fwd_col(x, w, b, stride, pad));
+  return std::get<0>(conv2d_fwd_col(x, w, b, stride, pad, false));
 }
 
 // Conv forward WITH fused per-tile BN stats from the GEMM epilogue:

@@ -1762,13 +1762,14 @@ torch::Tensor gemm_raw(torch::Tensor A, torch::Tensor B, bool ta, bool tb) {
 // linear layer entry points (reference main.py:120 fwd, 127-130 bwd)
 // ---------------------------------------------------------------------------
 
-torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b) {
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         bool relu) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
   long M = x.size(0), K = x.size(1), N = w.size(1);
   TORCH_CHECK(w.size(0) == K, "shape mismatch");
   auto y = torch::empty({M, N}, x.options());
   auto bc = b.contiguous();
-  gemm_bf16_raw(x, w, y, M, N, K, false, false, &bc, false,
+  gemm_bf16_raw(x, w, y, M, N, K, false, false, &bc, relu,
                 EpStore::kPlain, 0);
   return y;
 }

@@ -135,6 +135,28 @@ class TestConv:
 
 
 class TestPoolReluAcc:
+    def test_conv_fused_relu_matches_separate(self):
+        from bflc_amd.ops import functional as O
+        torch.manual_seed(7)
+        dev = torch.device("cuda", 0)
+        x = torch.randn(4, 14, 14, 32, device=dev,
+                        dtype=torch.bfloat16).requires_grad_(True)
+        w = torch.randn(64, 3, 3, 32, device=dev,
+                        dtype=torch.bfloat16).requires_grad_(True)
+        b = torch.randn(64, device=dev,
+                        dtype=torch.bfloat16).requires_grad_(True)
+        y = O.conv2d(x, w, b, 1, 1, relu=True)
+        dy = torch.randn_like(y)
+        (y.float() * dy.float()).sum().backward()
+        g = [t.grad.clone() for t in (x, w, b)]
+        for t in (x, w, b):
+            t.grad = None
+        y2 = O.relu(O.conv2d(x, w, b, 1, 1))
+        assert torch.equal(y, y2)
+        (y2.float() * dy.float()).sum().backward()
+        for a, t in zip(g, (x, w, b)):
+            assert torch.equal(a, t.grad)
+
     def test_maxpool(self):
         torch.manual_seed(4)
         x = torch.randn(3, 14, 14, 8)  # NHWC

@@ -48,6 +48,33 @@ class TestBatchNormOracle:
         assert torch.allclose(g.grad, g2.grad, atol=1e-5)
         assert torch.allclose(b.grad, b2.grad, atol=1e-5)
 
+    def test_fused_relu_epilogue_matches_unfused(self):
+        torch.manual_seed(5)
+        x = torch.randn(2, 7, 7, 3, requires_grad=True)
+        w = torch.randn(8, 3, 3, 3, requires_grad=True)
+        b = torch.randn(8, requires_grad=True)
+        y = O.conv2d(x, w, b, 1, 1, relu=True)
+        dy = torch.randn_like(y)
+        (y * dy).sum().backward()
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        y2 = O.relu(O.conv2d(x2, w2, b2, 1, 1))
+        assert torch.allclose(y, y2)
+        (y2 * dy).sum().backward()
+        assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+        assert torch.allclose(w.grad, w2.grad, atol=1e-5)
+        assert torch.allclose(b.grad, b2.grad, atol=1e-5)
+
+        a1 = torch.randn(16, 6, requires_grad=True)
+        w1 = torch.randn(6, 4, requires_grad=True)
+        b1 = torch.randn(4, requires_grad=True)
+        z = O.linear(a1, w1, b1, relu=True)
+        assert torch.allclose(z, torch.relu(a1 @ w1 + b1), atol=1e-5)
+        z.sum().backward()
+        mask = (a1 @ w1 + b1 > 0).float()
+        assert torch.allclose(w1.grad, a1.t() @ mask, atol=1e-5)
+
     def test_gap_and_add_relu(self):
         x = torch.randn(2, 5, 5, 4, requires_grad=True)  # NHWC
         y = O.global_avgpool(x)
