@@ -435,6 +435,21 @@ bool is_1x1_s1(const ConvShape& sh) {
   return sh.R == 1 && sh.S == 1 && sh.stride == 1 && sh.pad == 0;
 }
 
+// Measured implicit-vs-materialized crossovers, overridable for A/B
+// profiling runs without a rebuild.
+int env_gate(const char* name, int dflt) {
+  const char* e = getenv(name);
+  return e ? atoi(e) : dflt;
+}
+int wgrad_implicit_max_kout() {
+  static int v = env_gate("BFLC_WGRAD_IMPLICIT_MAX_KOUT", 64);
+  return v;
+}
+int dgrad_implicit_max_c() {
+  static int v = env_gate("BFLC_DGRAD_IMPLICIT_MAX_C", 128);
+  return v;
+}
+
 }  // namespace
 
 // Returns (y, col) so the autograd wrapper hands col back to wgrad
@@ -523,8 +538,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     gemm_bf16_raw(dy2, wT, dxv, sh.M(), sh.RSC(), sh.Kout, false, true,
                   nullptr, false, EpStore::kPlain, 0);
     dgrad_done = true;
-  } else if (sh.Kout % 8 == 0 && sh.C <= 128 && sh.C >= 8 &&
-             sh.stride == 1) {
+  } else if (sh.Kout % 8 == 0 && sh.C <= dgrad_implicit_max_c() &&
+             sh.C >= 8 && sh.stride == 1) {
     // (stride > 1 wastes stride^2 of the implicit MFMA work on
     // misaligned taps — the 7x7/2 stem forced implicit cost +17% of a
     // ResNet-50 round; tiny C additionally wastes the N tile)
@@ -570,7 +585,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   // GEMM does Kout flops per col element, so small Kout => col-bound
   // (measured: implicit at Kout>=128 ran 150us/call, slower than the
   // materialized dbuf path it replaced)
-  if (!is_1x1_s1(sh) && sh.Kout <= 64 &&
+  if (!is_1x1_s1(sh) && sh.Kout <= wgrad_implicit_max_kout() &&
       !(col_cache.has_value() && col_cache->numel() > 0))
     wgrad_done = gemm_conv_wgrad_raw(dy2, x, dw, sh);
   if (!wgrad_done) {
