@@ -75,6 +75,13 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
 
   __shared__ bf16 As[BM][BKP];
   __shared__ bf16 Bs[BN][BKP];
+  // Row-dependent XOR swizzle (multiple of 8, so 16-B groups and the
+  // b128 frag reads stay aligned): the b16 scatter stagings (TA / !TB /
+  // implicit wgrad) write 8-row-strided columns whose dword stride is
+  // 32 mod 64 banks — an 8-way conflict no 16-B-aligned padding can
+  // fix; XORing k by (row & 24) spreads them to 4 banks (2x), measured
+  // on the ResNet-20 implicit-wgrad kernel.
+  auto swz = [](int row, int k) { return k ^ (row & 24); };
 
   const int ntn = (int)((N + BN - 1) / BN);
   const int ntm = (int)((M + BM - 1) / BM);
@@ -142,11 +149,11 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
           src = (((long)nn * csh.OH + oh) * csh.OW + ow) * csh.Kout + kk8;
         }
         if (ok) {
-          *reinterpret_cast<bf16x8_t*>(&As[m][k8]) =
+          *reinterpret_cast<bf16x8_t*>(&As[m][swz(m, k8)]) =
               *reinterpret_cast<const bf16x8_t*>(&A[src]);
         } else {
           u16x8_t z = {};
-          *reinterpret_cast<bf16x8_t*>(&As[m][k8]) =
+          *reinterpret_cast<bf16x8_t*>(&As[m][swz(m, k8)]) =
               *reinterpret_cast<const bf16x8_t*>(&z);
         }
       }
@@ -160,12 +167,12 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
           const int m = g / (BK / 8), k8 = (g % (BK / 8)) * 8;
           const long gm = tile_m + m, gk = k0 + k8;
           if (gm < M && gk + 8 <= k_end) {
-            *reinterpret_cast<bf16x8_t*>(&As[m][k8]) =
+            *reinterpret_cast<bf16x8_t*>(&As[m][swz(m, k8)]) =
                 *reinterpret_cast<const bf16x8_t*>(&A[gm * K + gk]);
           } else {
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              As[m][k8 + j] = (gm < M && gk + j < k_end)
+              As[m][swz(m, k8) + j] = (gm < M && gk + j < k_end)
                                   ? A[gm * K + gk + j] : zero;
           }
         } else {    // A[K][M]: 8 consecutive m (same k) per thread
@@ -176,12 +183,13 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
                 *reinterpret_cast<const u16x8_t*>(&A[gk * M + gm]);
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              *reinterpret_cast<unsigned short*>(&As[m8 + j][k]) = v[j];
+              *reinterpret_cast<unsigned short*>(
+                  &As[m8 + j][swz(m8 + j, k)]) = v[j];
           } else {
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              As[m8 + j][k] = (gk < k_end && gm + j < M)
-                                  ? A[gk * M + gm + j] : zero;
+              As[m8 + j][swz(m8 + j, k)] =
+                  (gk < k_end && gm + j < M) ? A[gk * M + gm + j] : zero;
           }
         }
       }
@@ -191,7 +199,7 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         int idx = tid + i * THREADS;
         int m = idx / BK, k = idx % BK;
         long gm = tile_m + m, gk = k0 + k;
-        As[m][k] = (gm < M && gk < k_end)
+        As[m][swz(m, k)] = (gm < M && gk < k_end)
                        ? (TA ? A[gk * M + gm] : A[gm * K + gk]) : zero;
       }
     }
@@ -225,7 +233,8 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<unsigned short*>(&Bs[n8 + j][k]) = v[j];
+          *reinterpret_cast<unsigned short*>(
+              &Bs[n8 + j][swz(n8 + j, k)]) = v[j];
       }
     } else if (vecB) {
       constexpr int GROUPS = (BN * BK) / 8;
@@ -237,12 +246,12 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
           const int n = g / (BK / 8), k8 = (g % (BK / 8)) * 8;
           const long gn = tile_n + n, gk = k0 + k8;
           if (gn < N && gk + 8 <= k_end) {
-            *reinterpret_cast<bf16x8_t*>(&Bs[n][k8]) =
+            *reinterpret_cast<bf16x8_t*>(&Bs[n][swz(n, k8)]) =
                 *reinterpret_cast<const bf16x8_t*>(&B[gn * K + gk]);
           } else {
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              Bs[n][k8 + j] = (gn < N && gk + j < k_end)
+              Bs[n][swz(n, k8) + j] = (gn < N && gk + j < k_end)
                                   ? B[gn * K + gk + j] : zero;
           }
         } else {    // B[K][N]: 8 consecutive n (same k) per thread
@@ -253,12 +262,13 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
                 *reinterpret_cast<const u16x8_t*>(&B[gk * N + gn]);
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              *reinterpret_cast<unsigned short*>(&Bs[n8 + j][k]) = v[j];
+              *reinterpret_cast<unsigned short*>(
+                  &Bs[n8 + j][swz(n8 + j, k)]) = v[j];
           } else {
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              Bs[n8 + j][k] = (gk < k_end && gn + j < N)
-                                  ? B[gk * N + gn + j] : zero;
+              Bs[n8 + j][swz(n8 + j, k)] =
+                  (gk < k_end && gn + j < N) ? B[gk * N + gn + j] : zero;
           }
         }
       }
@@ -268,7 +278,7 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         int idx = tid + i * THREADS;
         int n = idx / BK, k = idx % BK;
         long gn = tile_n + n, gk = k0 + k;
-        Bs[n][k] = (gn < N && gk < k_end)
+        Bs[n][swz(n, k)] = (gn < N && gk < k_end)
                        ? (TB ? B[gn * K + gk] : B[gk * N + gn]) : zero;
       }
     }
@@ -278,11 +288,11 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm)
       a_frag[fm] = *reinterpret_cast<const bf16x8_t*>(
-          &As[wm0 + fm * 16 + l15][l4 * 8]);
+          &As[wm0 + fm * 16 + l15][swz(wm0 + fm * 16 + l15, l4 * 8)]);
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn)
       b_frag[fn] = *reinterpret_cast<const bf16x8_t*>(
-          &Bs[wn0 + fn * 16 + l15][l4 * 8]);
+          &Bs[wn0 + fn * 16 + l15][swz(wn0 + fn * 16 + l15, l4 * 8)]);
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
