@@ -1288,7 +1288,24 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   // N%64 floor: widening to N%8 routed mid shapes onto narrow-BN
   // configs that measured slower than the synchronous path (the edge
   // guards stay for M and the K tail).
-  if (N % 64 == 0 && K % 8 == 0 && M >= 48 && N >= 64 && K >= 32) {
+  // Route wgrad-layout GEMMs (ta, !tb — both operands K-major) straight
+  // to the small kernel's swizzled scatter staging when the two operand
+  // transposes the dbuf path would need outweigh the GEMM itself:
+  // t_transpose ~ 4K(M+N)/5e12 vs t_gemm ~ 2MNK/600e12, i.e. when
+  // 480(M+N) > MN (all wgrad outputs qualify; big square GEMMs do not).
+  // BFLC_WGRAD_SMALL: -1 never, 1 always, unset/0 = this gate.
+  // (A/B on ResNet-50: 81.2 default-off vs 80.7 ms/round routed small.)
+  static const int wgrad_small = [] {
+    const char* e = getenv("BFLC_WGRAD_SMALL");
+    return e ? atoi(e) : 0;
+  }();
+  const bool route_small =
+      ta && !tb &&
+      (wgrad_small == 1 ||
+       (wgrad_small == 0 && 480.0 * (double)(M + N) > (double)M * N));
+  const bool dbuf_ok = !route_small;
+  if (dbuf_ok &&
+      N % 64 == 0 && K % 8 == 0 && M >= 48 && N >= 64 && K >= 32) {
     const long ksteps64 = (K + BK2 - 1) / BK2;
     const int bn2 = (N % 256 == 0) ? 256
                     : (N % 128 == 0 ? 128 : 64);
