@@ -779,21 +779,58 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
 // (out-row, float4 granule) — coalesced, and gives the finalize kernel
 // a bounded chunk count regardless of how many tiles the producing
 // GEMM emitted.
-__global__ void bn_collapse_kernel(const float* __restrict__ in, int chunks,
-                                   int C, float* __restrict__ out) {
+// Collapse [chunks][C] partials to [64][C] for the serial final sweep.
+// Handles BOTH partial arrays (psum/psq) in one launch — the split-in-
+// two version ran 1-4 blocks per launch (64*C/4 threads) and was 11% of
+// a ResNet-20 round — and unrolls the strided chunk walk over 4
+// independent accumulators (fixed association order, so every rank
+// still reduces identically).
+__device__ inline void f4add(float4& a, const float4 v) {
+  a.x += v.x; a.y += v.y; a.z += v.z; a.w += v.w;
+}
+
+__global__ void bn_collapse_kernel(const float* __restrict__ in,
+                                   const float* __restrict__ in2, int chunks,
+                                   int C, float* __restrict__ out,
+                                   float* __restrict__ out2) {
   const int c4g = C / 4;
-  const long total = 64L * c4g;
+  const long per = 64L * c4g;
+  const long total = in2 ? 2 * per : per;
   long t = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (; t < total; t += stride) {
-    const int c4 = (int)(t % c4g) * 4;
-    const int j = (int)(t / c4g);
-    float4 acc = {0.f, 0.f, 0.f, 0.f};
-    for (int k = j; k < chunks; k += 64) {
-      const float4 v = *reinterpret_cast<const float4*>(&in[(long)k * C + c4]);
-      acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+    long u = t;
+    const float* src = in;
+    float* dst = out;
+    if (u >= per) { u -= per; src = in2; dst = out2; }
+    const int c4 = (int)(u % c4g) * 4;
+    const int j = (int)(u / c4g);
+    float4 a0 = {}, a1 = {}, a2 = {}, a3 = {};
+    int k = j;
+    for (; k + 192 < chunks; k += 256) {
+      f4add(a0, *reinterpret_cast<const float4*>(&src[(long)k * C + c4]));
+      f4add(a1, *reinterpret_cast<const float4*>(
+                    &src[(long)(k + 64) * C + c4]));
+      f4add(a2, *reinterpret_cast<const float4*>(
+                    &src[(long)(k + 128) * C + c4]));
+      f4add(a3, *reinterpret_cast<const float4*>(
+                    &src[(long)(k + 192) * C + c4]));
     }
-    *reinterpret_cast<float4*>(&out[(long)j * C + c4]) = acc;
+    if (k < chunks) {
+      f4add(a0, *reinterpret_cast<const float4*>(&src[(long)k * C + c4]));
+      k += 64;
+    }
+    if (k < chunks) {
+      f4add(a1, *reinterpret_cast<const float4*>(&src[(long)k * C + c4]));
+      k += 64;
+    }
+    if (k < chunks) {
+      f4add(a2, *reinterpret_cast<const float4*>(&src[(long)k * C + c4]));
+    }
+    f4add(a0, a1);
+    f4add(a2, a3);
+    f4add(a0, a2);
+    *reinterpret_cast<float4*>(&dst[(long)j * C + c4]) = a0;
   }
 }
 
@@ -857,14 +894,12 @@ std::tuple<torch::Tensor, torch::Tensor> bn_stats_finalize(
   if (chunks > 96 && C % 4 == 0) {  // bound the serial final sweep
     auto cs = torch::empty({64, C}, psum.options());
     auto cq = torch::empty({64, C}, psum.options());
-    const long total = 64L * (C / 4);
+    const long total = 2 * 64L * (C / 4);
     const int blocks = (int)std::min<long>((total + 255) / 256, 4096);
     hipLaunchKernelGGL(bn_collapse_kernel, dim3(blocks), dim3(256), 0,
-                       cur_stream(), psum.data_ptr<float>(), chunks, C,
-                       cs.data_ptr<float>());
-    hipLaunchKernelGGL(bn_collapse_kernel, dim3(blocks), dim3(256), 0,
-                       cur_stream(), psq.data_ptr<float>(), chunks, C,
-                       cq.data_ptr<float>());
+                       cur_stream(), psum.data_ptr<float>(),
+                       psq.data_ptr<float>(), chunks, C,
+                       cs.data_ptr<float>(), cq.data_ptr<float>());
     HIP_CHECK(hipGetLastError());
     psum = cs;
     psq = cq;

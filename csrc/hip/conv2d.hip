@@ -497,25 +497,26 @@ conv2d_fwd_bn(torch::Tensor x, torch::Tensor w, long stride, long pad) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
   auto sh = make_shape(x, w, stride, pad);
   auto w2 = w.view({(long)sh.Kout, sh.RSC()});
-  auto zb = torch::zeros({(long)sh.Kout}, x.options());
   auto y = torch::empty({(long)sh.N, (long)sh.OH, (long)sh.OW,
                          (long)sh.Kout}, x.options());
   std::pair<torch::Tensor, torch::Tensor> stats;
   auto none = torch::empty({0}, x.options());
+  // BN consumes the conv output unbiased: nullptr bias (a zeros tensor
+  // here cost a tiny fill launch per conv per step)
   if (is_1x1_s1(sh)) {
     auto col = x.view({sh.M(), (long)sh.C});
-    gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true, &zb,
-                  false, EpStore::kPlain, 0, &stats);
+    gemm_bf16_raw(col, w2, y, sh.M(), sh.Kout, sh.RSC(), false, true,
+                  nullptr, false, EpStore::kPlain, 0, &stats);
     return {y, col, stats.first.defined() ? stats.first : none,
             stats.second.defined() ? stats.second : none};
   }
-  if (gemm_conv_fwd_raw(x, w2, y, sh, &zb, false, &stats))
+  if (gemm_conv_fwd_raw(x, w2, y, sh, nullptr, false, &stats))
     return {y, none, stats.first.defined() ? stats.first : none,
             stats.second.defined() ? stats.second : none};
   auto col = im2col(x, sh);
   const long kp = col.size(1);
   gemm_bf16_raw(col, pad_w2(w2, kp), y, sh.M(), sh.Kout, kp, false, true,
-                &zb, false, EpStore::kPlain, 0, &stats);
+                nullptr, false, EpStore::kPlain, 0, &stats);
   return {y, col, stats.first.defined() ? stats.first : none,
           stats.second.defined() ? stats.second : none};
 }
