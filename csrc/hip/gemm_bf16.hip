@@ -979,6 +979,44 @@ __global__ void transpose_bf16_kernel(const bf16* __restrict__ in,
 // reduction tree is independent of timing/data, so replicas stay
 // bitwise identical (the old thread-per-element serial loop over up to
 // 512 slices was latency-bound at ~50us/call).
+// Plain-store fast path: one thread per 4 consecutive outputs, serial
+// ascending-s walk (fixed order, deterministic) over the slice streams
+// — all loads are 16-B and there is no LDS tree / barrier pair per 32
+// outputs like the general kernel below. Requires N % 4 == 0 so a
+// float4 never crosses a row (bias indexing stays affine).
+typedef __attribute__((ext_vector_type(4))) unsigned short u16x4_t;
+
+__global__ void splitk_reduce_vec_kernel(const float* __restrict__ Cpart,
+                                         int S, long M, long N,
+                                         bf16* __restrict__ C,
+                                         const bf16* __restrict__ bias,
+                                         int relu) {
+  const long total = M * N;
+  const long total4 = total / 4;
+  long t = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; t < total4; t += stride) {
+    const long i = t * 4;
+    float4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int s = 0; s < S; ++s) {
+      const float4 v =
+          *reinterpret_cast<const float4*>(&Cpart[(long)s * total + i]);
+      acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+    }
+    float vj[4] = {acc.x, acc.y, acc.z, acc.w};
+    const long col = i % N;
+    u16x4_t out;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (bias) vj[j] += b2f(bias[col + j]);
+      if (relu) vj[j] = fmaxf(vj[j], 0.f);
+      const bf16 b = f2b(vj[j]);
+      out[j] = *reinterpret_cast<const unsigned short*>(&b);
+    }
+    *reinterpret_cast<u16x4_t*>(&C[i]) = out;
+  }
+}
+
 __global__ void splitk_reduce_kernel(const float* __restrict__ Cpart, int S,
                                      long M, long N, bf16* __restrict__ C,
                                      const bf16* __restrict__ bias, int relu,
@@ -1011,6 +1049,29 @@ __global__ void splitk_reduce_kernel(const float* __restrict__ Cpart, int S,
     }
     __syncthreads();
   }
+}
+
+void launch_splitk_reduce(const float* part, long S, long M, long N,
+                          bf16* c, const bf16* bias, int relu,
+                          EpStore store, long ohw) {
+  const long total = M * N;
+  // vec path only when the output alone supplies enough threads: the
+  // serial ascending-s walk underfills the chip on the small-output,
+  // many-slice reduces (dw [16][144] with S~128 -> 3 blocks; routing
+  // those here cost ResNet-20 21.0 -> 25.5 ms/round). The 8-lane-tree
+  // kernel keeps 8-way S parallelism for them.
+  if (store == EpStore::kPlain && N % 4 == 0 && total >= 131072) {
+    const long total4 = total / 4;
+    const int blocks = (int)std::min<long>((total4 + 255) / 256, 16384);
+    hipLaunchKernelGGL(splitk_reduce_vec_kernel, dim3(blocks), dim3(256),
+                       0, cur_stream(), part, (int)S, M, N, c, bias, relu);
+  } else {
+    const int blocks = (int)std::min<long>((total + 31) / 32, 16384);
+    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
+                       cur_stream(), part, (int)S, M, N, c, bias, relu,
+                       (int)store, ohw);
+  }
+  HIP_CHECK(hipGetLastError());
 }
 
 // Hierarchical colsum: pass 1 tiles rows into fp32 partials, pass 2
@@ -1371,14 +1432,9 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
                          b2, c, part_ptr, bs, M, N, K, kslice,
                          relu ? 1 : 0, stp, stq);
       HIP_CHECK(hipGetLastError());
-      if (S > 1) {
-        const long total = M * N;
-        int blocks = (int)std::min<long>((total + 31) / 32, 16384);
-        hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256),
-                           0, cur_stream(), part_ptr, (int)S, M, N, c, bs,
-                           relu ? 1 : 0, (int)store, ohw);
-        HIP_CHECK(hipGetLastError());
-      }
+      if (S > 1)
+        launch_splitk_reduce(part_ptr, S, M, N, c, bs, relu ? 1 : 0,
+                             store, ohw);
       return;
     }
     auto launch2 = [&](auto bmv, auto bnv) {
@@ -1406,14 +1462,9 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
       case  64032: launch2(c64i{}, c32i{}); break;
     }
     HIP_CHECK(hipGetLastError());
-    if (S > 1) {
-      const long total = M * N;
-      int blocks = (int)std::min<long>((total + 31) / 32, 16384);
-      hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
-                         cur_stream(), part_ptr, (int)S, M, N, c, bs,
-                         relu ? 1 : 0, (int)store, ohw);
-      HIP_CHECK(hipGetLastError());
-    }
+    if (S > 1)
+      launch_splitk_reduce(part_ptr, S, M, N, c, bs, relu ? 1 : 0, store,
+                           ohw);
     return;
   }
 
@@ -1487,14 +1538,9 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
   }
   HIP_CHECK(hipGetLastError());
 
-  if (S > 1) {
-    const long total = M * N;
-    int blocks = (int)std::min<long>((total + 31) / 32, 16384);
-    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
-                       cur_stream(), part_ptr, (int)S, M, N, c, bs,
-                       relu ? 1 : 0, (int)store, ohw);
-    HIP_CHECK(hipGetLastError());
-  }
+  if (S > 1)
+    launch_splitk_reduce(part_ptr, S, M, N, c, bs, relu ? 1 : 0, store,
+                         ohw);
 }
 
 torch::Tensor colsum_bf16(const torch::Tensor& X) {
@@ -1612,14 +1658,9 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
       default: return false;
     }
     HIP_CHECK(hipGetLastError());
-    if (S > 1) {
-      const long total = M * N;
-      int blocks = (int)std::min<long>((total + 31) / 32, 16384);
-      hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
-                         cur_stream(), part_ptr, (int)S, M, N, c, bs,
-                         relu ? 1 : 0, (int)EpStore::kPlain, 0);
-      HIP_CHECK(hipGetLastError());
-    }
+    if (S > 1)
+      launch_splitk_reduce(part_ptr, S, M, N, c, bs, relu ? 1 : 0,
+                           EpStore::kPlain, 0);
     return true;
   }
   const long ksteps64 = (K + BK2 - 1) / BK2;
@@ -1688,14 +1729,9 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
     case  64064: launchc(c64i{}, c64i{}); break;
   }
   HIP_CHECK(hipGetLastError());
-  if (S > 1) {
-    const long total = M * N;
-    int blocks = (int)std::min<long>((total + 31) / 32, 16384);
-    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
-                       cur_stream(), part_ptr, (int)S, M, N, c, bs,
-                       relu ? 1 : 0, (int)EpStore::kPlain, 0);
-    HIP_CHECK(hipGetLastError());
-  }
+  if (S > 1)
+    launch_splitk_reduce(part_ptr, S, M, N, c, bs, relu ? 1 : 0,
+                         EpStore::kPlain, 0);
   return true;
 }
 
@@ -1763,14 +1799,9 @@ bool gemm_conv_wgrad_raw(const torch::Tensor& dy2, const torch::Tensor& x,
     default: return false;
   }
   HIP_CHECK(hipGetLastError());
-  if (S > 1) {
-    const long total = M * N;
-    int blocks = (int)std::min<long>((total + 31) / 32, 16384);
-    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
-                       cur_stream(), part_ptr, (int)S, M, N, c, nullptr, 0,
-                       (int)EpStore::kPlain, 0);
-    HIP_CHECK(hipGetLastError());
-  }
+  if (S > 1)
+    launch_splitk_reduce(part_ptr, S, M, N, c, nullptr, 0,
+                         EpStore::kPlain, 0);
   return true;
 }
 
