@@ -109,6 +109,7 @@ class FLEngine:
         # Adam, or if capture fails on this ROCm build.
         import os
         self._stepper = None
+        self._scorers: Dict[int, object] = {}
         self._use_graphs = (cfg.use_graphs and self.device.type == "cuda"
                             and cfg.optimizer == "sgd"
                             and os.environ.get("BFLC_GRAPHS", "1") != "0")
@@ -207,15 +208,36 @@ class FLEngine:
         candidate = W0 - lr*delta, scored by accuracy on the scorer's own
         local shard."""
         shard = self.shards[scorer]
+        gs = self._graphed_scorer(scorer) if self._use_graphs else None
         accs: List[Tuple[str, torch.Tensor]] = []
         for origin, delta in updates:
             cand = self.global_flat.clone()
             O.axpy_(cand, -self.cfg.learning_rate, delta)
-            self.model.set_flat(cand)
-            # device-resident: ONE host sync per scorer (below), not
-            # one per candidate
-            accs.append((origin, self.model.accuracy_t(shard.x, shard.y)))
+            if gs is not None:
+                accs.append((origin, gs.score(cand)))
+            else:
+                self.model.set_flat(cand)
+                # device-resident: ONE host sync per scorer (below),
+                # not one per candidate
+                accs.append((origin,
+                             self.model.accuracy_t(shard.x, shard.y)))
         return {origin: float(a) for origin, a in accs}
+
+    def _graphed_scorer(self, scorer: int):
+        """Per-scorer captured scoring graph (fl/graphs.py); capture
+        mutates the scratch model, which every phase reloads anyway."""
+        if scorer in self._scorers:
+            return self._scorers[scorer]
+        try:
+            from bflc_amd.fl.graphs import GraphedScore
+            shard = self.shards[scorer]
+            self._scorers[scorer] = GraphedScore(self.model, shard.x,
+                                                 shard.y)
+        except Exception as e:
+            import warnings
+            warnings.warn(f"score-graph capture failed, running eager: {e}")
+            self._scorers[scorer] = None
+        return self._scorers[scorer]
 
     # ------------------------------------------------------------------
     def run_round(self, eval_global: bool = False) -> RoundStats:

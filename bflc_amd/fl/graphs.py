@@ -82,3 +82,35 @@ class GraphedTrainStep:
         self.sx.copy_(xb)
         self.sy.copy_(yb)
         self.graph.replay()
+
+
+class GraphedScore:
+    """Captured committee-scoring evaluation: load a candidate flat
+    vector into the model (copy + shadow refresh) and run accuracy over
+    the scorer's own shard — the shard tensors are baked into the graph
+    (they never change), only the candidate is copied in per replay.
+    The per-round scoring cost scales with the number of admitted
+    candidates (committee size x quota at 8 nodes), so this is the
+    phase that grows with world size."""
+
+    def __init__(self, model, shard_x: torch.Tensor,
+                 shard_y: torch.Tensor) -> None:
+        self.model = model
+        self.cand = torch.empty_like(model.flat)
+        self.cand.copy_(model.flat.detach())
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            model.set_flat(self.cand)
+            model.accuracy_t(shard_x, shard_y)
+        torch.cuda.current_stream().wait_stream(side)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            model.set_flat(self.cand)
+            self.acc = model.accuracy_t(shard_x, shard_y)
+
+    def score(self, cand: torch.Tensor) -> torch.Tensor:
+        """Returns a device scalar snapshot (no host sync)."""
+        self.cand.copy_(cand)
+        self.graph.replay()
+        return self.acc.clone()
