@@ -107,46 +107,95 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
 
   f32x4_t acc[FM][FN] = {};
 
+  // Implicit-gather m-side decode hoisted out of the K loop: each
+  // staging slot's output row gm is FIXED across k-steps, so its
+  // (n, oh/ow or ih/iw) divides and the row base/offsets are computed
+  // once here instead of every k-step (they were ~half the staging
+  // instruction count on the ResNet-20 implicit kernels).
+  constexpr int AGI =
+      (CMODE == 1 || CMODE == 2)
+          ? ((BM * BK / 8) + THREADS - 1) / THREADS : 1;
+  long cv_nbase[AGI];
+  int cv_i0[AGI], cv_j0[AGI];
+  bool cv_mok[AGI];
+  if (CMODE == 1 || CMODE == 2) {
+    constexpr int GROUPS = (BM * BK) / 8;
+#pragma unroll
+    for (int i = 0; i < AGI; ++i) {
+      const int g = tid + i * THREADS;
+      const int m = g / (BK / 8);
+      const long gm = tile_m + m;
+      cv_mok[i] = g < GROUPS && gm < M;
+      const long gmc = cv_mok[i] ? gm : 0;
+      if (CMODE == 1) {
+        const int ow = (int)(gmc % csh.OW);
+        const int oh = (int)((gmc / csh.OW) % csh.OH);
+        const int nn = (int)(gmc / ((long)csh.OW * csh.OH));
+        cv_i0[i] = oh * csh.stride - csh.pad;  // ih0
+        cv_j0[i] = ow * csh.stride - csh.pad;  // iw0
+        cv_nbase[i] = (long)nn * csh.H;
+      } else {
+        const int iw = (int)(gmc % csh.W);
+        const int ih = (int)((gmc / csh.W) % csh.H);
+        const int nn = (int)(gmc / ((long)csh.W * csh.H));
+        cv_i0[i] = ih + csh.pad;  // oh_num0
+        cv_j0[i] = iw + csh.pad;  // ow_num0
+        cv_nbase[i] = (long)nn * csh.OH;
+      }
+    }
+  }
+
+  // CMODE 3: n'-side (r, s, c8) decode per staging slot, k-invariant
+  constexpr int BGI =
+      CMODE == 3 ? ((BN * BK / 8) + THREADS - 1) / THREADS : 1;
+  int cv3_rr[BGI], cv3_ss[BGI], cv3_c8[BGI];
+  if (CMODE == 3) {
+#pragma unroll
+    for (int i = 0; i < BGI; ++i) {
+      const int g = tid + i * THREADS;
+      const int n8 = (g % (BN / 8)) * 8;
+      const long gn = tile_n + n8;
+      const long gnc = gn + 8 <= N ? gn : 0;
+      const int rs = (int)(gnc / csh.C);
+      cv3_c8[i] = (int)(gnc - (long)rs * csh.C);
+      cv3_rr[i] = rs / csh.S;
+      cv3_ss[i] = rs - cv3_rr[i] * csh.S;
+    }
+  }
+
   const bf16 zero = f2b(0.f);
   for (long k0 = k_begin; k0 < k_end; k0 += BK) {
     // ---- stage A tile (16-byte vector path when layout permits) ----
     if (CMODE == 1 || CMODE == 2) {  // implicit A gather (x / dy)
       constexpr int GROUPS = (BM * BK) / 8;
 #pragma unroll
-      for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
+      for (int i = 0; i < AGI; ++i) {
         const int g = tid + i * THREADS;
         if (GROUPS % THREADS != 0 && g >= GROUPS) break;
         const int m = g / (BK / 8), k8 = (g % (BK / 8)) * 8;
-        const long gm = tile_m + m;
         const long gk = k0 + k8;
-        bool ok = gm < M && gk < k_end;
+        bool ok = cv_mok[i] && gk < k_end;
         long src = 0;
         if (ok && CMODE == 1) {  // fwd: x gather
           const int rs = (int)(gk / csh.C);
           const int c8 = (int)(gk - (long)rs * csh.C);
           const int rr = rs / csh.S, ss = rs - rr * csh.S;
-          const int ow = (int)(gm % csh.OW);
-          const int oh = (int)((gm / csh.OW) % csh.OH);
-          const int nn = (int)(gm / ((long)csh.OW * csh.OH));
-          const int ih = oh * csh.stride - csh.pad + rr;
-          const int iw = ow * csh.stride - csh.pad + ss;
+          const int ih = cv_i0[i] + rr;
+          const int iw = cv_j0[i] + ss;
           ok = ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W;
-          src = (((long)nn * csh.H + ih) * csh.W + iw) * csh.C + c8;
+          src = ((cv_nbase[i] + ih) * csh.W + iw) * csh.C + c8;
         } else if (ok) {  // dgrad: dy gather, flipped correlation
           const int rs = (int)(gk / csh.Kout);
           const int kk8 = (int)(gk - (long)rs * csh.Kout);
           const int rr = rs / csh.S, ss = rs - rr * csh.S;
-          const int iw = (int)(gm % csh.W);
-          const int ih = (int)((gm / csh.W) % csh.H);
-          const int nn = (int)(gm / ((long)csh.W * csh.H));
-          const int oh_num = ih + csh.pad - rr;
-          const int ow_num = iw + csh.pad - ss;
+          const int oh_num = cv_i0[i] - rr;
+          const int ow_num = cv_j0[i] - ss;
           const int oh = oh_num / csh.stride;
           const int ow = ow_num / csh.stride;
           ok = oh_num >= 0 && ow_num >= 0 &&
                oh_num % csh.stride == 0 && ow_num % csh.stride == 0 &&
                oh < csh.OH && ow < csh.OW;
-          src = (((long)nn * csh.OH + oh) * csh.OW + ow) * csh.Kout + kk8;
+          src = ((cv_nbase[i] + oh) * csh.OW + ow) * csh.Kout + kk8;
         }
         if (ok) {
           *reinterpret_cast<bf16x8_t*>(&As[m][swz(m, k8)]) =
@@ -207,7 +256,9 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
     if (CMODE == 3) {
       // wgrad implicit col: B[k'=m][n'=(r,s,c)]; a granule is 8
       // consecutive c at fixed (m, r, s) -> one contiguous x load,
-      // scatter-written like the vecB !TB path (col never exists)
+      // scatter-written like the vecB !TB path (col never exists).
+      // The n'-side (r, s, c8) decode is fixed per slot across
+      // k-steps; only the m decode varies.
       constexpr int GROUPS = (BN * BK) / 8;
 #pragma unroll
       for (int i = 0; i < (GROUPS + THREADS - 1) / THREADS; ++i) {
@@ -219,17 +270,15 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         bool ok = gm < k_end && gn + 8 <= N;
         u16x8_t v = {};
         if (ok) {
-          const int rs = (int)(gn / csh.C);
-          const int c8 = (int)(gn - (long)rs * csh.C);
-          const int rr = rs / csh.S, ss = rs - rr * csh.S;
           const int ow = (int)(gm % csh.OW);
           const int oh = (int)((gm / csh.OW) % csh.OH);
           const int nn = (int)(gm / ((long)csh.OW * csh.OH));
-          const int ih = oh * csh.stride - csh.pad + rr;
-          const int iw = ow * csh.stride - csh.pad + ss;
+          const int ih = oh * csh.stride - csh.pad + cv3_rr[i];
+          const int iw = ow * csh.stride - csh.pad + cv3_ss[i];
           if (ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W)
             v = *reinterpret_cast<const u16x8_t*>(
-                &B[(((long)nn * csh.H + ih) * csh.W + iw) * csh.C + c8]);
+                &B[(((long)nn * csh.H + ih) * csh.W + iw) * csh.C +
+                   cv3_c8[i]]);
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j)
