@@ -118,12 +118,17 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
   long cv_nbase[AGI];
   int cv_i0[AGI], cv_j0[AGI];
   bool cv_mok[AGI];
+  // k-side decode kept as incremental per-slot state: gk advances by
+  // BK each step, so (c8, r, s) update with adds and wraps — no
+  // divides left anywhere in the staging loop.
+  int st_c8[AGI], st_rr[AGI], st_ss[AGI];
   if (CMODE == 1 || CMODE == 2) {
     constexpr int GROUPS = (BM * BK) / 8;
 #pragma unroll
     for (int i = 0; i < AGI; ++i) {
       const int g = tid + i * THREADS;
       const int m = g / (BK / 8);
+      const int k8 = (g % (BK / 8)) * 8;
       const long gm = tile_m + m;
       cv_mok[i] = g < GROUPS && gm < M;
       const long gmc = cv_mok[i] ? gm : 0;
@@ -142,6 +147,12 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         cv_j0[i] = iw + csh.pad;  // ow_num0
         cv_nbase[i] = (long)nn * csh.OH;
       }
+      const int inner = CMODE == 1 ? csh.C : csh.Kout;
+      const long gk0 = k_begin + k8;
+      const int rs = (int)(gk0 / inner);
+      st_c8[i] = (int)(gk0 - (long)rs * inner);
+      st_rr[i] = rs / csh.S;
+      st_ss[i] = rs - st_rr[i] * csh.S;
     }
   }
 
@@ -177,25 +188,19 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         bool ok = cv_mok[i] && gk < k_end;
         long src = 0;
         if (ok && CMODE == 1) {  // fwd: x gather
-          const int rs = (int)(gk / csh.C);
-          const int c8 = (int)(gk - (long)rs * csh.C);
-          const int rr = rs / csh.S, ss = rs - rr * csh.S;
-          const int ih = cv_i0[i] + rr;
-          const int iw = cv_j0[i] + ss;
+          const int ih = cv_i0[i] + st_rr[i];
+          const int iw = cv_j0[i] + st_ss[i];
           ok = ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W;
-          src = ((cv_nbase[i] + ih) * csh.W + iw) * csh.C + c8;
+          src = ((cv_nbase[i] + ih) * csh.W + iw) * csh.C + st_c8[i];
         } else if (ok) {  // dgrad: dy gather, flipped correlation
-          const int rs = (int)(gk / csh.Kout);
-          const int kk8 = (int)(gk - (long)rs * csh.Kout);
-          const int rr = rs / csh.S, ss = rs - rr * csh.S;
-          const int oh_num = cv_i0[i] - rr;
-          const int ow_num = cv_j0[i] - ss;
+          const int oh_num = cv_i0[i] - st_rr[i];
+          const int ow_num = cv_j0[i] - st_ss[i];
           const int oh = oh_num / csh.stride;
           const int ow = ow_num / csh.stride;
           ok = oh_num >= 0 && ow_num >= 0 &&
                oh_num % csh.stride == 0 && ow_num % csh.stride == 0 &&
                oh < csh.OH && ow < csh.OW;
-          src = ((cv_nbase[i] + oh) * csh.OW + ow) * csh.Kout + kk8;
+          src = ((cv_nbase[i] + oh) * csh.OW + ow) * csh.Kout + st_c8[i];
         }
         if (ok) {
           *reinterpret_cast<bf16x8_t*>(&As[m][swz(m, k8)]) =
@@ -204,6 +209,15 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
           u16x8_t z = {};
           *reinterpret_cast<bf16x8_t*>(&As[m][swz(m, k8)]) =
               *reinterpret_cast<const bf16x8_t*>(&z);
+        }
+        {  // advance (c8, s, r) by BK along the gather row
+          const int inner = CMODE == 1 ? csh.C : csh.Kout;
+          int c8n = st_c8[i] + BK;
+          while (c8n >= inner) {
+            c8n -= inner;
+            if (++st_ss[i] == csh.S) { st_ss[i] = 0; ++st_rr[i]; }
+          }
+          st_c8[i] = c8n;
         }
       }
     } else if (vecA) {
