@@ -156,10 +156,14 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
     }
   }
 
-  // CMODE 3: n'-side (r, s, c8) decode per staging slot, k-invariant
+  // CMODE 3: n'-side (r, s, c8) decode per staging slot, k-invariant;
+  // the m-side (n, oh, ow) decode advances by BK each k-step, so it is
+  // carried incrementally too — no divides in the staging loop.
   constexpr int BGI =
       CMODE == 3 ? ((BN * BK / 8) + THREADS - 1) / THREADS : 1;
   int cv3_rr[BGI], cv3_ss[BGI], cv3_c8[BGI];
+  int st3_ow[BGI], st3_oh[BGI];
+  long st3_nb[BGI];
   if (CMODE == 3) {
 #pragma unroll
     for (int i = 0; i < BGI; ++i) {
@@ -171,6 +175,11 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
       cv3_c8[i] = (int)(gnc - (long)rs * csh.C);
       cv3_rr[i] = rs / csh.S;
       cv3_ss[i] = rs - cv3_rr[i] * csh.S;
+      const int k = g / (BN / 8);
+      const long gm0 = k_begin + k;
+      st3_ow[i] = (int)(gm0 % csh.OW);
+      st3_oh[i] = (int)((gm0 / csh.OW) % csh.OH);
+      st3_nb[i] = (long)(gm0 / ((long)csh.OW * csh.OH)) * csh.H;
     }
   }
 
@@ -284,20 +293,24 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
         bool ok = gm < k_end && gn + 8 <= N;
         u16x8_t v = {};
         if (ok) {
-          const int ow = (int)(gm % csh.OW);
-          const int oh = (int)((gm / csh.OW) % csh.OH);
-          const int nn = (int)(gm / ((long)csh.OW * csh.OH));
-          const int ih = oh * csh.stride - csh.pad + cv3_rr[i];
-          const int iw = ow * csh.stride - csh.pad + cv3_ss[i];
+          const int ih = st3_oh[i] * csh.stride - csh.pad + cv3_rr[i];
+          const int iw = st3_ow[i] * csh.stride - csh.pad + cv3_ss[i];
           if (ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W)
             v = *reinterpret_cast<const u16x8_t*>(
-                &B[(((long)nn * csh.H + ih) * csh.W + iw) * csh.C +
-                   cv3_c8[i]]);
+                &B[((st3_nb[i] + ih) * csh.W + iw) * csh.C + cv3_c8[i]]);
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           *reinterpret_cast<unsigned short*>(
               &Bs[n8 + j][swz(n8 + j, k)]) = v[j];
+        {  // advance m = (n, oh, ow) by BK
+          int ow = st3_ow[i] + BK;
+          while (ow >= csh.OW) {
+            ow -= csh.OW;
+            if (++st3_oh[i] == csh.OH) { st3_oh[i] = 0; st3_nb[i] += csh.H; }
+          }
+          st3_ow[i] = ow;
+        }
       }
     } else if (vecB) {
       constexpr int GROUPS = (BN * BK) / 8;
