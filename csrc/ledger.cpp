@@ -31,6 +31,8 @@
 #include <pybind11/stl.h>
 
 #include <algorithm>
+#include <cstdlib>
+#include <iostream>
 #include <map>
 #include <optional>
 #include <stdexcept>
@@ -75,6 +77,19 @@ struct AggregationDecision {
   std::vector<std::string> next_committee;  // top comm_count scored trainers
   std::map<std::string, double> median_scores;  // per-trainer medians
 };
+
+// Protocol trace, the reference's OUTPUT-gated log lines
+// (CommitteePrecompiled.h:4, .cpp:240-243,255-257,291-293,422-425):
+// BFLC_LEDGER_TRACE=1 prints the same progress markers to stderr so an
+// operator can eyeball replicas the way the reference README documents
+// (tail-ing 4 chain-node logs).
+inline bool ledger_trace() {
+  static const bool on = [] {
+    const char* e = std::getenv("BFLC_LEDGER_TRACE");
+    return e && std::atoi(e) != 0;
+  }();
+  return on;
+}
 
 // Median with the exact semantics of GetMid (CommitteePrecompiled.cpp:81-115):
 // sorted ascending, odd n -> v[n/2]; even n -> (v[n/2-1] + v[n/2]) / 2.
@@ -140,6 +155,10 @@ class CommitteeLedger {
     ++update_count_;
     local_updates_[origin] = Update{blob, n_samples, avg_cost};
     update_order_.push_back(origin);
+    if (ledger_trace())
+      std::clog << "the update of local model is collected! ("
+                << update_count_ << "/" << cfg_.needed_update_count
+                << ", epoch " << epoch_ << ", " << origin << ")\n";
     return Admit::kAccepted;
   }
 
@@ -171,6 +190,9 @@ class CommitteeLedger {
     const bool existed = local_scores_.count(origin) > 0;
     local_scores_[origin] = scores;
     if (!existed) ++score_count_;
+    if (ledger_trace())
+      std::clog << score_count_ << " scores has been uploaded (epoch "
+                << epoch_ << ")\n";
     if (score_count_ == cfg_.comm_count) {
       pending_ = decide_aggregation();
       return pending_;
@@ -185,6 +207,9 @@ class CommitteeLedger {
     if (!pending_) throw std::logic_error("no pending aggregation");
     global_model_ = new_global_blob;
     global_loss_ = pending_->avg_cost;
+    if (ledger_trace())
+      std::clog << "the " << epoch_ << " epoch , global loss : "
+                << global_loss_ << "\n";
     epoch_ += 1;
     local_updates_.clear();
     update_order_.clear();
