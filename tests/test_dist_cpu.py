@@ -25,9 +25,10 @@ from bflc_amd.comm import Transport
 from bflc_amd.data import make_federated
 from bflc_amd.fl import FLEngine
 
-cfg = FLConfig.for_world(2, model="mlp", n_features=784, n_class=10,
+cfg = FLConfig.for_world({world}, model="mlp", n_features=784, n_class=10,
                          samples_per_client=128, batch_size=32,
-                         eval_samples=256)
+                         eval_samples=256, partition={partition!r},
+                         byzantine_clients={byz})
 shards, test = make_federated(cfg)
 t = Transport(backend="gloo", device=torch.device("cpu"))
 eng = FLEngine(cfg, t, shards, test)
@@ -47,15 +48,17 @@ t.close()
 """
 
 
-def run_world2(tmp_path, rounds=3):
+def run_world2(tmp_path, rounds=3, world=2, partition="iid", byz=0,
+               port="29541"):
     script = tmp_path / "worker.py"
     script.write_text(WORKER.format(repo=REPO, rounds=rounds,
-                                    outdir=str(tmp_path)))
+                                    outdir=str(tmp_path), world=world,
+                                    partition=partition, byz=byz))
     procs = []
-    for rank in range(2):
+    for rank in range(world):
         env = dict(os.environ,
-                   RANK=str(rank), WORLD_SIZE="2",
-                   MASTER_ADDR="127.0.0.1", MASTER_PORT="29541",
+                   RANK=str(rank), WORLD_SIZE=str(world),
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT=port,
                    OMP_NUM_THREADS="2")
         procs.append(subprocess.Popen(
             [sys.executable, str(script)], env=env,
@@ -63,7 +66,8 @@ def run_world2(tmp_path, rounds=3):
     for p in procs:
         out, err = p.communicate(timeout=300)
         assert p.returncode == 0, err.decode()[-3000:]
-    return [json.load(open(tmp_path / f"rank{r}.json")) for r in range(2)]
+    return [json.load(open(tmp_path / f"rank{r}.json"))
+            for r in range(world)]
 
 
 def test_world2_replicas_identical(tmp_path):
@@ -93,3 +97,14 @@ def test_world2_matches_single_process(tmp_path):
     eng.run(3)
     assert torch.sum(eng.global_flat.double()).item() == r0["digest"]
     assert eng.global_flat[:8].tolist() == r0["flat0"]
+
+
+def test_world4_byzantine_replicas_identical(tmp_path):
+    """4 ranks, non-IID shards, one label-flip attacker: the committee
+    protocol must stay live and every replica must
+    agree bitwise despite the hostile update."""
+    res = run_world2(tmp_path, rounds=4, world=4, partition="dirichlet",
+                     byz=1, port="29551")
+    assert all(r["epoch"] == 4 for r in res)
+    assert all(r["digest"] == res[0]["digest"] for r in res)
+    assert all(r["roles"] == res[0]["roles"] for r in res)
