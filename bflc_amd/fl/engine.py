@@ -28,7 +28,9 @@ ABI-compatible JSON blob path lives in bflc_amd.chain.client).
 """
 from __future__ import annotations
 
+import os
 import time
+import warnings
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
@@ -107,7 +109,6 @@ class FLEngine:
         # captured kernel DAG per minibatch instead of relaunching
         # ~dozens of kernels. SGD-only; falls back to eager on CPU, on
         # Adam, or if capture fails on this ROCm build.
-        import os
         self._stepper = None
         self._scorers: Dict[int, object] = {}
         self._use_graphs = (cfg.use_graphs and self.device.type == "cuda"
@@ -144,7 +145,6 @@ class FLEngine:
                 self.model, self.cfg.learning_rate, shard.x[:bs],
                 shard.y[:bs])
         except Exception as e:  # capture unsupported: eager fallback
-            import warnings
             warnings.warn(f"hipGraph capture failed, running eager: {e}")
             self._use_graphs = False
             return None
@@ -234,7 +234,6 @@ class FLEngine:
             self._scorers[scorer] = GraphedScore(self.model, shard.x,
                                                  shard.y)
         except Exception as e:
-            import warnings
             warnings.warn(f"score-graph capture failed, running eager: {e}")
             self._scorers[scorer] = None
         return self._scorers[scorer]
