@@ -67,6 +67,12 @@ def main() -> None:
     spc = args.samples_per_client or md["spc"]
     bs = args.batch_size or md["bs"]
 
+    if "WORLD_SIZE" not in os.environ and args.gpus > 1:
+        raise SystemExit(
+            "bench.py: --gpus N>1 must run under torchrun (one process "
+            "per GPU over RCCL), e.g.\n  python -m torch.distributed.run "
+            f"--nnodes=1 --nproc-per-node {args.gpus} "
+            f"--master-addr 127.0.0.1 bench.py --gpus {args.gpus} ...")
     n = int(os.environ.get("WORLD_SIZE", args.gpus))
     n_clients = args.clients or n
     cfg = FLConfig.for_world(
