@@ -119,3 +119,19 @@ class TestNonIID:
                                  eval_samples=128)
         eng, stats, acc = run_engine(cfg, rounds=2)
         assert stats[-1].epoch == 1
+
+
+class TestClientAssignment:
+    def test_uneven_split_covers_all_clients(self):
+        from bflc_amd.fl.engine import client_rank
+        for n_clients in (1, 3, 7, 20):
+            for world in (1, 2, 3, 8):
+                owners = [client_rank(i, n_clients, world)
+                          for i in range(n_clients)]
+                # every client owned by exactly one valid rank,
+                # contiguous ascending (the property the deterministic
+                # admission order relies on)
+                assert all(0 <= r < world for r in owners)
+                assert owners == sorted(owners)
+                if n_clients >= world:
+                    assert set(owners) == set(range(world))

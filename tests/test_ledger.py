@@ -224,3 +224,31 @@ class TestSelfScoringWorld1:
         assert dec is not None and dec.selected == [("node_0", 10)]
         led.commit_aggregate(b"")
         assert led.epoch == 1
+
+
+class TestProtocolTrace:
+    def test_reference_log_lines(self):
+        """BFLC_LEDGER_TRACE=1 reprints the reference's OUTPUT-gated
+        markers (CommitteePrecompiled.h:4, .cpp:255-257,291-293,
+        422-425) from the replica."""
+        import subprocess
+        import sys
+        code = (
+            "import os; os.environ['BFLC_LEDGER_TRACE']='1';"
+            "from bflc_amd._ledger import CommitteeLedger;"
+            "from bflc_amd.config import FLConfig;"
+            "cfg=FLConfig(client_num=4,comm_count=1,needed_update_count=2,"
+            "aggregate_count=1);"
+            "led=CommitteeLedger(cfg.ledger_config());"
+            "[led.register_node(f'node_{i}') for i in range(4)];"
+            "led.set_global_model(b'');"
+            "led.upload_local_update('node_1',b'',0,10,0.5);"
+            "led.upload_local_update('node_2',b'',0,10,0.4);"
+            "led.upload_scores('node_0',0,{'node_1':0.9,'node_2':0.8});"
+            "led.commit_aggregate(b'')")
+        r = subprocess.run([sys.executable, "-c", code],
+                           capture_output=True, text=True, timeout=120)
+        assert r.returncode == 0, r.stderr
+        assert "the update of local model is collected!" in r.stderr
+        assert "scores has been uploaded" in r.stderr
+        assert "epoch , global loss :" in r.stderr
