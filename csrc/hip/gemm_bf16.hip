@@ -1327,15 +1327,21 @@ TileCfg pick_tile(long M, long N) {
 // reads, conflict-free), vectorized A reads and C writes — is bound by
 // the A/C streams instead. KT is compile-time so the A row and the dot
 // fully unroll in registers.
+typedef __attribute__((ext_vector_type(8))) float f32x8v_t;
+
 template <int KT>
 __global__ __launch_bounds__(256) void gemm_thin_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     bf16* __restrict__ C, const bf16* __restrict__ bias, long M, int N,
     int relu) {
-  __shared__ float Bl[64 * 32];
+  // B panel TRANSPOSED in LDS ([k][n], n contiguous) so the dot runs as
+  // 8-wide vector FMAs per k with a broadcast a[k]: the [n][k] row-dot
+  // form compiled to 125 v_mov per 64 v_pk_fma (operand shuffles around
+  // the packed FMAs — 2/3 of the VALU slots).
+  __shared__ float Bl[32 * 64];
   __shared__ float bl[64];
   for (int i = threadIdx.x; i < N * KT; i += blockDim.x)
-    Bl[i] = b2f(B[i]);
+    Bl[(i % KT) * 64 + i / KT] = b2f(B[i]);
   for (int i = threadIdx.x; i < N; i += blockDim.x)
     bl[i] = bias ? b2f(bias[i]) : 0.f;
   __syncthreads();
@@ -1355,15 +1361,19 @@ __global__ __launch_bounds__(256) void gemm_thin_kernel(
       for (int j = 0; j < 8; ++j) a[k8 + j] = b2f(v[j]);
     }
     for (int n8 = 0; n8 < N; n8 += 8) {
+      f32x8v_t acc = *reinterpret_cast<const f32x8v_t*>(&bl[n8]);
+#pragma unroll
+      for (int k = 0; k < KT; ++k) {
+        const f32x8v_t bv =
+            *reinterpret_cast<const f32x8v_t*>(&Bl[k * 64 + n8]);
+        acc += bv * a[k];
+      }
       bf16x8_t out;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float acc = bl[n8 + j];
-        const float* brow = &Bl[(n8 + j) * KT];
-#pragma unroll
-        for (int k = 0; k < KT; ++k) acc += a[k] * brow[k];
-        if (relu && acc < 0.f) acc = 0.f;
-        out[j] = f2b(acc);
+        float v = acc[j];
+        if (relu && v < 0.f) v = 0.f;
+        out[j] = f2b(v);
       }
       *reinterpret_cast<bf16x8_t*>(&C[m * N + n8]) = out;
     }
