@@ -50,6 +50,7 @@ namespace {
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
 typedef __attribute__((ext_vector_type(8))) unsigned short u16x8_t;
+typedef __attribute__((ext_vector_type(2))) unsigned short u16x2_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 constexpr int BK = 32;      // K per MFMA instruction / per LDS stage
@@ -156,15 +157,19 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
     }
   }
 
-  // CMODE 3: n'-side (r, s, c8) decode per staging slot, k-invariant;
+  // CMODE 3/4: n'-side (r, s, c8) decode per staging slot, k-invariant;
   // the m-side (n, oh, ow) decode advances by BK each k-step, so it is
   // carried incrementally too — no divides in the staging loop.
+  // CMODE 4 = the k-pair experiment (BFLC_WGRAD_KPAIR): each slot owns
+  // TWO adjacent k's of one n-granule so the LDS writes are b32 —
+  // half the write instructions at the same (4-way) conflict degree.
   constexpr int BGI =
-      CMODE == 3 ? ((BN * BK / 8) + THREADS - 1) / THREADS : 1;
+      CMODE == 3 ? ((BN * BK / 8) + THREADS - 1) / THREADS
+      : (CMODE == 4 ? ((BN * BK / 16) + THREADS - 1) / THREADS : 1);
   int cv3_rr[BGI], cv3_ss[BGI], cv3_c8[BGI];
   int st3_ow[BGI], st3_oh[BGI];
   long st3_nb[BGI];
-  if (CMODE == 3) {
+  if (CMODE == 3 || CMODE == 4) {
 #pragma unroll
     for (int i = 0; i < BGI; ++i) {
       const int g = tid + i * THREADS;
@@ -175,7 +180,7 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
       cv3_c8[i] = (int)(gnc - (long)rs * csh.C);
       cv3_rr[i] = rs / csh.S;
       cv3_ss[i] = rs - cv3_rr[i] * csh.S;
-      const int k = g / (BN / 8);
+      const int k = (CMODE == 4 ? 2 : 1) * (g / (BN / 8));
       const long gm0 = k_begin + k;
       st3_ow[i] = (int)(gm0 % csh.OW);
       st3_oh[i] = (int)((gm0 / csh.OW) % csh.OH);
@@ -276,7 +281,56 @@ __global__ void gemm_kernel(const bf16* __restrict__ A,
       }
     }
     // ---- stage B tile ----
-    if (CMODE == 3) {
+    if (CMODE == 4) {  // k-pair wgrad staging (b32 LDS writes)
+      constexpr int GROUPS = (BN * BK) / 16;
+#pragma unroll
+      for (int i = 0; i < BGI; ++i) {
+        const int g = tid + i * THREADS;
+        if (GROUPS % THREADS != 0 && g >= GROUPS) break;
+        const int k2 = g / (BN / 8), n8 = (g % (BN / 8)) * 8;
+        const int k = 2 * k2;
+        const long gn = tile_n + n8;
+        const long gm0 = k0 + k;
+        const bool okn = gn + 8 <= N;
+        u16x8_t v0 = {}, v1 = {};
+        if (okn && gm0 < k_end) {
+          const int ih = st3_oh[i] * csh.stride - csh.pad + cv3_rr[i];
+          const int iw = st3_ow[i] * csh.stride - csh.pad + cv3_ss[i];
+          if (ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W)
+            v0 = *reinterpret_cast<const u16x8_t*>(
+                &B[((st3_nb[i] + ih) * csh.W + iw) * csh.C + cv3_c8[i]]);
+        }
+        if (okn && gm0 + 1 < k_end) {
+          int ow1 = st3_ow[i] + 1, oh1 = st3_oh[i];
+          long nb1 = st3_nb[i];
+          if (ow1 == csh.OW) {
+            ow1 = 0;
+            if (++oh1 == csh.OH) { oh1 = 0; nb1 += csh.H; }
+          }
+          const int ih = oh1 * csh.stride - csh.pad + cv3_rr[i];
+          const int iw = ow1 * csh.stride - csh.pad + cv3_ss[i];
+          if (ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W)
+            v1 = *reinterpret_cast<const u16x8_t*>(
+                &B[((nb1 + ih) * csh.W + iw) * csh.C + cv3_c8[i]]);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          u16x2_t w2v;
+          w2v[0] = v0[j];
+          w2v[1] = v1[j];
+          *reinterpret_cast<u16x2_t*>(
+              &Bs[n8 + j][swz(n8 + j, k)]) = w2v;
+        }
+        {  // advance m pair by BK
+          int ow = st3_ow[i] + BK;
+          while (ow >= csh.OW) {
+            ow -= csh.OW;
+            if (++st3_oh[i] == csh.OH) { st3_oh[i] = 0; st3_nb[i] += csh.H; }
+          }
+          st3_ow[i] = ow;
+        }
+      }
+    } else if (CMODE == 3) {
       // wgrad implicit col: B[k'=m][n'=(r,s,c)]; a granule is 8
       // consecutive c at fixed (m, r, s) -> one contiguous x load,
       // scatter-written like the vecB !TB path (col never exists).
@@ -1857,14 +1911,26 @@ bool gemm_conv_wgrad_raw(const torch::Tensor& dy2, const torch::Tensor& x,
     part_ptr = part.data_ptr<float>();
   }
   dim3 grid((unsigned)tiles, (unsigned)S);
+  // BFLC_WGRAD_KPAIR=1: k-pair staging experiment (CMODE 4), default
+  // off until GPU-measured — values are bit-identical either way.
+  static const bool kpair = [] {
+    const char* e = getenv("BFLC_WGRAD_KPAIR");
+    return e && atoi(e) != 0;
+  }();
   auto launchw = [&](auto bm, auto bn, auto wr, auto wc) {
     constexpr int BMv = decltype(bm)::value, BNv = decltype(bn)::value;
     constexpr int WRv = decltype(wr)::value, WCv = decltype(wc)::value;
     dim3 block(WRv * WCv * 64);
-    hipLaunchKernelGGL(
-        (gemm_kernel<BMv, BNv, WRv, WCv, true, false, 3>), grid, block, 0,
-        cur_stream(), a, b, c, part_ptr, nullptr, M, N, K, kslice, 0,
-        (int)EpStore::kPlain, 0, 1, 0, sh);
+    if (kpair)
+      hipLaunchKernelGGL(
+          (gemm_kernel<BMv, BNv, WRv, WCv, true, false, 4>), grid, block,
+          0, cur_stream(), a, b, c, part_ptr, nullptr, M, N, K, kslice, 0,
+          (int)EpStore::kPlain, 0, 1, 0, sh);
+    else
+      hipLaunchKernelGGL(
+          (gemm_kernel<BMv, BNv, WRv, WCv, true, false, 3>), grid, block,
+          0, cur_stream(), a, b, c, part_ptr, nullptr, M, N, K, kslice, 0,
+          (int)EpStore::kPlain, 0, 1, 0, sh);
   };
   using c32 = std::integral_constant<int, 32>;
   using c64 = std::integral_constant<int, 64>;
