@@ -166,3 +166,44 @@ class TestCompatDemoEndToEnd:
                 out.stdout.splitlines() if l.startswith("Epoch:")]
         assert len(accs) >= 8
         assert max(accs) >= 0.92
+
+
+class TestRecordsProperty:
+    def test_update_roundtrip_fuzz(self):
+        """Wire-format round trip over random shapes/values, both the
+        object form (main.py:155-158) and the nested-string form
+        (.h:99-104) the reference emits."""
+        try:
+            from hypothesis import given, settings, strategies as st
+        except ImportError:
+            import pytest
+            pytest.skip("hypothesis not installed")
+        import json
+
+        from bflc_amd.chain import records
+
+        floats = st.floats(allow_nan=False, allow_infinity=False,
+                           width=32)
+
+        @settings(max_examples=100, deadline=None)
+        @given(w=st.lists(st.lists(floats, min_size=1, max_size=4),
+                          min_size=1, max_size=4),
+               b=st.lists(floats, min_size=1, max_size=4),
+               n=st.integers(min_value=0, max_value=10**9),
+               cost=floats)
+        def check(w, b, n, cost):
+            blob = records.update_record(w, b, n, cost)
+            up = records.parse_update(blob)
+            assert up["delta_model"]["ser_W"] == w
+            assert up["delta_model"]["ser_b"] == b
+            assert up["meta"]["n_samples"] == n
+            # nested-string form round trip
+            nested = json.dumps({
+                "delta_model": json.dumps({"ser_W": w, "ser_b": b}),
+                "meta": json.dumps({"n_samples": n, "avg_cost": cost}),
+            })
+            up2 = records.parse_update(nested)
+            assert up2["delta_model"]["ser_W"] == w
+            assert up2["meta"]["n_samples"] == n
+
+        check()
