@@ -135,3 +135,32 @@ class TestClientAssignment:
                 assert owners == sorted(owners)
                 if n_clients >= world:
                     assert set(owners) == set(range(world))
+
+
+class TestMetricsJsonl:
+    def test_per_round_records(self, tmp_path):
+        import json
+
+        import torch
+
+        from bflc_amd.comm import Transport
+        from bflc_amd.config import FLConfig
+        from bflc_amd.data import make_federated
+        from bflc_amd.fl import FLEngine
+
+        path = str(tmp_path / "m.jsonl")
+        cfg = FLConfig.for_world(1, model="logreg", n_features=8,
+                                 n_class=2, samples_per_client=64,
+                                 batch_size=32, eval_samples=64)
+        eng = FLEngine(cfg, Transport(device=torch.device("cpu")),
+                       *make_federated(cfg), metrics_path=path)
+        eng.run(5, eval_every=2)
+        recs = [json.loads(l) for l in open(path)]
+        rounds = [r for r in recs if r["kind"] == "round"]
+        assert len(rounds) == 5
+        for r in rounds:
+            for k in ("epoch", "wall_s", "train_s", "gather_s",
+                      "score_s", "aggregate_s", "global_loss"):
+                assert k in r, k
+        # eval_every=2 attaches test_acc to those rounds
+        assert sum(1 for r in rounds if r.get("test_acc") is not None) >= 2
