@@ -164,3 +164,17 @@ class TestMetricsJsonl:
                 assert k in r, k
         # eval_every=2 attaches test_acc to those rounds
         assert sum(1 for r in rounds if r.get("test_acc") is not None) >= 2
+
+
+def test_reference_default_constants():
+    """The reference's protocol constants survive as the default profile
+    (CommitteePrecompiled.h:6-19, main.py:52-88)."""
+    from bflc_amd.config import FLConfig
+    c = FLConfig.reference_defaults()
+    assert c.client_num == 20
+    assert c.comm_count == 4
+    assert c.needed_update_count == 10
+    assert c.aggregate_count == 6
+    assert c.learning_rate == 1e-3
+    assert c.batch_size == 100
+    assert c.n_features == 5 and c.n_class == 2
