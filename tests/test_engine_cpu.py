@@ -178,3 +178,51 @@ def test_reference_default_constants():
     assert c.learning_rate == 1e-3
     assert c.batch_size == 100
     assert c.n_features == 5 and c.n_class == 2
+
+
+def test_random_config_protocol_soak():
+    """Randomized protocol-liveness sweep: any well-formed config must
+    run N rounds with epochs advancing, a full committee after every
+    rotation, and two runs bitwise identical (the PBFT-replacement
+    property)."""
+    try:
+        from hypothesis import given, settings, strategies as st
+    except ImportError:
+        import pytest
+        pytest.skip("hypothesis not installed")
+    import torch
+
+    from bflc_amd.comm import Transport
+    from bflc_amd.config import FLConfig
+    from bflc_amd.data import make_federated
+    from bflc_amd.fl import FLEngine
+
+    @settings(max_examples=12, deadline=None)
+    @given(data=st.data())
+    def check(data):
+        n = data.draw(st.integers(min_value=2, max_value=8))
+        comm = data.draw(st.integers(min_value=1,
+                                     max_value=max(1, n // 2)))
+        trainers = n - comm
+        quota = data.draw(st.integers(min_value=comm, max_value=trainers))
+        agg = data.draw(st.integers(min_value=1, max_value=quota))
+        byz = data.draw(st.integers(min_value=0, max_value=n - 1))
+        part = data.draw(st.sampled_from(["iid", "dirichlet"]))
+        cfg = FLConfig(client_num=n, comm_count=comm,
+                       needed_update_count=quota, aggregate_count=agg,
+                       byzantine_clients=byz, partition=part,
+                       model="logreg", n_features=6, n_class=2,
+                       samples_per_client=32, batch_size=16,
+                       eval_samples=32)
+        runs = []
+        for _ in range(2):
+            eng = FLEngine(cfg, Transport(device=torch.device("cpu")),
+                           *make_federated(cfg))
+            eng.run(8)
+            assert eng.ledger.epoch == 8
+            roles = eng.ledger.roles()
+            assert sum(1 for r in roles.values() if r == "comm") == comm
+            runs.append(eng.global_flat.clone())
+        assert torch.equal(runs[0], runs[1])
+
+    check()
