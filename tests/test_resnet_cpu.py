@@ -122,7 +122,7 @@ class TestResNet20:
             loss = m.loss(x, y)
             loss.backward()
             m.sgd_step(0.05)
-            losses.append(float(loss))
+            losses.append(float(loss.detach()))
         assert losses[-1] < losses[0]
 
 
