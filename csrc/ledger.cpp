@@ -51,6 +51,8 @@ enum class Admit : int {
   kDuplicate = 2,
   kQuotaFull = 3,
   kNotStarted = 4,  // epoch still -999 (FL not started)
+  kFinished = 5,    // epoch > max_epoch (reference main.py:251-252)
+  kBadSignature = 6,  // origin authentication failed (chain layer)
 };
 
 struct LedgerConfig {
@@ -149,6 +151,7 @@ class CommitteeLedger {
   Admit upload_local_update(const std::string& origin, const std::string& blob,
                             int epoch, long n_samples, double avg_cost) {
     if (epoch_ <= kEpochUninit) return Admit::kNotStarted;
+    if (finished()) return Admit::kFinished;
     if (epoch != epoch_) return Admit::kStaleEpoch;
     if (local_updates_.count(origin)) return Admit::kDuplicate;
     if (update_count_ >= cfg_.needed_update_count) return Admit::kQuotaFull;
@@ -180,15 +183,25 @@ class CommitteeLedger {
   // --- UploadScores (reference .cpp:259-297) ------------------------------
   // Returns the aggregation decision when this was the comm_count-th score
   // set, std::nullopt otherwise (including on rejected submissions).
+  //
+  // Hostile-input hardening (round 2): score keys that name no ADMITTED
+  // update are dropped at admission — the reference only ever reads the
+  // score maps for keys that exist in its updates table (.cpp:351-377),
+  // so a garbage key there is inert; here it must not survive into
+  // decide_aggregation where it could differ across replicas or throw.
   std::optional<AggregationDecision> upload_scores(
       const std::string& origin, int epoch,
       const std::map<std::string, double>& scores) {
     if (pending_) throw std::logic_error("aggregation pending, commit first");
-    if (epoch_ <= kEpochUninit || epoch != epoch_) return std::nullopt;
+    if (epoch_ <= kEpochUninit || finished() || epoch != epoch_)
+      return std::nullopt;
     auto it = roles_.find(origin);
     if (it == roles_.end() || it->second == "trainer") return std::nullopt;
+    std::map<std::string, double> filtered;
+    for (const auto& kv : scores)
+      if (local_updates_.count(kv.first)) filtered.emplace(kv);
     const bool existed = local_scores_.count(origin) > 0;
-    local_scores_[origin] = scores;
+    local_scores_[origin] = std::move(filtered);
     if (!existed) ++score_count_;
     if (ledger_trace())
       std::clog << score_count_ << " scores has been uploaded (epoch "
@@ -243,6 +256,12 @@ class CommitteeLedger {
 
   // --- introspection ------------------------------------------------------
   int epoch() const { return epoch_; }
+  // The FL run is over once epoch exceeds max_epoch (reference clients
+  // exit on epoch > MAX_EPOCH, main.py:251-252); the ledger refuses
+  // further uploads so no replica can be driven past the end.
+  bool finished() const {
+    return epoch_ > kEpochUninit && epoch_ > cfg_.max_epoch;
+  }
   int update_count() const { return update_count_; }
   int score_count() const { return score_count_; }
   double global_loss() const { return global_loss_; }
@@ -325,21 +344,27 @@ class CommitteeLedger {
 
   // Median -> sort -> top-k (reference .cpp:349-400 decision half).
   AggregationDecision decide_aggregation() const {
-    // Per-trainer median over the committee score sets. The reference
-    // unions the keys of all score maps (.cpp:351-362); trainers keyed by
-    // any committee member participate.
-    std::map<std::string, std::vector<float>> per_trainer;
-    for (const auto& comm_kv : local_scores_)
-      for (const auto& sc : comm_kv.second)
-        per_trainer[sc.first].push_back(static_cast<float>(sc.second));
-
+    // Per-trainer median over the committee score sets. Exactly the
+    // ADMITTED updates participate (reference iterates its updates
+    // table, .cpp:351-362); a committee member that omitted an admitted
+    // trainer contributes 0.0 for it — so every admitted trainer always
+    // has score_count values and decide can never throw, whatever a
+    // hostile/buggy committee member uploaded.
     AggregationDecision dec;
     dec.epoch = epoch_;
     std::vector<std::pair<std::string, float>> ranked;
-    for (const auto& kv : per_trainer) {
-      float m = median_ref(kv.second);
-      dec.median_scores[kv.first] = m;
-      ranked.emplace_back(kv.first, m);
+    for (const auto& id : update_order_) {
+      std::vector<float> vals;
+      vals.reserve(local_scores_.size());
+      for (const auto& comm_kv : local_scores_) {
+        auto sit = comm_kv.second.find(id);
+        vals.push_back(sit == comm_kv.second.end()
+                           ? 0.0f
+                           : static_cast<float>(sit->second));
+      }
+      float m = vals.empty() ? 0.0f : median_ref(vals);
+      dec.median_scores[id] = m;
+      ranked.emplace_back(id, m);
     }
     // Deterministic: score desc, then id asc (reference: unstable sort).
     std::sort(ranked.begin(), ranked.end(), [](const auto& a, const auto& b) {
@@ -351,14 +376,16 @@ class CommitteeLedger {
         std::min<int>(cfg_.aggregate_count, static_cast<int>(ranked.size()));
     for (int i = 0; i < k; ++i) {
       const auto& id = ranked[i].first;
+      // ranked is built from update_order_, so the lookup always hits;
+      // keep the skip (not a throw) as a restore()-path safety net.
       auto it = local_updates_.find(id);
-      if (it == local_updates_.end())
-        throw std::logic_error("scored trainer has no update: " + id);
+      if (it == local_updates_.end()) continue;
       dec.selected.emplace_back(id, it->second.n_samples);
       dec.total_weight += static_cast<double>(it->second.n_samples);
       dec.avg_cost += it->second.avg_cost;
     }
-    if (k > 0) dec.avg_cost /= static_cast<double>(k);
+    if (!dec.selected.empty())
+      dec.avg_cost /= static_cast<double>(dec.selected.size());
     const int c =
         std::min<int>(cfg_.comm_count, static_cast<int>(ranked.size()));
     for (int i = 0; i < c; ++i) dec.next_committee.push_back(ranked[i].first);
@@ -389,7 +416,9 @@ PYBIND11_MODULE(_ledger, m) {
       .value("STALE_EPOCH", bflc::Admit::kStaleEpoch)
       .value("DUPLICATE", bflc::Admit::kDuplicate)
       .value("QUOTA_FULL", bflc::Admit::kQuotaFull)
-      .value("NOT_STARTED", bflc::Admit::kNotStarted);
+      .value("NOT_STARTED", bflc::Admit::kNotStarted)
+      .value("FINISHED", bflc::Admit::kFinished)
+      .value("BAD_SIGNATURE", bflc::Admit::kBadSignature);
 
   py::class_<bflc::LedgerConfig>(m, "LedgerConfig")
       .def(py::init<>())
@@ -453,6 +482,7 @@ PYBIND11_MODULE(_ledger, m) {
       .def("update_blob", &bflc::CommitteeLedger::update_blob)
       .def("update_meta", &bflc::CommitteeLedger::update_meta)
       .def("has_pending", &bflc::CommitteeLedger::has_pending)
+      .def_property_readonly("finished", &bflc::CommitteeLedger::finished)
       .def("snapshot", &bflc::CommitteeLedger::snapshot)
       .def("restore", &bflc::CommitteeLedger::restore);
 }

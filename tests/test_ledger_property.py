@@ -150,6 +150,14 @@ def test_ledger_matches_python_model(data):
             smap = {tr: data.draw(st.floats(min_value=0, max_value=1,
                                             allow_nan=False, width=32))
                     for tr in model.updates}
+            # hostile input: sometimes key scores by ids with no admitted
+            # update — the ledger must drop them at admission and the
+            # round must proceed identically (VERDICT weak #3)
+            if data.draw(st.booleans()):
+                smap[data.draw(st.sampled_from(
+                    ["ghost", "", "node_999", origins[0] + "x"]))] = \
+                    data.draw(st.floats(min_value=0, max_value=10,
+                                        allow_nan=False, width=32))
             d = led.upload_scores(o, model.epoch, smap)
             ref_sel = model.upload_scores(o, model.epoch, smap)
             if ref_sel is not None:
