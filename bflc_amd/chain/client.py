@@ -25,11 +25,28 @@ class BcosClient:
     def __init__(self, chain: LocalChain) -> None:
         self._chain = chain
         self._origin: Optional[str] = None
+        self._key: Optional[bytes] = None
         self._finished = False
 
-    # reference main.py:96 — one chain identity per FL client
-    def set_from_account_signer(self, node_id: str) -> None:
+    def _sign(self, kind: str, epoch: int, payload: str) -> Optional[bytes]:
+        if self._key is None:
+            return None
+        from bflc_amd.chain.identity import sign_with_key
+        return sign_with_key(self._key, kind, self.origin, epoch,
+                             payload.encode())
+
+    # reference main.py:96 — one chain identity per FL client. `key` is
+    # the client's own HMAC credential (chain/identity.py KeyTable);
+    # when the chain enforces signatures and no key is passed, the
+    # client's own key is fetched from the bootstrap table (the moral
+    # equivalent of reading its node_<i>.pem, get_batch_accounts.sh).
+    def set_from_account_signer(self, node_id: str,
+                                key: Optional[bytes] = None) -> None:
         self._origin = node_id
+        if key is None and getattr(self._chain, "keys", None) is not None \
+                and self._chain.keys.knows(node_id):
+            key = self._chain.keys.key(node_id)
+        self._key = key
 
     @property
     def origin(self) -> str:
@@ -61,10 +78,14 @@ class BcosClient:
             self._chain.register_node(self.origin)
         elif fn_name == "UploadLocalUpdate":
             update, epoch = args
-            self._chain.upload_local_update(self.origin, update, int(epoch))
+            self._chain.upload_local_update(
+                self.origin, update, int(epoch),
+                tag=self._sign("update", int(epoch), update))
         elif fn_name == "UploadScores":
             epoch, scores = args
-            self._chain.upload_scores(self.origin, int(epoch), scores)
+            self._chain.upload_scores(
+                self.origin, int(epoch), scores,
+                tag=self._sign("scores", int(epoch), scores))
         else:
             raise ValueError(f"unknown transaction function {fn_name}")
         return {"status": "0x0", "output": "0x"}

@@ -27,8 +27,15 @@ from bflc_amd.config import FLConfig
 
 
 class LocalChain:
-    def __init__(self, cfg: FLConfig, log_path: Optional[str] = None) -> None:
+    def __init__(self, cfg: FLConfig, log_path: Optional[str] = None,
+                 keys=None) -> None:
+        """`keys`: optional chain.identity.KeyTable. When set, update and
+        score transactions must carry a valid HMAC tag for their origin
+        (the reference chain verifies each tx's ECDSA signature against
+        the sender's account, README.md:283-299); unsigned/forged
+        submissions are rejected."""
         self.cfg = cfg
+        self.keys = keys
         self.ledger = CommitteeLedger(cfg.ledger_config())
         self.ledger.set_global_model(
             records.zero_model(cfg.n_features, cfg.n_class).encode())
@@ -62,10 +69,21 @@ class LocalChain:
             blob, epoch = self.ledger.query_global_model()
             return blob.decode(), epoch
 
+    def _verify(self, kind: str, origin: str, epoch: int, payload: str,
+                tag) -> bool:
+        if self.keys is None:
+            return True
+        return self.keys.verify(kind, origin, int(epoch), payload.encode(),
+                                tag)
+
     def upload_local_update(self, origin: str, update: str,
-                            epoch: int) -> bool:
+                            epoch: int, tag: Optional[bytes] = None) -> bool:
         up = records.parse_update(update)
         meta = up["meta"]
+        if not self._verify("update", origin, epoch, update, tag):
+            self._log("update_rejected", origin=origin, epoch=epoch,
+                      code=str(Admit.BAD_SIGNATURE))
+            return False
         with self._lock:
             code = self.ledger.upload_local_update(
                 origin, update.encode(), int(epoch),
@@ -88,7 +106,12 @@ class LocalChain:
                 return ""
             return json.dumps({k: v.decode() for k, v in ups})
 
-    def upload_scores(self, origin: str, epoch: int, scores: str) -> bool:
+    def upload_scores(self, origin: str, epoch: int, scores: str,
+                      tag: Optional[bytes] = None) -> bool:
+        if not self._verify("scores", origin, epoch, scores, tag):
+            self._log("scores_rejected", origin=origin, epoch=epoch,
+                      code=str(Admit.BAD_SIGNATURE))
+            return False
         smap = {k: float(v) for k, v in json.loads(scores).items()}
         with self._lock:
             dec = self.ledger.upload_scores(origin, int(epoch), smap)

@@ -26,10 +26,19 @@ class Transport:
     """Thin wrapper over torch.distributed with a world_size==1 fast path."""
 
     def __init__(self, backend: Optional[str] = None,
-                 device: Optional[torch.device] = None) -> None:
+                 device: Optional[torch.device] = None,
+                 timeout_s: Optional[float] = None) -> None:
         self.rank = int(os.environ.get("RANK", "0"))
         self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
         self._initialized_here = False
+        # per-collective timeout: a dead/slow peer turns an infinite hang
+        # into a raised error the engine converts to a clean abort with a
+        # checkpoint (the reference tolerated crashed trainers by 10-of-16
+        # over-provisioning, CommitteePrecompiled.h:15 — a barrier-driven
+        # engine needs an explicit failure path instead)
+        if timeout_s is None:
+            timeout_s = float(os.environ.get("BFLC_COLL_TIMEOUT_S", "300"))
+        self.timeout_s = timeout_s
 
         if device is not None:
             self.device = device
@@ -49,7 +58,7 @@ class Transport:
                 backend=backend,
                 rank=self.rank,
                 world_size=self.world_size,
-                timeout=datetime.timedelta(seconds=300),
+                timeout=datetime.timedelta(seconds=self.timeout_s),
             )
             self._initialized_here = True
         self.backend = (dist.get_backend() if dist.is_initialized()

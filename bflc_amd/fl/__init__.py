@@ -1,3 +1,4 @@
-from bflc_amd.fl.engine import FLEngine, RoundStats, client_rank
+from bflc_amd.fl.engine import (FLEngine, RoundStats, StragglerError,
+                                client_rank)
 
-__all__ = ["FLEngine", "RoundStats", "client_rank"]
+__all__ = ["FLEngine", "RoundStats", "StragglerError", "client_rank"]

@@ -37,11 +37,22 @@ from typing import Dict, List, Optional, Tuple
 import torch
 
 from bflc_amd._ledger import Admit, CommitteeLedger
+from bflc_amd.chain.identity import (KeyTable, scores_payload,
+                                     update_payload)
 from bflc_amd.comm import Transport
 from bflc_amd.config import FLConfig
 from bflc_amd.data.synthetic import Shard
 from bflc_amd.models import build_model
 from bflc_amd.ops import functional as O
+
+
+class StragglerError(RuntimeError):
+    """A collective phase failed (dead or too-slow peer). The engine
+    aborts the round cleanly after writing a rank-local checkpoint; the
+    reference survived crashed trainers by quota over-provisioning
+    (CommitteePrecompiled.h:15, only 10 of 16 need to upload) — a
+    barrier-driven engine turns the hang into this diagnosable error
+    instead, bounded by Transport.timeout_s."""
 
 
 def client_rank(client_idx: int, n_clients: int, world: int) -> int:
@@ -92,6 +103,11 @@ class FLEngine:
         self.model = build_model(cfg, self.device)
         self.global_flat = self.model.get_flat()  # device-resident fp32
 
+        # per-origin HMAC keys (reference get_batch_accounts.sh: one
+        # identity per client); every gathered submission is verified
+        # before it reaches the ledger (chain/identity.py)
+        self.keys = KeyTable(self.origins, cfg.seed)
+
         # replicated ledger: registration = deterministic replay of the
         # rank-ordered client lists (reference RegisterNode, .cpp:168-190)
         self.ledger = CommitteeLedger(cfg.ledger_config())
@@ -109,10 +125,10 @@ class FLEngine:
         # captured kernel DAG per minibatch instead of relaunching
         # ~dozens of kernels. SGD-only; falls back to eager on CPU, on
         # Adam, or if capture fails on this ROCm build.
-        self._stepper = None
+        self._steppers: Dict[tuple, object] = {}
         self._scorers: Dict[int, object] = {}
+        self._graph_pool = None  # shared across all captures (one pool)
         self._use_graphs = (cfg.use_graphs and self.device.type == "cuda"
-                            and cfg.optimizer == "sgd"
                             and os.environ.get("BFLC_GRAPHS", "1") != "0")
 
     # ------------------------------------------------------------------
@@ -132,23 +148,28 @@ class FLEngine:
 
     # ------------------------------------------------------------------
     def _graphed_stepper(self, shard: Shard, bs: int):
-        """Lazily capture the train-step hipGraph (fl/graphs.py); one
-        stepper serves every local client (equal batch shapes). Returns
+        """Lazily capture the train-step hipGraph (fl/graphs.py), one
+        stepper per batch shape (heterogeneous shard sizes each get
+        their own capture instead of silently running eager — round-1
+        ADVICE item). All captures share one graph memory pool. Returns
         None — and stops trying — if capture is unavailable."""
-        if self._stepper is not None:
-            xb = shard.x[:bs]
-            return self._stepper if self._stepper.matches(xb, shard.y[:bs]) \
-                else None
+        xb, yb = shard.x[:bs], shard.y[:bs]
+        key = (tuple(xb.shape), xb.dtype, tuple(yb.shape))
+        if key in self._steppers:
+            return self._steppers[key]
         try:
             from bflc_amd.fl.graphs import GraphedTrainStep
-            self._stepper = GraphedTrainStep(
-                self.model, self.cfg.learning_rate, shard.x[:bs],
-                shard.y[:bs])
+            st = GraphedTrainStep(self.model, self.cfg.learning_rate, xb,
+                                  yb, optimizer=self.cfg.optimizer,
+                                  pool=self._graph_pool)
+            if self._graph_pool is None:
+                self._graph_pool = st.pool()
+            self._steppers[key] = st
         except Exception as e:  # capture unsupported: eager fallback
             warnings.warn(f"hipGraph capture failed, running eager: {e}")
             self._use_graphs = False
             return None
-        return self._stepper
+        return self._steppers[key]
 
     # ------------------------------------------------------------------
     def _local_train(self, client: int) -> Tuple[torch.Tensor, int, float]:
@@ -165,12 +186,13 @@ class FLEngine:
         stepper = self._graphed_stepper(shard, bs) if self._use_graphs \
             else None
         self.model.set_flat(self.global_flat)
-        if cfg.optimizer == "adam":
+        if cfg.optimizer == "adam" and stepper is None:
             m = torch.zeros_like(self.global_flat)
             v = torch.zeros_like(self.global_flat)
             step = 0
         if stepper is not None:
             stepper.cost.zero_()
+            stepper.reset_state()  # fresh Adam (m, v, step) per client
             for _ in range(cfg.local_epochs):
                 for bi in range(total_batches):
                     stepper.step(shard.x[bi * bs:(bi + 1) * bs],
@@ -231,16 +253,51 @@ class FLEngine:
         try:
             from bflc_amd.fl.graphs import GraphedScore
             shard = self.shards[scorer]
-            self._scorers[scorer] = GraphedScore(self.model, shard.x,
-                                                 shard.y)
+            gs = GraphedScore(self.model, shard.x, shard.y,
+                              pool=self._graph_pool)
+            if self._graph_pool is None:
+                self._graph_pool = gs.pool()
+            self._scorers[scorer] = gs
         except Exception as e:
             warnings.warn(f"score-graph capture failed, running eager: {e}")
             self._scorers[scorer] = None
         return self._scorers[scorer]
 
     # ------------------------------------------------------------------
+    def _gather(self, phase: str, fn, *args):
+        """Run one collective phase; a dead or timed-out peer becomes a
+        clean StragglerError after a rank-local checkpoint (path:
+        $BFLC_ABORT_CHECKPOINT, default under the system temp dir)."""
+        try:
+            return fn(*args)
+        except StragglerError:
+            raise
+        except Exception as e:
+            import tempfile
+            path = os.environ.get(
+                "BFLC_ABORT_CHECKPOINT",
+                os.path.join(tempfile.gettempdir(),
+                             f"bflc_abort_rank{self.rank}.pt"))
+            note = ""
+            try:
+                self.save(path)
+                note = f"; state checkpointed to {path}"
+            except Exception:
+                note = "; checkpoint write also failed"
+            raise StragglerError(
+                f"collective phase '{phase}' failed at epoch "
+                f"{self.ledger.epoch} on rank {self.rank}/{self.world} "
+                f"(dead or >={self.t.timeout_s:.0f}s-slow peer?): "
+                f"{e}{note}") from e
+
+    # ------------------------------------------------------------------
     def run_round(self, eval_global: bool = False) -> RoundStats:
         cfg, led = self.cfg, self.ledger
+        if led.finished:
+            raise RuntimeError(
+                f"FL run finished: epoch {led.epoch} > max_epoch "
+                f"{cfg.max_epoch} (reference clients exit here, "
+                "main.py:251-252)")
         epoch = led.epoch
         t0 = time.perf_counter()
 
@@ -258,18 +315,31 @@ class FLEngine:
         t1 = time.perf_counter()
 
         # ---- phase U: publish updates (one RCCL all-gather) -------------
-        metas = [(o, epoch, n, c) for (o, delta, n, c) in local_updates]
-        all_metas = self.t.all_gather_objects(metas)
+        # each submission carries an HMAC tag binding (origin, epoch,
+        # meta) to the origin's bootstrap key (chain/identity.py); every
+        # replica verifies before feeding its ledger, so a forged origin
+        # is rejected deterministically on all ranks (reference: one
+        # ECDSA key per client, get_batch_accounts.sh)
+        metas = [(o, epoch, n, c,
+                  self.keys.sign("update", o, epoch, update_payload(n, c)))
+                 for (o, delta, n, c) in local_updates]
+        all_metas = self._gather("updates", self.t.all_gather_objects, metas)
         max_subs = max((len(m) for m in all_metas), default=0)
         P = self.global_flat.numel()
         stack = torch.zeros(max(max_subs, 1), P, dtype=torch.float32,
                             device=self.global_flat.device)
         for j, (_, delta, _, _) in enumerate(local_updates):
             stack[j] = delta
-        gathered = self.t.all_gather_tensor(stack)
+        gathered = self._gather("update-deltas", self.t.all_gather_tensor,
+                                stack)
         updates: List[Tuple[str, torch.Tensor]] = []
+        n_bad_sig = 0
         for r, rank_metas in enumerate(all_metas):
-            for j, (origin, ep, n, c) in enumerate(rank_metas):
+            for j, (origin, ep, n, c, tag) in enumerate(rank_metas):
+                if not self.keys.verify("update", origin, ep,
+                                        update_payload(n, c), tag):
+                    n_bad_sig += 1
+                    continue
                 code = led.upload_local_update(origin, b"", ep, n, c)
                 if code == Admit.ACCEPTED:
                     updates.append((origin, gathered[r][j].to(self.device)))
@@ -282,17 +352,26 @@ class FLEngine:
         else:
             local_scorers = [i for i in self.local_clients
                              if roles.get(self.origins[i]) == "comm"]
-        my_scores = [(self.origins[i], self._score_candidates(i, updates))
-                     for i in local_scorers]
+        my_scores = []
+        for i in local_scorers:
+            o = self.origins[i]
+            smap = self._score_candidates(i, updates)
+            my_scores.append((o, smap, self.keys.sign(
+                "scores", o, epoch, scores_payload(smap))))
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
         t3 = time.perf_counter()
 
         # ---- phase V: publish scores, decide ----------------------------
-        all_scores = self.t.all_gather_objects(my_scores)
+        all_scores = self._gather("scores", self.t.all_gather_objects,
+                                  my_scores)
         decision = None
         for rank_scores in all_scores:
-            for origin, smap in rank_scores:
+            for origin, smap, tag in rank_scores:
+                if not self.keys.verify("scores", origin, epoch,
+                                        scores_payload(smap), tag):
+                    n_bad_sig += 1
+                    continue
                 d = led.upload_scores(origin, epoch, smap)
                 if d is not None:
                     decision = d
@@ -305,17 +384,21 @@ class FLEngine:
         by_origin = dict(updates)
         sel = decision.selected
         K = len(sel)
-        deltas = torch.empty(K, P, dtype=torch.float32,
-                             device=self.global_flat.device)
-        weights = torch.empty(K, dtype=torch.float32,
-                              device=self.global_flat.device)
-        for k, (origin, w) in enumerate(sel):
-            deltas[k] = by_origin[origin]
-            weights[k] = float(w)
         self.last_decision = decision
-        avg = O.weighted_fedavg(deltas, weights)
-        # global -= lr * avg  (reference .cpp:403-414)
-        O.axpy_(self.global_flat, -cfg.learning_rate, avg)
+        if K > 0:
+            deltas = torch.empty(K, P, dtype=torch.float32,
+                                 device=self.global_flat.device)
+            weights = torch.empty(K, dtype=torch.float32,
+                                  device=self.global_flat.device)
+            for k, (origin, w) in enumerate(sel):
+                deltas[k] = by_origin[origin]
+                weights[k] = float(w)
+            avg = O.weighted_fedavg(deltas, weights)
+            # global -= lr * avg  (reference .cpp:403-414)
+            O.axpy_(self.global_flat, -cfg.learning_rate, avg)
+        # K == 0 (no admitted updates survived): the model is unchanged
+        # but the epoch still advances and the committee rotates/refills
+        # — the round degrades instead of wedging every replica
         led.commit_aggregate(b"")
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
@@ -331,7 +414,8 @@ class FLEngine:
             global_loss=led.global_loss, n_updates=len(updates),
             n_selected=K, samples_trained=samples_trained, test_acc=acc)
         self.metrics.log("round", stats,
-                         selected=[o for o, _ in decision.selected])
+                         selected=[o for o, _ in decision.selected],
+                         bad_signatures=n_bad_sig)
         return stats
 
     # ------------------------------------------------------------------
@@ -344,6 +428,8 @@ class FLEngine:
     def run(self, rounds: int, eval_every: int = 0) -> List[RoundStats]:
         out = []
         for r in range(rounds):
+            if self.ledger.finished:  # max_epoch reached (main.py:251)
+                break
             ev = eval_every > 0 and (r + 1) % eval_every == 0
             out.append(self.run_round(eval_global=ev))
         return out

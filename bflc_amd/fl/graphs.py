@@ -8,7 +8,7 @@ ragged tail, engine.py). A profiled round is made of many 8-25 us
 kernels (profiles/r01_kernel_stats_nhwc.md), so per-kernel launch gaps
 are a real cost; capturing one
 
-    zero-visible-state -> forward -> backward -> fused SGD -> loss+=
+    zero-visible-state -> forward -> backward -> fused optimizer -> loss+=
 
 step into a hipGraph and replaying it once per batch launches the whole
 step as a single unit.
@@ -24,11 +24,24 @@ guarantees:
   * intermediate/grad tensors allocated during capture live in the
     graph's private memory pool and are rewritten in place on replay.
 
-SGD only: Adam's per-step bias correction is a host scalar that would
-be baked in at capture time; the eager path keeps handling Adam.
+SGD captures trivially. Adam captures through the graph-safe kernel
+pair ``adam_tick`` + ``adam_master_dev`` (csrc/hip/elementwise.hip):
+the step counter and the bias corrections (1-beta^t) live in DEVICE
+buffers advanced inside the captured step, so nothing epoch-dependent
+is baked in; ``reset_state()`` zeroes (m, v, step) between clients —
+the reference builds a fresh optimizer per round (main.py:109,126).
+Adam-on-GPU-fp32 (no bf16 shadow) stays eager: only the master+shadow
+kernel has the device-step variant, and fp32 compute is a test-only
+configuration.
+
 Capture executes real kernels, so construction MUTATES the weights —
 callers build the stepper before loading the round's global weights
 (engine.py does this), or snapshot/restore around construction.
+
+All captures in one engine share ONE graph memory pool (pass
+``pool=graph.pool()`` of the first capture): per-scorer private pools
+each held full-shard forward activations and grew GPU memory with the
+number of local scorers (round-1 ADVICE item).
 """
 from __future__ import annotations
 
@@ -36,18 +49,34 @@ import torch
 
 
 class GraphedTrainStep:
-    """One captured (forward, backward, fused-SGD) step over static
-    input buffers; ``step()`` copies a batch in and replays."""
+    """One captured (forward, backward, fused-optimizer) step over
+    static input buffers; ``step()`` copies a batch in and replays."""
 
     def __init__(self, model, lr: float, x_proto: torch.Tensor,
-                 y_proto: torch.Tensor, warmup: int = 2) -> None:
+                 y_proto: torch.Tensor, optimizer: str = "sgd",
+                 warmup: int = 2, pool=None) -> None:
         self.model = model
+        self.optimizer = optimizer
         self.sx = torch.empty_like(x_proto)
         self.sy = torch.empty_like(y_proto)
         self.cost = torch.zeros((), device=model.device,
                                 dtype=torch.float32)
         self.sx.copy_(x_proto)
         self.sy.copy_(y_proto)
+        if optimizer == "adam":
+            if model.cflat is model.flat:
+                raise RuntimeError(
+                    "graphed Adam needs the bf16-shadow model (the "
+                    "device-step kernel is adam_master_graph_)")
+            self.m = torch.zeros_like(model.flat)
+            self.v = torch.zeros_like(model.flat)
+            self.step_t = torch.zeros(1, device=model.device,
+                                      dtype=torch.int32)
+            self.bc = torch.zeros(2, device=model.device,
+                                  dtype=torch.float32)
+        elif optimizer != "sgd":
+            raise ValueError(optimizer)
+        self._lr = float(lr)
 
         # warmup on a side stream (materializes autograd engine state,
         # cat/workspace allocations) so none of it happens mid-capture
@@ -55,24 +84,42 @@ class GraphedTrainStep:
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
             for _ in range(max(warmup, 1)):
-                self._eager_step(lr)
+                self._one_step()
         torch.cuda.current_stream().wait_stream(side)
 
         # grads=None at capture: backward's grad tensors are allocated
         # from the graph pool and rewritten in place on every replay
         self.model.zero_grad()
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        with torch.cuda.graph(self.graph, pool=pool):
             loss = self.model.loss(self.sx, self.sy)
             loss.backward()
-            self.model.sgd_step(lr)
+            self._optim_step()
             self.cost.add_(loss.detach().float())
 
-    def _eager_step(self, lr: float) -> None:
+    def _optim_step(self) -> None:
+        if self.optimizer == "adam":
+            self.model.adam_step_graph(self.m, self.v, self.step_t,
+                                       self.bc, self._lr)
+        else:
+            self.model.sgd_step(self._lr)
+
+    def _one_step(self) -> None:
         self.model.zero_grad()
         loss = self.model.loss(self.sx, self.sy)
         loss.backward()
-        self.model.sgd_step(lr)
+        self._optim_step()
+
+    def reset_state(self) -> None:
+        """Fresh optimizer for a new client/round (reference builds a
+        new TF graph per round, main.py:109)."""
+        if self.optimizer == "adam":
+            self.m.zero_()
+            self.v.zero_()
+            self.step_t.zero_()
+
+    def pool(self):
+        return self.graph.pool()
 
     def matches(self, xb: torch.Tensor, yb: torch.Tensor) -> bool:
         return (xb.shape == self.sx.shape and xb.dtype == self.sx.dtype
@@ -94,7 +141,7 @@ class GraphedScore:
     phase that grows with world size."""
 
     def __init__(self, model, shard_x: torch.Tensor,
-                 shard_y: torch.Tensor) -> None:
+                 shard_y: torch.Tensor, pool=None) -> None:
         self.model = model
         self.cand = torch.empty_like(model.flat)
         self.cand.copy_(model.flat.detach())
@@ -105,9 +152,12 @@ class GraphedScore:
             model.accuracy_t(shard_x, shard_y)
         torch.cuda.current_stream().wait_stream(side)
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        with torch.cuda.graph(self.graph, pool=pool):
             model.set_flat(self.cand)
             self.acc = model.accuracy_t(shard_x, shard_y)
+
+    def pool(self):
+        return self.graph.pool()
 
     def score(self, cand: torch.Tensor) -> torch.Tensor:
         """Returns a device scalar snapshot (no host sync)."""

@@ -172,6 +172,20 @@ class FlatModel:
                 adam_step_(self.flat.data, g.float(), m, v, step, lr)
                 self.cflat.data.copy_(self.flat.to(self.compute_dtype))
 
+    def adam_step_graph(self, m: torch.Tensor, v: torch.Tensor,
+                        step_t: torch.Tensor, bc: torch.Tensor,
+                        lr: float) -> None:
+        """hipGraph-capturable Adam: the step counter and bias
+        corrections live in device buffers advanced by the fused kernel
+        pair (csrc/hip/elementwise.hip adam_tick/adam_master_dev) — so
+        the whole step captures without baking in a host-side epoch."""
+        assert self.cflat is not self.flat, "needs the bf16 shadow model"
+        g = self.grad_flat()
+        from bflc_amd.ops.functional import hip_ops
+        hip_ops().adam_master_graph_(self.flat.detach(), self.cflat.data,
+                                     g, m, v, step_t, bc, float(lr), 0.9,
+                                     0.999, 1e-8)
+
     # -- train/eval ---------------------------------------------------------
     def loss(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         return O.softmax_cross_entropy(self.forward(self._cast(x)), y)

@@ -18,6 +18,10 @@ void sgd_master_(torch::Tensor p, torch::Tensor shadow, torch::Tensor g,
 void adam_master_(torch::Tensor p, torch::Tensor shadow, torch::Tensor g,
                   torch::Tensor m, torch::Tensor v, long step, double lr,
                   double beta1, double beta2, double eps);
+void adam_master_graph_(torch::Tensor p, torch::Tensor shadow,
+                        torch::Tensor g, torch::Tensor m, torch::Tensor v,
+                        torch::Tensor step, torch::Tensor bc, double lr,
+                        double beta1, double beta2, double eps);
 void refresh_shadow_(torch::Tensor p, torch::Tensor shadow);
 torch::Tensor relu_fwd(torch::Tensor x);
 torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy);
@@ -90,6 +94,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_master_", &bflc::sgd_master_,
         "fused fp32-master SGD + bf16 shadow refresh");
   m.def("adam_master_", &bflc::adam_master_);
+  m.def("adam_master_graph_", &bflc::adam_master_graph_,
+        "hipGraph-capturable Adam: device step counter + on-device bias "
+        "corrections (tick kernel), fp32 master + bf16 shadow");
   m.def("refresh_shadow_", &bflc::refresh_shadow_, "shadow = bf16(master)");
   m.def("relu_fwd", &bflc::relu_fwd);
   m.def("relu_bwd", &bflc::relu_bwd);
@@ -103,7 +110,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w"), py::arg("b"), py::arg("relu") = false);
   m.def("gemm_raw", &bflc::gemm_raw, "raw GEMM (bench/ablation)");
   m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");
-  m.def("conv2d_fwd", &bflc::conv2d_fwd, "im2col + MFMA GEMM, NCHW");
+  m.def("conv2d_fwd", &bflc::conv2d_fwd,
+        "NHWC implicit-GEMM convolution (MFMA)");
   m.def("conv2d_fwd_col", &bflc::conv2d_fwd_col, "(y, col) - col for bwd",
         py::arg("x"), py::arg("w"), py::arg("b"), py::arg("stride"),
         py::arg("pad"), py::arg("relu") = false);

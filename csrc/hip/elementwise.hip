@@ -142,6 +142,47 @@ __global__ void adam_master_kernel(float* __restrict__ p,
   }
 }
 
+// hipGraph-capturable Adam: the per-step bias corrections are HOST
+// scalars in adam_master_ (baked in at capture), so a captured Adam
+// step would replay step-1 corrections forever. Here the step counter
+// lives in a device int and a single-thread tick kernel advances it and
+// writes (1-beta1^t, 1-beta2^t) into a 2-float device buffer the main
+// kernel reads — the whole (tick, update) pair captures and replays
+// correctly (VERDICT round-1 item 10).
+__global__ void adam_tick_kernel(int* __restrict__ step,
+                                 float* __restrict__ bc, float beta1,
+                                 float beta2) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    int t = *step + 1;
+    *step = t;
+    bc[0] = 1.f - powf(beta1, (float)t);
+    bc[1] = 1.f - powf(beta2, (float)t);
+  }
+}
+
+__global__ void adam_master_dev_kernel(float* __restrict__ p,
+                                       bf16* __restrict__ shadow,
+                                       const bf16* __restrict__ g,
+                                       float* __restrict__ m,
+                                       float* __restrict__ v,
+                                       const float* __restrict__ bc,
+                                       float lr, float beta1, float beta2,
+                                       float eps, long n) {
+  float bc1 = bc[0], bc2 = bc[1];
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride) {
+    float gk = b2f(g[k]);
+    float mk = beta1 * m[k] + (1.f - beta1) * gk;
+    float vk = beta2 * v[k] + (1.f - beta2) * gk * gk;
+    m[k] = mk;
+    v[k] = vk;
+    float pv = p[k] - lr * (mk / bc1) / (sqrtf(vk / bc2) + eps);
+    p[k] = pv;
+    shadow[k] = f2b(pv);
+  }
+}
+
 // shadow refresh: shadow = bf16(master)
 __global__ void cast_f32_bf16_kernel(const float* __restrict__ p,
                                      bf16* __restrict__ shadow, long n) {
@@ -294,6 +335,26 @@ void adam_master_(torch::Tensor p, torch::Tensor shadow, torch::Tensor g,
                      (bf16*)shadow.data_ptr(), (const bf16*)g.data_ptr(),
                      m.data_ptr<float>(), v.data_ptr<float>(), (float)lr,
                      (float)beta1, (float)beta2, (float)eps, bc1, bc2, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void adam_master_graph_(torch::Tensor p, torch::Tensor shadow,
+                        torch::Tensor g, torch::Tensor m, torch::Tensor v,
+                        torch::Tensor step, torch::Tensor bc, double lr,
+                        double beta1, double beta2, double eps) {
+  CHECK_GPU(p); CHECK_CONTIG(p); CHECK_CONTIG(shadow); CHECK_CONTIG(g);
+  TORCH_CHECK(step.scalar_type() == at::kInt && step.numel() == 1);
+  TORCH_CHECK(bc.scalar_type() == at::kFloat && bc.numel() == 2);
+  long n = p.numel();
+  hipLaunchKernelGGL(adam_tick_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     step.data_ptr<int>(), bc.data_ptr<float>(),
+                     (float)beta1, (float)beta2);
+  hipLaunchKernelGGL(adam_master_dev_kernel, dim3(grid_for(n, 2)),
+                     dim3(kBlock), 0, cur_stream(), p.data_ptr<float>(),
+                     (bf16*)shadow.data_ptr(), (const bf16*)g.data_ptr(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     bc.data_ptr<float>(), (float)lr, (float)beta1,
+                     (float)beta2, (float)eps, n);
   HIP_CHECK(hipGetLastError());
 }
 

@@ -108,3 +108,119 @@ def test_world4_byzantine_replicas_identical(tmp_path):
     assert all(r["epoch"] == 4 for r in res)
     assert all(r["digest"] == res[0]["digest"] for r in res)
     assert all(r["roles"] == res[0]["roles"] for r in res)
+
+
+# ---------------------------------------------------------------------------
+# Straggler/crash tolerance (VERDICT round-1 item 5): the reference
+# tolerated dead trainers by quota over-provisioning
+# (CommitteePrecompiled.h:15 — any 10 of 16 suffice); the barrier-driven
+# engine instead converts the hang into a diagnosable StragglerError
+# with a rank-local checkpoint, bounded by the collective timeout.
+
+STRAGGLER_WORKER = r"""
+import os, sys, time
+import torch
+sys.path.insert(0, {repo!r})
+from bflc_amd.config import FLConfig
+from bflc_amd.comm import Transport
+from bflc_amd.data import make_federated
+from bflc_amd.fl import FLEngine, StragglerError
+
+cfg = FLConfig.for_world(4, model="mlp", n_features=32, n_class=4,
+                         samples_per_client=64, batch_size=32,
+                         eval_samples=64)
+shards, test = make_federated(cfg)
+t = Transport(backend="gloo", device=torch.device("cpu"), timeout_s=15)
+eng = FLEngine(cfg, t, shards, test)
+eng.run_round()  # round 0 completes on all ranks
+if t.rank == 3:
+    os._exit(0)  # this rank dies before round 1
+try:
+    eng.run_round()
+except StragglerError as e:
+    print(f"STRAGGLER_CAUGHT rank={{t.rank}}: {{e}}", flush=True)
+    sys.exit(7)
+sys.exit(1)  # the round must NOT silently complete without rank 3
+"""
+
+
+def test_dead_rank_raises_straggler_error(tmp_path):
+    import time as _time
+    script = tmp_path / "straggler.py"
+    script.write_text(STRAGGLER_WORKER.format(repo=REPO))
+    ckpt = tmp_path / "abort_ckpt.pt"
+    procs = []
+    for rank in range(4):
+        env = dict(os.environ, RANK=str(rank), WORLD_SIZE="4",
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT="29547",
+                   OMP_NUM_THREADS="2",
+                   BFLC_ABORT_CHECKPOINT=str(ckpt) + f".{rank}")
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE))
+    t0 = _time.time()
+    outs = [p.communicate(timeout=180) for p in procs]
+    elapsed = _time.time() - t0
+    assert procs[3].returncode == 0  # the dead rank exited cleanly
+    for r in range(3):
+        out, err = outs[r]
+        assert procs[r].returncode == 7, \
+            (r, out.decode()[-500:], err.decode()[-2000:])
+        msg = out.decode()
+        assert "STRAGGLER_CAUGHT" in msg
+        assert "dead or" in msg and "epoch 1" in msg
+        # clean abort wrote the rank-local checkpoint
+        assert os.path.exists(str(ckpt) + f".{r}")
+    # diagnosable WITHIN the timeout budget, not an unbounded hang
+    assert elapsed < 120
+
+
+# ---------------------------------------------------------------------------
+# Variable-length blob gather at ResNet-50 delta size (~102 MB): the
+# padding/offset logic must round-trip big, unequal payloads (VERDICT
+# round-1 item 8 — de-risk the first real 8-GPU run).
+
+BIGBLOB_WORKER = r"""
+import hashlib, os, sys
+import torch
+sys.path.insert(0, {repo!r})
+from bflc_amd.comm import Transport
+
+t = Transport(backend="gloo", device=torch.device("cpu"))
+if t.rank == 0:
+    blob = os.urandom(102_000_000)  # ~ResNet-50 flat fp32 delta
+else:
+    blob = os.urandom(1_234_567)    # very unequal: exercises padding
+got = t.all_gather_blobs(blob)
+assert len(got) == 2
+assert got[t.rank] == blob
+digests = [hashlib.sha256(b).hexdigest() for b in got]
+lens = [len(b) for b in got]
+print(f"RANK{{t.rank}} lens={{lens}} d0={{digests[0][:12]}} "
+      f"d1={{digests[1][:12]}}", flush=True)
+t.barrier()
+t.close()
+"""
+
+
+def test_bigblob_gather_roundtrip(tmp_path):
+    script = tmp_path / "bigblob.py"
+    script.write_text(BIGBLOB_WORKER.format(repo=REPO))
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ, RANK=str(rank), WORLD_SIZE="2",
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT="29548",
+                   OMP_NUM_THREADS="2")
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE))
+    lines = []
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        assert p.returncode == 0, err.decode()[-3000:]
+        lines.append(out.decode().strip())
+    # both ranks saw the same bytes (identical digests, correct sizes)
+    f0 = lines[0].split("lens=")[1]
+    f1 = lines[1].split("lens=")[1]
+    assert f0 == f1
+    assert "[102000000, 1234567]" in lines[0]
