@@ -3,11 +3,20 @@
 (BASELINE.json metric: global-model test acc + wall-clock/FL-round at
 1/2/4/8 FL nodes = GPUs).
 
-One process per GPU (torchrun), one FL node per GPU, non-IID Dirichlet
-shards of synthetic FEMNIST-shaped data (28x28x1, 62 classes), random
-init, bf16 compute. A step = one full FL round (local train -> RCCL
-all-gather of updates -> committee scoring -> all-gather of scores ->
-weighted FedAvg commit). Weak scaling: per-GPU work fixed as N grows.
+One process per GPU (torchrun), non-IID Dirichlet shards of synthetic
+FEMNIST-shaped data (28x28x1, 62 classes), random init, bf16 compute.
+A step = one full FL round (local train -> RCCL all-gather of updates ->
+committee scoring -> all-gather of scores -> weighted FedAvg commit).
+
+The protocol is FIXED at every GPU count: 8 FL clients, committee 4,
+update quota 4, top-3 aggregation (the for_world(8) shape of the
+reference's 20-client/committee-4/quota-10/top-6 protocol,
+CommitteePrecompiled.h:11-17), hosted on however many GPUs run — so the
+timed region always contains trainer racing, committee scoring of
+multiple candidates, median -> top-k, and weighted FedAvg, and the
+1-GPU and 8-GPU points of the scaling curve measure the SAME protocol
+(round-1 VERDICT item 1: the old default degenerated to 1 client
+self-scoring at N=1). Total work is fixed as N grows => strong scaling.
 
 Prints ONE JSON line on rank 0 (driver contract).
 """
@@ -30,9 +39,11 @@ def main() -> None:
                              "mlp", "logreg"])
     ap.add_argument("--samples-per-client", type=int, default=None)
     ap.add_argument("--batch-size", type=int, default=None)
-    ap.add_argument("--clients", type=int, default=None,
-                    help="FL clients (default: one per rank; more than "
-                         "ranks = multiple clients per GPU, reference-style)")
+    ap.add_argument("--clients", type=int, default=8,
+                    help="FL clients (default 8 = fixed committee-4/"
+                         "quota-4 protocol at every GPU count; more "
+                         "clients than ranks = multiple clients per GPU, "
+                         "reference-style)")
     ap.add_argument("--byzantine", type=int, default=0,
                     help="label-flip attacker clients (BASELINE config 4; "
                          "committee scoring is the defense)")
@@ -75,6 +86,8 @@ def main() -> None:
             f"--master-addr 127.0.0.1 bench.py --gpus {args.gpus} ...")
     n = int(os.environ.get("WORLD_SIZE", args.gpus))
     n_clients = args.clients or n
+    if n_clients < n:
+        raise SystemExit(f"bench.py: --clients {n_clients} < {n} ranks")
     cfg = FLConfig.for_world(
         n_clients, model=args.model, n_class=md["n_class"],
         samples_per_client=spc,
@@ -88,6 +101,15 @@ def main() -> None:
     eng = FLEngine(cfg, t, shards, test, metrics_path=args.metrics)
 
     use_cuda = t.device.type == "cuda"
+    # per-rank sanity line (stderr): makes the first real multi-GPU run
+    # debuggable from its log — device binding, backend, client split
+    import sys
+    print(f"# bflc rank {t.rank}/{t.world_size} device={t.device} "
+          f"backend={t.backend} "
+          f"visible_gpus={torch.cuda.device_count() if use_cuda else 0} "
+          f"local_clients={eng.local_clients} "
+          f"committee={eng.ledger.committee()}",
+          file=sys.stderr, flush=True)
 
     # ---- warmup (untimed) ----
     for _ in range(args.warmup):
@@ -132,10 +154,9 @@ def main() -> None:
 
     if t.rank == 0:
         # BASELINE.json's named metric is wall-clock per FL round (plus
-        # global-model test acc, reported alongside). Weak scaling: one
-        # FL node per GPU with fixed per-client work; the committee
-        # protocol itself fixes how many of the N nodes train vs score
-        # each round (BASELINE config 2 pins committee=4 at 8 nodes).
+        # global-model test acc, reported alongside). The protocol is
+        # the same at every N (8 clients, committee 4, quota 4), so
+        # total per-round work is fixed as GPUs grow: strong scaling.
         out = {
             "metric": "fl_round_wall_clock_ms",
             "value": elapsed / args.steps * 1e3,
@@ -146,7 +167,7 @@ def main() -> None:
             "warmup": args.warmup,
             "ms_per_step": elapsed / args.steps * 1e3,
             "higher_is_better": False,
-            "scaling": "weak",
+            "scaling": "strong",
             "vs_baseline": None,  # reference publishes no perf numbers
             "dtype": "bf16" if use_cuda else "fp32",
             "data": "synthetic",
