@@ -124,11 +124,20 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--clients", type=int, default=20)
     ap.add_argument("--epochs", type=int, default=10)
+    ap.add_argument("--data", default="synthetic",
+                    choices=["synthetic", "occupancy"],
+                    help="occupancy = the reference's real UCI CSV "
+                         "(reference main.py:32-53; its published "
+                         "0.9214 accuracy is on this data)")
     args = ap.parse_args()
 
     cfg = FLConfig() if args.clients == 20 else \
         FLConfig.for_world(args.clients)
-    shards, test = make_federated(cfg)
+    if args.data == "occupancy":
+        from bflc_amd.data import load_occupancy
+        shards, test = load_occupancy(clients=cfg.client_num)
+    else:
+        shards, test = make_federated(cfg)
     chain = LocalChain(cfg, log_path="bflc_chain_log.jsonl")
 
     stop = threading.Event()
