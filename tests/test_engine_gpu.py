@@ -60,16 +60,84 @@ def test_graphed_step_matches_eager():
         eng.run(10)
         runs[graphs] = eng.global_flat.clone()
         if graphs:  # the graph path must actually have been taken
-            assert eng._stepper is not None, \
+            assert len(eng._steppers) > 0, \
                 "hipGraph capture fell back to eager on the GPU"
     assert torch.allclose(runs[True], runs[False], atol=1e-6, rtol=1e-5)
 
 
-def test_graphed_step_off_for_adam():
+def test_graphed_adam_matches_eager():
+    """Graphed Adam (device step counter + on-device bias corrections,
+    csrc/hip/elementwise.hip adam_tick/adam_master_dev) must track the
+    eager fused-Adam path (VERDICT round-1 item 10)."""
     from bflc_amd.config import FLConfig
-    cfg = FLConfig.for_world(1, model="mlp", n_features=16, n_class=4,
-                             samples_per_client=256, batch_size=128,
-                             eval_samples=128, optimizer="adam")
-    eng = _engine(cfg)
-    eng.run(3)
-    assert eng._stepper is None
+    base = dict(model="femnist_cnn", n_class=62, samples_per_client=512,
+                batch_size=256, eval_samples=512, partition="dirichlet",
+                learning_rate=0.001, optimizer="adam")
+    runs = {}
+    for graphs in (True, False):
+        cfg = FLConfig.for_world(1, use_graphs=graphs, **base)
+        eng = _engine(cfg)
+        eng.run(5)
+        runs[graphs] = eng.global_flat.clone()
+        if graphs:
+            assert len(eng._steppers) > 0, \
+                "graphed Adam fell back to eager on the GPU"
+    # identical math modulo device-vs-host powf for the bias corrections
+    assert torch.allclose(runs[True], runs[False], atol=1e-4, rtol=1e-4)
+
+
+def test_adam_engine_matches_cpu_oracle():
+    """Fused adam_kernel rounds (GPU, fp32 compute so dtype matches)
+    against the engine's CPU Adam oracle: same seed, same rounds, same
+    weights (VERDICT round-1 item 10)."""
+    from bflc_amd.comm import Transport
+    from bflc_amd.config import FLConfig
+    from bflc_amd.data import make_federated
+    from bflc_amd.fl import FLEngine
+    base = dict(model="mlp", n_features=16, n_class=4,
+                samples_per_client=256, batch_size=128, eval_samples=128,
+                optimizer="adam", dtype="fp32", learning_rate=0.001)
+    outs = {}
+    for dev in (DEV, torch.device("cpu")):
+        cfg = FLConfig.for_world(1, **base)
+        shards, test = make_federated(cfg)
+        eng = FLEngine(cfg, Transport(device=dev), shards, test)
+        eng.run(3)
+        outs[dev.type] = eng.global_flat.cpu()
+    assert torch.allclose(outs["cuda"], outs["cpu"], atol=1e-5, rtol=1e-5)
+
+
+def test_byzantine_defense_on_gpu():
+    """BASELINE config 4 on hardware: 2 of 8 clients label-flip; the
+    committee-score defense must (a) never select an attacker update and
+    (b) beat the attacked-FedAvg control (top-k widened to admit every
+    update) on final accuracy (VERDICT round-1 item 7)."""
+    from bflc_amd.config import FLConfig
+    from bflc_amd.data import make_federated
+    from bflc_amd.fl import FLEngine
+    from bflc_amd.comm import Transport
+
+    base = dict(client_num=8, comm_count=2, needed_update_count=6,
+                byzantine_clients=2, model="mlp", n_features=32,
+                n_class=8, samples_per_client=256, batch_size=128,
+                eval_samples=1024, learning_rate=0.05)
+    attackers = {"node_6", "node_7"}
+
+    def run(agg):
+        cfg = FLConfig(aggregate_count=agg, **base)
+        shards, test = make_federated(cfg)
+        assert shards[6].byzantine and shards[7].byzantine
+        eng = FLEngine(cfg, Transport(device=DEV), shards, test)
+        picked = []
+        for _ in range(12):
+            eng.run_round()
+            picked += [o for o, _ in eng.last_decision.selected]
+        return eng.evaluate_global(), picked
+
+    defense_acc, defense_sel = run(agg=3)   # top-3 of 6: honest margin
+    control_acc, control_sel = run(agg=6)   # everything aggregates
+    assert not (set(defense_sel) & attackers), \
+        f"attacker selected under defense: {set(defense_sel) & attackers}"
+    assert set(control_sel) & attackers  # control really is attacked
+    assert defense_acc > 0.9
+    assert defense_acc > control_acc + 0.04, (defense_acc, control_acc)
