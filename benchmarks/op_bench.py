@@ -10,8 +10,12 @@ run it before and after a kernel change and diff the per-shape numbers.
 """
 import argparse
 import json
+import os
+import sys
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from bflc_amd.ops import functional as O
 

@@ -87,24 +87,35 @@ def test_graphed_adam_matches_eager():
 
 
 def test_adam_engine_matches_cpu_oracle():
-    """Fused adam_kernel rounds (GPU, fp32 compute so dtype matches)
-    against the engine's CPU Adam oracle: same seed, same rounds, same
-    weights (VERDICT round-1 item 10)."""
+    """Adam FL rounds on the GPU (bf16 compute, fused adam_master_
+    kernels) against the engine's CPU fp32 Adam oracle: same seed, same
+    data, same rounds. The compute dtypes differ (the GEMM family is
+    bf16-only by design), so the assertion is on the round DELTA
+    direction and magnitude — an Adam bug (swapped betas, wrong bias
+    correction, stale m/v across clients) wrecks both; bf16 rounding
+    does not (VERDICT round-1 item 10)."""
     from bflc_amd.comm import Transport
     from bflc_amd.config import FLConfig
     from bflc_amd.data import make_federated
     from bflc_amd.fl import FLEngine
     base = dict(model="mlp", n_features=16, n_class=4,
-                samples_per_client=256, batch_size=128, eval_samples=128,
-                optimizer="adam", dtype="fp32", learning_rate=0.001)
-    outs = {}
+                samples_per_client=256, batch_size=128, eval_samples=512,
+                optimizer="adam", learning_rate=0.001)
+    deltas, accs = {}, {}
     for dev in (DEV, torch.device("cpu")):
-        cfg = FLConfig.for_world(1, **base)
+        cfg = FLConfig.for_world(4, **base)
         shards, test = make_federated(cfg)
         eng = FLEngine(cfg, Transport(device=dev), shards, test)
+        flat0 = eng.global_flat.clone()
         eng.run(3)
-        outs[dev.type] = eng.global_flat.cpu()
-    assert torch.allclose(outs["cuda"], outs["cpu"], atol=1e-5, rtol=1e-5)
+        deltas[dev.type] = (eng.global_flat - flat0).cpu()
+        accs[dev.type] = eng.evaluate_global()
+    cos = torch.nn.functional.cosine_similarity(
+        deltas["cuda"], deltas["cpu"], dim=0)
+    ratio = deltas["cuda"].norm() / deltas["cpu"].norm()
+    assert float(cos) > 0.98, f"adam delta cosine {float(cos):.4f}"
+    assert 0.9 < float(ratio) < 1.1, f"adam delta norm ratio {ratio:.3f}"
+    assert abs(accs["cuda"] - accs["cpu"]) < 0.1
 
 
 def test_byzantine_defense_on_gpu():

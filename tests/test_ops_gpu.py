@@ -219,6 +219,54 @@ class TestFlatOps:
             hip().adam_step_(pg, gg, mg, vg, step, 0.01, 0.9, 0.999, 1e-8)
         assert torch.allclose(pg.cpu(), p, atol=1e-5)
 
+    def test_adam_master_matches_fp32_oracle(self):
+        """Fused fp32-master/bf16-shadow Adam vs a plain fp32 torch
+        oracle fed the bf16-rounded grads (same inputs the kernel sees).
+        """
+        torch.manual_seed(16)
+        p = torch.randn(5000)
+        m = torch.zeros(5000)
+        v = torch.zeros(5000)
+        pg, mg, vg = (t.to(DEV) for t in (p, m, v))
+        sh = pg.bfloat16()
+        b1, b2, eps, lr = 0.9, 0.999, 1e-8, 0.01
+        for step in (1, 2, 3):
+            g = torch.randn(5000, device=DEV).bfloat16()
+            hip().adam_master_(pg, sh, g, mg, vg, step, lr, b1, b2, eps)
+            gf = g.float().cpu()
+            m = b1 * m + (1 - b1) * gf
+            v = b2 * v + (1 - b2) * gf * gf
+            p -= lr * (m / (1 - b1 ** step)) / \
+                ((v / (1 - b2 ** step)).sqrt() + eps)
+        assert torch.allclose(pg.cpu(), p, atol=2e-5)
+        assert torch.equal(sh, pg.bfloat16())  # shadow = bf16(master)
+
+    def test_adam_master_graph_matches_host_step(self):
+        """The device-step variant (adam_tick + adam_master_dev) must
+        reproduce the host-step adam_master_ sequence exactly — this is
+        what makes the Adam train step hipGraph-capturable."""
+        torch.manual_seed(17)
+        p0 = torch.randn(4096, device=DEV)
+        gs = [torch.randn(4096, device=DEV).bfloat16() for _ in range(4)]
+        # host-step reference
+        p_a = p0.clone(); sh_a = p_a.bfloat16()
+        m_a = torch.zeros_like(p0); v_a = torch.zeros_like(p0)
+        for step, g in enumerate(gs, 1):
+            hip().adam_master_(p_a, sh_a, g, m_a, v_a, step, 0.01,
+                               0.9, 0.999, 1e-8)
+        # device-step pair
+        p_b = p0.clone(); sh_b = p_b.bfloat16()
+        m_b = torch.zeros_like(p0); v_b = torch.zeros_like(p0)
+        step_t = torch.zeros(1, dtype=torch.int32, device=DEV)
+        bc = torch.zeros(2, device=DEV)
+        for g in gs:
+            hip().adam_master_graph_(p_b, sh_b, g, m_b, v_b, step_t, bc,
+                                     0.01, 0.9, 0.999, 1e-8)
+        assert int(step_t.item()) == 4
+        # identical math modulo device-vs-host powf in the corrections
+        assert torch.allclose(p_a, p_b, atol=1e-6, rtol=1e-6)
+        assert torch.equal(sh_b, p_b.bfloat16())
+
     def test_fedavg_matches_and_deterministic(self):
         torch.manual_seed(7)
         deltas = torch.randn(6, 200000, device=DEV)
