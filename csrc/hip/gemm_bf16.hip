@@ -1911,11 +1911,16 @@ bool gemm_conv_wgrad_raw(const torch::Tensor& dy2, const torch::Tensor& x,
     part_ptr = part.data_ptr<float>();
   }
   dim3 grid((unsigned)tiles, (unsigned)S);
-  // BFLC_WGRAD_KPAIR=1: k-pair staging experiment (CMODE 4), default
-  // off until GPU-measured — values are bit-identical either way.
+  // k-pair scatter staging (CMODE 4): each LDS slot owns two adjacent
+  // k's of one n-granule, halving ds_write instructions in the staging
+  // loop. Default ON since the round-2 A/B on MI355X: every wgrad-heavy
+  // conv bwd shape improved (-5..-11% in benchmarks/op_bench.py,
+  // profiles/r02_kpair_ab.md; ResNet-20 round 92.8 -> 90.2 ms) with
+  // bit-identical values (51 conv/resnet GPU tests pass under either
+  // mode). BFLC_WGRAD_KPAIR=0 restores the old staging for A/B.
   static const bool kpair = [] {
     const char* e = getenv("BFLC_WGRAD_KPAIR");
-    return e && atoi(e) != 0;
+    return !e || atoi(e) != 0;
   }();
   auto launchw = [&](auto bm, auto bn, auto wr, auto wc) {
     constexpr int BMv = decltype(bm)::value, BNv = decltype(bn)::value;
