@@ -128,6 +128,12 @@ class FLEngine:
         self._steppers: Dict[tuple, object] = {}
         self._scorers: Dict[int, object] = {}
         self._graph_pool = None  # shared across all captures (one pool)
+        # round-persistent scratch (avoid per-round alloc/fill of
+        # O(model)-sized tensors: the gather stack is ~400 MB/round on
+        # ResNet-50)
+        self._cand_buf: Optional[torch.Tensor] = None
+        self._stack_buf: Optional[torch.Tensor] = None
+        self._fedavg_buf: Optional[torch.Tensor] = None
         self._use_graphs = (cfg.use_graphs and self.device.type == "cuda"
                             and os.environ.get("BFLC_GRAPHS", "1") != "0")
 
@@ -231,14 +237,22 @@ class FLEngine:
         local shard."""
         shard = self.shards[scorer]
         gs = self._graphed_scorer(scorer) if self._use_graphs else None
+        lr = self.cfg.learning_rate
         accs: List[Tuple[str, torch.Tensor]] = []
-        for origin, delta in updates:
-            cand = self.global_flat.clone()
-            O.axpy_(cand, -self.cfg.learning_rate, delta)
-            if gs is not None:
-                accs.append((origin, gs.score(cand)))
-            else:
-                self.model.set_flat(cand)
+        if gs is not None:
+            for origin, delta in updates:
+                # candidate built in the graph's own input buffer — no
+                # per-candidate clone
+                accs.append((origin,
+                             gs.score_candidate(self.global_flat, lr,
+                                                delta)))
+        else:
+            if self._cand_buf is None:
+                self._cand_buf = torch.empty_like(self.global_flat)
+            for origin, delta in updates:
+                self._cand_buf.copy_(self.global_flat)
+                O.axpy_(self._cand_buf, -lr, delta)
+                self.model.set_flat(self._cand_buf)
                 # device-resident: ONE host sync per scorer (below),
                 # not one per candidate
                 accs.append((origin,
@@ -326,8 +340,13 @@ class FLEngine:
         all_metas = self._gather("updates", self.t.all_gather_objects, metas)
         max_subs = max((len(m) for m in all_metas), default=0)
         P = self.global_flat.numel()
-        stack = torch.zeros(max(max_subs, 1), P, dtype=torch.float32,
-                            device=self.global_flat.device)
+        rows = max(max_subs, 1)
+        # cached, never zeroed: rows beyond a rank's own submission count
+        # are never read back (unpack walks rank_metas, not the padding)
+        if self._stack_buf is None or self._stack_buf.shape[0] < rows:
+            self._stack_buf = torch.empty(rows, P, dtype=torch.float32,
+                                          device=self.global_flat.device)
+        stack = self._stack_buf[:rows]
         for j, (_, delta, _, _) in enumerate(local_updates):
             stack[j] = delta
         gathered = self._gather("update-deltas", self.t.all_gather_tensor,
@@ -386,8 +405,11 @@ class FLEngine:
         K = len(sel)
         self.last_decision = decision
         if K > 0:
-            deltas = torch.empty(K, P, dtype=torch.float32,
-                                 device=self.global_flat.device)
+            if self._fedavg_buf is None or self._fedavg_buf.shape[0] < K:
+                self._fedavg_buf = torch.empty(
+                    K, P, dtype=torch.float32,
+                    device=self.global_flat.device)
+            deltas = self._fedavg_buf[:K]
             weights = torch.empty(K, dtype=torch.float32,
                                   device=self.global_flat.device)
             for k, (origin, w) in enumerate(sel):

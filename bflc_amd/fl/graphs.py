@@ -164,3 +164,15 @@ class GraphedScore:
         self.cand.copy_(cand)
         self.graph.replay()
         return self.acc.clone()
+
+    def score_candidate(self, global_flat: torch.Tensor, lr: float,
+                        delta: torch.Tensor) -> torch.Tensor:
+        """Build candidate = global - lr*delta DIRECTLY in the graph's
+        input buffer (reference main.py:215-216) — no per-candidate
+        clone+copy round trip (the engine's scoring loop runs
+        committee x quota of these per round)."""
+        from bflc_amd.ops import functional as O
+        self.cand.copy_(global_flat)
+        O.axpy_(self.cand, -lr, delta)
+        self.graph.replay()
+        return self.acc.clone()
