@@ -61,7 +61,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          long stride, long pad);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
-    long pad, c10::optional<torch::Tensor> col_cache);
+    long pad, c10::optional<torch::Tensor> col_cache, bool want_db);
 std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
                                                        long kernel,
                                                        long stride);
@@ -125,9 +125,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("invstd"), py::arg("relu"),
         py::arg("residual") = py::none(),
         "normalization pass with known stats");
-  m.def("conv2d_bwd", &bflc::conv2d_bwd, "(dx, dw, db)",
+  m.def("conv2d_bwd", &bflc::conv2d_bwd,
+        "(dx, dw, db) - db empty unless want_db",
         py::arg("x"), py::arg("w"), py::arg("dy"), py::arg("stride"),
-        py::arg("pad"), py::arg("col_cache") = py::none());
+        py::arg("pad"), py::arg("col_cache") = py::none(),
+        py::arg("want_db") = true);
   m.def("maxpool2d_fwd", &bflc::maxpool2d_fwd);
   m.def("maxpool2d_bwd", &bflc::maxpool2d_bwd);
   m.def("batchnorm_fwd", &bflc::batchnorm_fwd,

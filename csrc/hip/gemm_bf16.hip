@@ -1382,6 +1382,7 @@ TileCfg pick_tile(long M, long N) {
 // the A/C streams instead. KT is compile-time so the A row and the dot
 // fully unroll in registers.
 typedef __attribute__((ext_vector_type(8))) float f32x8v_t;
+typedef __attribute__((ext_vector_type(2))) float f32x2v_t;
 
 template <int KT>
 __global__ __launch_bounds__(256) void gemm_thin_kernel(
@@ -1406,21 +1407,34 @@ __global__ __launch_bounds__(256) void gemm_thin_kernel(
   long m = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (; m < M; m += stride) {
-    float a[KT];
+    // A row pre-DUPLICATED into aligned fp32 pairs: v_pk_fma_f32 wants
+    // a 64-bit (2xf32) src for the broadcast operand, and with a plain
+    // float a[KT] the compiler re-materialized the (a[k], a[k]) pair
+    // INSIDE the n8 loop — 120 v_mov_b32 per 64 v_pk_fma, 2/3 of the
+    // VALU slots burned on operand shuffles (profiles/r01_kernel_
+    // resources.md). Duplicating once per row hoists all of it: the
+    // same body now compiles to 8 v_mov per 64 v_pk_fma.
+    f32x2v_t a2[KT];
     const bf16* arow = &A[m * KT];
 #pragma unroll
     for (int k8 = 0; k8 < KT; k8 += 8) {
       const bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(&arow[k8]);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) a[k8 + j] = b2f(v[j]);
+      for (int j = 0; j < 8; ++j) {
+        const float f = b2f(v[j]);
+        a2[k8 + j] = f32x2v_t{f, f};
+      }
     }
     for (int n8 = 0; n8 < N; n8 += 8) {
       f32x8v_t acc = *reinterpret_cast<const f32x8v_t*>(&bl[n8]);
+      f32x2v_t* acc2 = reinterpret_cast<f32x2v_t*>(&acc);
 #pragma unroll
       for (int k = 0; k < KT; ++k) {
         const f32x8v_t bv =
             *reinterpret_cast<const f32x8v_t*>(&Bl[k * 64 + n8]);
-        acc += bv * a[k];
+        const f32x2v_t* bv2 = reinterpret_cast<const f32x2v_t*>(&bv);
+#pragma unroll
+        for (int p = 0; p < 4; ++p) acc2[p] += bv2[p] * a2[k];
       }
       bf16x8_t out;
 #pragma unroll
