@@ -81,6 +81,9 @@ def main():
         dy = torch.randn_like(y)
         col_arg = col if col.numel() else None
         out[f"conv.{name}.fwd"] = timeit(
+            lambda: h_ops.conv2d_fwd_col(x, w, b, stride, pad, False,
+                                         True), args.iters)
+        out[f"conv.{name}.fwd_eval"] = timeit(
             lambda: h_ops.conv2d_fwd(x, w, b, stride, pad), args.iters)
         out[f"conv.{name}.bwd"] = timeit(
             lambda: h_ops.conv2d_bwd(x, w, dy, stride, pad, col_arg),

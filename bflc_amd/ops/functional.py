@@ -167,10 +167,15 @@ class _Conv2dFn(torch.autograd.Function):
         ctx.has_bias = b is not None
         ctx.relu = relu
         if x.is_cuda:
+            # grad-free forwards (committee scoring / sponsor eval) skip
+            # the col materialization entirely — thin shapes gather the
+            # window inside the GEMM (gemm_thin_conv_kernel), identical
+            # output, no 2x O(M*RSC) col traffic
+            want_col = torch.is_grad_enabled()
             y, col = hip_ops().conv2d_fwd_col(
                 x, w, b if b is not None else
                 torch.zeros(w.shape[0], device=x.device, dtype=x.dtype),
-                stride, padding, relu)
+                stride, padding, relu, want_col)
             # keep col for wgrad when the fwd materialized one (the
             # implicit-GEMM path returns an empty marker; bwd then
             # builds its own)

@@ -46,7 +46,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
 // conv_im2col.hip
 std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
     torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride, long pad,
-    bool relu);
+    bool relu, bool want_col);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 conv2d_fwd_bn(torch::Tensor x, torch::Tensor w, long stride, long pad);
 std::tuple<torch::Tensor, torch::Tensor> bn_stats(torch::Tensor x,
@@ -58,7 +58,7 @@ torch::Tensor batchnorm_norm(torch::Tensor x, torch::Tensor gamma,
                              torch::Tensor invstd, bool relu,
                              c10::optional<torch::Tensor> residual);
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
-                         long stride, long pad);
+                         long stride, long pad, bool relu);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
     long pad, c10::optional<torch::Tensor> col_cache, bool want_db);
@@ -111,10 +111,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_raw", &bflc::gemm_raw, "raw GEMM (bench/ablation)");
   m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");
   m.def("conv2d_fwd", &bflc::conv2d_fwd,
-        "NHWC implicit-GEMM convolution (MFMA)");
-  m.def("conv2d_fwd_col", &bflc::conv2d_fwd_col, "(y, col) - col for bwd",
+        "NHWC implicit-GEMM convolution (MFMA); inference entry, no col",
         py::arg("x"), py::arg("w"), py::arg("b"), py::arg("stride"),
         py::arg("pad"), py::arg("relu") = false);
+  m.def("conv2d_fwd_col", &bflc::conv2d_fwd_col, "(y, col) - col for bwd",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("stride"),
+        py::arg("pad"), py::arg("relu") = false,
+        py::arg("want_col") = true);
   m.def("conv2d_fwd_bn", &bflc::conv2d_fwd_bn,
         "(y, col, psum, psq) - conv with fused epilogue BN stats");
   m.def("bn_stats", &bflc::bn_stats, "(mean, invstd) of x [.., C]");

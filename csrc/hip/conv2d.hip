@@ -457,7 +457,7 @@ int dgrad_implicit_max_c() {
 // cache free). For 1x1 stride-1 convs col is just a VIEW of x.
 std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
     torch::Tensor x, torch::Tensor w, torch::Tensor b, long stride,
-    long pad, bool relu) {
+    long pad, bool relu, bool want_col) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv: bf16 only");
   auto sh = make_shape(x, w, stride, pad);
@@ -475,16 +475,26 @@ std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
   // no col matrix exists (wgrad materializes its own in the backward).
   if (gemm_conv_fwd_raw(x, w2, y, sh, &bc, relu))
     return {y, torch::empty({0}, x.options())};
+  const long kp = (sh.RSC() + 7) / 8 * 8;
+  // grad-free forward (committee scoring): nobody reads col back, so
+  // thin shapes gather the window inside the GEMM instead of paying
+  // the im2col write + re-read (bitwise-identical output)
+  if (!want_col && kp <= 32) {
+    auto w2p = pad_w2(w2, kp);
+    if (gemm_thin_conv_raw(x, w2p, y, sh, &bc, relu))
+      return {y, torch::empty({0}, x.options())};
+  }
   auto col = im2col(x, sh);
-  const long kp = col.size(1);  // may be K-padded to %8 (C<8 convs)
+  TORCH_CHECK(col.size(1) == kp);
   gemm_bf16_raw(col, pad_w2(w2, kp), y, sh.M(), sh.Kout, kp, false, true,
                 &bc, relu, EpStore::kPlain, 0);
   return {y, col};
 }
 
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
-                         long stride, long pad) {
-  return std::get<0>(conv2d_fwd_col(x, w, b, stride, pad, false));
+                         long stride, long pad, bool relu) {
+  // inference entry: no col wanted
+  return std::get<0>(conv2d_fwd_col(x, w, b, stride, pad, relu, false));
 }
 
 // Conv forward WITH fused per-tile BN stats from the GEMM epilogue:

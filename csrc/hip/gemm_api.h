@@ -46,6 +46,12 @@ void gemm_bf16_raw(const torch::Tensor& A, const torch::Tensor& B,
 // the im2col gather fused into the GEMM's A staging (no col matrix).
 // Requires sh.C % 8 == 0 and Kout % 64 == 0; returns false if the
 // shape cannot take this path (caller materializes col instead).
+// col-free thin conv forward (grad-free path): A gathered from the
+// NHWC window, no im2col matrix. w2p is the KT-padded weight.
+bool gemm_thin_conv_raw(const torch::Tensor& x, const torch::Tensor& w2p,
+                        torch::Tensor& y, const ConvShape& sh,
+                        const torch::Tensor* bias, bool relu);
+
 bool gemm_conv_fwd_raw(const torch::Tensor& x, const torch::Tensor& w2,
                        torch::Tensor& y, const ConvShape& sh,
                        const torch::Tensor* bias, bool relu,

@@ -1448,7 +1448,106 @@ __global__ __launch_bounds__(256) void gemm_thin_kernel(
   }
 }
 
+// Implicit-gather variant for grad-free conv forwards (committee
+// scoring runs 16 of the ~28 forwards per FL round and never reads the
+// col buffer back): the A row is gathered straight from the NHWC input
+// window instead of a materialized im2col matrix — for FEMNIST conv1
+// that removes a 77 MB col write + 77 MB re-read per forward and keeps
+// only the (L2-resident) x reads. Same dot order as the col-backed
+// kernel over the same KT-padded taps => bitwise-identical output.
+template <int KT>
+__global__ __launch_bounds__(256) void gemm_thin_conv_kernel(
+    const bf16* __restrict__ X, const bf16* __restrict__ B,
+    bf16* __restrict__ C, const bf16* __restrict__ bias, ConvShape sh,
+    long M, int N, int relu) {
+  __shared__ float Bl[32 * 64];
+  __shared__ float bl[64];
+  for (int i = threadIdx.x; i < N * KT; i += blockDim.x)
+    Bl[(i % KT) * 64 + i / KT] = b2f(B[i]);
+  for (int i = threadIdx.x; i < N; i += blockDim.x)
+    bl[i] = bias ? b2f(bias[i]) : 0.f;
+  __syncthreads();
+  const int OW = sh.OW, OHW = sh.OH * sh.OW;
+  long m = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; m < M; m += stride) {
+    const int n = (int)(m / OHW);
+    const int rem = (int)(m % OHW);
+    const int oh = rem / OW, ow = rem % OW;
+    const int ih0 = oh * sh.stride - sh.pad;
+    const int iw0 = ow * sh.stride - sh.pad;
+    f32x2v_t a2[KT];
+#pragma unroll
+    for (int t = 0; t < KT; ++t) a2[t] = f32x2v_t{0.f, 0.f};
+    int t = 0;
+    for (int r = 0; r < sh.R; ++r) {
+      const int ih = ih0 + r;
+      const bool okh = ih >= 0 && ih < sh.H;
+      for (int s = 0; s < sh.S; ++s) {
+        const int iw = iw0 + s;
+        const bool ok = okh && iw >= 0 && iw < sh.W;
+        const long base =
+            (((long)n * sh.H + ih) * sh.W + iw) * sh.C;
+        for (int c = 0; c < sh.C; ++c, ++t) {
+          const float f = ok ? b2f(X[base + c]) : 0.f;
+          a2[t] = f32x2v_t{f, f};
+        }
+      }
+    }
+    for (int n8 = 0; n8 < N; n8 += 8) {
+      f32x8v_t acc = *reinterpret_cast<const f32x8v_t*>(&bl[n8]);
+      f32x2v_t* acc2 = reinterpret_cast<f32x2v_t*>(&acc);
+#pragma unroll
+      for (int k = 0; k < KT; ++k) {
+        const f32x8v_t bv =
+            *reinterpret_cast<const f32x8v_t*>(&Bl[k * 64 + n8]);
+        const f32x2v_t* bv2 = reinterpret_cast<const f32x2v_t*>(&bv);
+#pragma unroll
+        for (int p = 0; p < 4; ++p) acc2[p] += bv2[p] * a2[k];
+      }
+      bf16x8_t out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = acc[j];
+        if (relu && v < 0.f) v = 0.f;
+        out[j] = f2b(v);
+      }
+      *reinterpret_cast<bf16x8_t*>(&C[m * N + n8]) = out;
+    }
+  }
+}
+
 }  // namespace
+
+// Col-free thin conv fwd: true if this shape is handled (the caller
+// falls back to im2col + GEMM otherwise). w2p must already be KT-padded
+// ([Kout][KT], KT = RSC rounded up to 8).
+bool gemm_thin_conv_raw(const torch::Tensor& x, const torch::Tensor& w2p,
+                        torch::Tensor& y, const ConvShape& sh,
+                        const torch::Tensor* bias, bool relu) {
+  const long M = sh.M();
+  const long KT = w2p.size(1);
+  const long N = sh.Kout;
+  if (KT > 32 || KT % 8 != 0 || N > 64 || N % 8 != 0 || M < 65536)
+    return false;
+  const bf16* bs = bias ? (const bf16*)bias->data_ptr() : nullptr;
+  const int blocks = (int)std::min<long>((M + 255) / 256, 16384);
+  auto launch = [&](auto kv) {
+    hipLaunchKernelGGL((gemm_thin_conv_kernel<decltype(kv)::value>),
+                       dim3(blocks), dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(),
+                       (const bf16*)w2p.data_ptr(), (bf16*)y.data_ptr(),
+                       bs, sh, M, (int)N, relu ? 1 : 0);
+  };
+  switch (KT) {
+    case 8: launch(std::integral_constant<int, 8>{}); break;
+    case 16: launch(std::integral_constant<int, 16>{}); break;
+    case 24: launch(std::integral_constant<int, 24>{}); break;
+    default: launch(std::integral_constant<int, 32>{}); break;
+  }
+  HIP_CHECK(hipGetLastError());
+  return true;
+}
 
 torch::Tensor transpose_bf16(const torch::Tensor& X) {
   const long R = X.size(0), C = X.size(1);

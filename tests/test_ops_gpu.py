@@ -219,6 +219,31 @@ class TestFlatOps:
             hip().adam_step_(pg, gg, mg, vg, step, 0.01, 0.9, 0.999, 1e-8)
         assert torch.allclose(pg.cpu(), p, atol=1e-5)
 
+    def test_thin_conv_implicit_matches_col_path(self):
+        """The grad-free conv forward (gemm_thin_conv_kernel: window
+        gathered inside the GEMM, no im2col materialization) must be
+        BITWISE equal to the col-backed path — same taps, same KT
+        padding, same dot order."""
+        torch.manual_seed(11)
+        # M = N*OH*OW must clear the 65536 implicit gate (first four);
+        # the small fifth shape checks the im2col fallback stays wired
+        for (n, h, c, k, r, stride, pad) in [
+            (96, 28, 1, 32, 3, 1, 1),    # FEMNIST conv1 (C=1, kp=16)
+            (96, 32, 3, 16, 3, 1, 1),    # CIFAR stem (C=3, kp=32)
+            (512, 28, 1, 32, 3, 2, 1),   # strided
+            (128, 28, 1, 32, 3, 1, 0),   # no padding
+            (16, 28, 1, 32, 3, 1, 1),    # below the gate: fallback
+        ]:
+            x = torch.randn(n, h, h, c, device=DEV).bfloat16()
+            w = (torch.randn(k, r, r, c, device=DEV) * 0.1).bfloat16()
+            b = torch.randn(k, device=DEV).bfloat16()
+            for relu in (False, True):
+                y_col, col = hip().conv2d_fwd_col(x, w, b, stride, pad,
+                                                  relu, True)
+                y_imp = hip().conv2d_fwd(x, w, b, stride, pad, relu)
+                assert torch.equal(y_col, y_imp), \
+                    (n, h, c, k, stride, pad, relu)
+
     def test_adam_master_matches_fp32_oracle(self):
         """Fused fp32-master/bf16-shadow Adam vs a plain fp32 torch
         oracle fed the bf16-rounded grads (same inputs the kernel sees).
