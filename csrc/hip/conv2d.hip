@@ -478,8 +478,11 @@ std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
   const long kp = (sh.RSC() + 7) / 8 * 8;
   // grad-free forward (committee scoring): nobody reads col back, so
   // thin shapes gather the window inside the GEMM instead of paying
-  // the im2col write + re-read (bitwise-identical output)
-  if (!want_col && kp <= 32) {
+  // the im2col write + re-read (bitwise-identical output). RSC <= 16
+  // only: at 27 scalar taps (CIFAR stem) the per-row gather costs more
+  // than the col round trip it saves (A/B: stem eval 53 -> 66 us,
+  // FEMNIST conv1 9 taps 50.7 -> 35.4 us).
+  if (!want_col && sh.RSC() <= 16 && kp <= 32) {
     auto w2p = pad_w2(w2, kp);
     if (gemm_thin_conv_raw(x, w2p, y, sh, &bc, relu))
       return {y, torch::empty({0}, x.options())};
