@@ -25,6 +25,8 @@ void adam_master_graph_(torch::Tensor p, torch::Tensor shadow,
 void refresh_shadow_(torch::Tensor p, torch::Tensor shadow);
 torch::Tensor relu_fwd(torch::Tensor x);
 torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy);
+std::tuple<torch::Tensor, torch::Tensor> relu_bwd_colsum(
+    const torch::Tensor& y, const torch::Tensor& dy);
 
 // softmax_ce.hip
 std::tuple<torch::Tensor, torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
@@ -100,6 +102,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("refresh_shadow_", &bflc::refresh_shadow_, "shadow = bf16(master)");
   m.def("relu_fwd", &bflc::relu_fwd);
   m.def("relu_bwd", &bflc::relu_bwd);
+  m.def("relu_bwd_colsum", &bflc::relu_bwd_colsum,
+        "fused relu backward + bias-grad column sum: (dx, db)");
   m.def("softmax_ce_fwd", &bflc::softmax_ce_fwd,
         "fused softmax cross-entropy fwd (loss, probs)");
   m.def("softmax_ce_bwd", &bflc::softmax_ce_bwd);

@@ -73,6 +73,9 @@ bool gemm_conv_wgrad_raw(const torch::Tensor& dy2, const torch::Tensor& x,
 
 // colsum: out[n] = sum_m X[m,n]  (bias gradient)
 torch::Tensor colsum_bf16(const torch::Tensor& X);
+// fused relu-backward + bias-grad column sum: (dx, db)
+std::tuple<torch::Tensor, torch::Tensor> relu_bwd_colsum(
+    const torch::Tensor& y, const torch::Tensor& dy);
 
 // LDS-tiled bf16 transpose: out[C][R] = X[R][C]^T (16-B coalesced both
 // sides). Used to put GEMM operands into the vector-staging layout.

@@ -244,6 +244,21 @@ class TestFlatOps:
                 assert torch.equal(y_col, y_imp), \
                     (n, h, c, k, stride, pad, relu)
 
+    def test_relu_bwd_colsum_matches_separate_ops(self):
+        """Fused relu-backward+colsum == relu_bwd then the fp32 oracle
+        colsum (bitwise dx; db vs fp32 oracle within bf16-out rounding)."""
+        torch.manual_seed(12)
+        for (m, n) in [(100000, 32), (524288, 64), (4096, 8)]:
+            y = torch.randn(m, n, device=DEV).bfloat16()
+            dy = torch.randn(m, n, device=DEV).bfloat16()
+            dx_f, db_f = hip().relu_bwd_colsum(y, dy)
+            dx_ref = hip().relu_bwd(y, dy)
+            assert torch.equal(dx_f, dx_ref)
+            db_oracle = dx_ref.float().sum(0)
+            assert torch.allclose(db_f.float(), db_oracle,
+                                  atol=max(1.0, 3e-2 * m ** 0.5),
+                                  rtol=2e-2)
+
     def test_adam_master_matches_fp32_oracle(self):
         """Fused fp32-master/bf16-shadow Adam vs a plain fp32 torch
         oracle fed the bf16-rounded grads (same inputs the kernel sees).
