@@ -224,3 +224,61 @@ def test_bigblob_gather_roundtrip(tmp_path):
     f1 = lines[1].split("lens=")[1]
     assert f0 == f1
     assert "[102000000, 1234567]" in lines[0]
+
+
+# ---------------------------------------------------------------------------
+# The exact bench protocol shape (8 clients / committee 4 / quota 4 /
+# top-3) split over 4 ranks — the same client->rank split the driver's
+# N=4 scaling point uses; replicas must stay bitwise identical.
+
+BENCHSHAPE_WORKER = r"""
+import json, os, sys
+import torch
+sys.path.insert(0, {repo!r})
+from bflc_amd.config import FLConfig
+from bflc_amd.comm import Transport
+from bflc_amd.data import make_federated
+from bflc_amd.fl import FLEngine
+
+cfg = FLConfig.for_world(8, model="mlp", n_features=32, n_class=8,
+                         samples_per_client=96, batch_size=32,
+                         eval_samples=128, partition="dirichlet")
+assert (cfg.client_num, cfg.comm_count, cfg.needed_update_count,
+        cfg.aggregate_count) == (8, 4, 4, 3)
+shards, test = make_federated(cfg)
+t = Transport(backend="gloo", device=torch.device("cpu"))
+eng = FLEngine(cfg, t, shards, test)
+eng.run(4)
+out = {{"rank": t.rank, "epoch": eng.ledger.epoch,
+        "roles": eng.ledger.roles(),
+        "digest": torch.sum(eng.global_flat.double()).item(),
+        "local_clients": eng.local_clients}}
+with open(os.path.join({outdir!r}, f"rank{{t.rank}}.json"), "w") as f:
+    json.dump(out, f)
+t.barrier()
+t.close()
+"""
+
+
+def test_world4_bench_protocol_shape(tmp_path):
+    script = tmp_path / "w4.py"
+    script.write_text(BENCHSHAPE_WORKER.format(repo=REPO,
+                                               outdir=str(tmp_path)))
+    procs = []
+    for rank in range(4):
+        env = dict(os.environ, RANK=str(rank), WORLD_SIZE="4",
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT="29551",
+                   OMP_NUM_THREADS="2")
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE))
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        assert p.returncode == 0, err.decode()[-3000:]
+    rs = [json.load(open(tmp_path / f"rank{r}.json")) for r in range(4)]
+    assert all(r["epoch"] == 4 for r in rs)
+    assert all(r["digest"] == rs[0]["digest"] for r in rs)
+    assert all(r["roles"] == rs[0]["roles"] for r in rs)
+    # contiguous 2-client slices
+    assert [r["local_clients"] for r in rs] == \
+        [[0, 1], [2, 3], [4, 5], [6, 7]]
