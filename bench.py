@@ -94,7 +94,10 @@ def main() -> None:
         batch_size=bs, partition="dirichlet",
         dirichlet_alpha=0.3, eval_samples=min(4096, 2 * spc),
         learning_rate=0.01, byzantine_clients=args.byzantine,
-        optimizer=args.optimizer, use_graphs=not args.no_graphs)
+        optimizer=args.optimizer, use_graphs=not args.no_graphs,
+        # the ledger enforces max_epoch (reference main.py:251); size it
+        # to the requested run so long soaks don't trip the guard
+        max_epoch=max(1000, args.steps + args.warmup + 10))
 
     t = Transport()
     shards, test = make_federated(cfg)
