@@ -127,6 +127,13 @@ class FLEngine:
         # Adam, or if capture fails on this ROCm build.
         self._steppers: Dict[tuple, object] = {}
         self._scorers: Dict[int, object] = {}
+        # whole-phase graphs: one per client (full local training pass)
+        # and one per scorer (all quota candidates in one replay) —
+        # launch/copy overhead per round collapses to one dispatch per
+        # phase participant (fl/graphs.py)
+        self._train_graphs: Dict[int, object] = {}
+        self._score_graphs: Dict[int, object] = {}
+        self._cand_stack: Optional[torch.Tensor] = None
         self._graph_pool = None  # shared across all captures (one pool)
         # round-persistent scratch (avoid per-round alloc/fill of
         # O(model)-sized tensors: the gather stack is ~400 MB/round on
@@ -178,6 +185,28 @@ class FLEngine:
         return self._steppers[key]
 
     # ------------------------------------------------------------------
+    def _whole_train_graph(self, client: int):
+        """Per-client whole-local-training graph (shard slices baked in,
+        zero per-batch copies); None if capture is unavailable."""
+        if client in self._train_graphs:
+            return self._train_graphs[client]
+        try:
+            from bflc_amd.fl.graphs import GraphedLocalTrain
+            shard = self.shards[client]
+            g = GraphedLocalTrain(self.model, self.cfg.learning_rate,
+                                  shard.x, shard.y, self.cfg.batch_size,
+                                  self.cfg.local_epochs,
+                                  optimizer=self.cfg.optimizer,
+                                  pool=self._graph_pool)
+            if self._graph_pool is None:
+                self._graph_pool = g.pool()
+        except Exception as e:
+            warnings.warn(f"whole-train graph capture failed for client "
+                          f"{client}, using per-batch path: {e}")
+            g = None
+        self._train_graphs[client] = g
+        return g
+
     def _local_train(self, client: int) -> Tuple[torch.Tensor, int, float]:
         """Local train step (reference local_training, main.py:103-158):
         start from the global model, run local_epochs passes of
@@ -185,6 +214,17 @@ class FLEngine:
         cfg = self.cfg
         shard = self.shards[client]
         n = shard.n
+        # whole-pass graph: capture BEFORE loading the round's weights
+        # (capture executes real steps and mutates flat/cflat)
+        wg = self._whole_train_graph(client) if self._use_graphs else None
+        if wg is not None:
+            self.model.set_flat(self.global_flat)
+            cost = wg.run()
+            avg_cost = float(cost) / wg.n_steps
+            delta = self.global_flat.clone()
+            O.axpy_(delta, -1.0, self.model.flat.data)
+            delta.div_(cfg.learning_rate)
+            return delta, n, avg_cost
         bs = min(cfg.batch_size, n)
         total_batches = max(n // bs, 1)
         # capture BEFORE loading the round's weights: graph warmup and
@@ -229,12 +269,55 @@ class FLEngine:
         return delta, n, avg_cost
 
     # ------------------------------------------------------------------
+    def _pack_candidates(self, updates) -> bool:
+        """Pack the admitted deltas into the persistent [quota, P]
+        buffer the whole-scoring graphs bake in; False when the shape
+        doesn't fit (degraded round) => per-candidate fallback."""
+        K = len(updates)
+        if not self._use_graphs or K != self.cfg.needed_update_count \
+                or K == 0 or self.device.type != "cuda":
+            return False
+        if self._cand_stack is None:
+            self._cand_stack = torch.empty(
+                K, self.global_flat.numel(), dtype=torch.float32,
+                device=self.device)
+        for k, (_, delta) in enumerate(updates):
+            self._cand_stack[k].copy_(delta)
+        return True
+
+    def _phase_scorer(self, scorer: int):
+        """Per-scorer whole-phase scoring graph (all quota candidates in
+        one replay); None if capture is unavailable."""
+        if scorer in self._score_graphs:
+            return self._score_graphs[scorer]
+        try:
+            from bflc_amd.fl.graphs import GraphedScorePhase
+            shard = self.shards[scorer]
+            g = GraphedScorePhase(self.model, shard.x, shard.y,
+                                  self._cand_stack, self.global_flat,
+                                  self.cfg.learning_rate,
+                                  pool=self._graph_pool)
+            if self._graph_pool is None:
+                self._graph_pool = g.pool()
+        except Exception as e:
+            warnings.warn(f"whole-scoring graph capture failed for "
+                          f"scorer {scorer}, per-candidate path: {e}")
+            g = None
+        self._score_graphs[scorer] = g
+        return g
+
     def _score_candidates(self, scorer: int,
-                          updates: List[Tuple[str, torch.Tensor]]
-                          ) -> Dict[str, float]:
+                          updates: List[Tuple[str, torch.Tensor]],
+                          packed: bool = False) -> Dict[str, float]:
         """Committee scoring (reference local_scoring, main.py:196-217):
         candidate = W0 - lr*delta, scored by accuracy on the scorer's own
         local shard."""
+        if packed:
+            g = self._phase_scorer(scorer)
+            if g is not None:
+                accs = g.run().tolist()  # one host sync for all K
+                return {origin: accs[k]
+                        for k, (origin, _) in enumerate(updates)}
         shard = self.shards[scorer]
         gs = self._graphed_scorer(scorer) if self._use_graphs else None
         lr = self.cfg.learning_rate
@@ -371,10 +454,11 @@ class FLEngine:
         else:
             local_scorers = [i for i in self.local_clients
                              if roles.get(self.origins[i]) == "comm"]
+        packed = bool(local_scorers) and self._pack_candidates(updates)
         my_scores = []
         for i in local_scorers:
             o = self.origins[i]
-            smap = self._score_candidates(i, updates)
+            smap = self._score_candidates(i, updates, packed=packed)
             my_scores.append((o, smap, self.keys.sign(
                 "scores", o, epoch, scores_payload(smap))))
         if self.device.type == "cuda":
@@ -470,5 +554,8 @@ class FLEngine:
         ck = torch.load(path, map_location="cpu", weights_only=False)
         assert ck["config"]["model"] == self.cfg.model
         self.ledger.restore(ck["ledger"])
-        self.global_flat = ck["global_flat"].to(self.device)
+        # copy INTO the existing buffer: captured graphs bake its
+        # device pointer (global_flat must never reallocate)
+        with torch.no_grad():
+            self.global_flat.copy_(ck["global_flat"].to(self.device))
         self._round = ck["round"]

@@ -131,6 +131,130 @@ class GraphedTrainStep:
         self.graph.replay()
 
 
+class GraphedLocalTrain:
+    """A client's ENTIRE local training pass as one graph: every
+    minibatch step (forward, backward, fused optimizer, loss
+    accumulate) of `local_epochs x total_batches` captured back to back
+    with the client's OWN shard slices baked in (shard tensors are
+    engine-lifetime persistent, so no per-batch input copies at all).
+    One replay per client per round instead of one per batch — the
+    per-round launch/copy overhead of the train phase collapses to a
+    single dispatch."""
+
+    def __init__(self, model, lr: float, shard_x: torch.Tensor,
+                 shard_y: torch.Tensor, batch_size: int,
+                 local_epochs: int, optimizer: str = "sgd",
+                 pool=None) -> None:
+        self.model = model
+        self.optimizer = optimizer
+        self.cost = torch.zeros((), device=model.device,
+                                dtype=torch.float32)
+        n = shard_x.shape[0]
+        bs = min(batch_size, n)
+        self.total_batches = max(n // bs, 1)
+        self.n_steps = self.total_batches * local_epochs
+        if optimizer == "adam":
+            if model.cflat is model.flat:
+                raise RuntimeError("graphed Adam needs the bf16 shadow")
+            self.m = torch.zeros_like(model.flat)
+            self.v = torch.zeros_like(model.flat)
+            self.step_t = torch.zeros(1, device=model.device,
+                                      dtype=torch.int32)
+            self.bc = torch.zeros(2, device=model.device,
+                                  dtype=torch.float32)
+        elif optimizer != "sgd":
+            raise ValueError(optimizer)
+        self._lr = float(lr)
+        self._batches = [(shard_x[b * bs:(b + 1) * bs],
+                          shard_y[b * bs:(b + 1) * bs])
+                         for b in range(self.total_batches)] * local_epochs
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):
+                self._one_step(*self._batches[0])
+        torch.cuda.current_stream().wait_stream(side)
+        self.model.zero_grad()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph, pool=pool):
+            for xb, yb in self._batches:
+                self._one_step(xb, yb, accumulate=True)
+
+    def _optim_step(self) -> None:
+        if self.optimizer == "adam":
+            self.model.adam_step_graph(self.m, self.v, self.step_t,
+                                       self.bc, self._lr)
+        else:
+            self.model.sgd_step(self._lr)
+
+    def _one_step(self, xb, yb, accumulate: bool = False) -> None:
+        self.model.zero_grad()
+        loss = self.model.loss(xb, yb)
+        loss.backward()
+        self._optim_step()
+        if accumulate:
+            self.cost.add_(loss.detach().float())
+
+    def pool(self):
+        return self.graph.pool()
+
+    def run(self) -> torch.Tensor:
+        """One replay = the whole local training pass. Caller loads the
+        round's global weights first (set_flat). Returns the summed
+        per-batch loss (device scalar)."""
+        self.cost.zero_()
+        if self.optimizer == "adam":
+            self.m.zero_()
+            self.v.zero_()
+            self.step_t.zero_()
+        self.graph.replay()
+        return self.cost
+
+
+class GraphedScorePhase:
+    """One scorer's ENTIRE committee-scoring pass as one graph: for
+    each of the K quota slots, build candidate = global - lr*stack[k]
+    in place, load it, and evaluate accuracy on the scorer's shard —
+    K candidate evaluations and their accuracy reads in a single
+    replay. `cand_stack` must be the engine's persistent [K, P] packed
+    buffer (same device pointers every round)."""
+
+    def __init__(self, model, shard_x: torch.Tensor,
+                 shard_y: torch.Tensor, cand_stack: torch.Tensor,
+                 global_flat: torch.Tensor, lr: float, pool=None) -> None:
+        from bflc_amd.ops import functional as O
+        self.model = model
+        self.K = cand_stack.shape[0]
+        self.cand = torch.empty_like(model.flat)
+        self.accs = torch.zeros(self.K, device=model.device,
+                                dtype=torch.float32)
+        lr = float(lr)
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            self.cand.copy_(global_flat)
+            model.set_flat(self.cand)
+            model.accuracy_t(shard_x, shard_y)
+        torch.cuda.current_stream().wait_stream(side)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph, pool=pool):
+            for k in range(self.K):
+                self.cand.copy_(global_flat)
+                O.axpy_(self.cand, -lr, cand_stack[k])
+                model.set_flat(self.cand)
+                self.accs[k] = model.accuracy_t(shard_x, shard_y)
+
+    def pool(self):
+        return self.graph.pool()
+
+    def run(self) -> torch.Tensor:
+        """Replay; returns a snapshot of the K accuracies (device)."""
+        self.graph.replay()
+        return self.accs.clone()
+
+
 class GraphedScore:
     """Captured committee-scoring evaluation: load a candidate flat
     vector into the model (copy + shadow refresh) and run accuracy over
