@@ -60,7 +60,9 @@ def test_graphed_step_matches_eager():
         eng.run(10)
         runs[graphs] = eng.global_flat.clone()
         if graphs:  # the graph path must actually have been taken
-            assert len(eng._steppers) > 0, \
+            assert any(g is not None
+                       for g in eng._train_graphs.values()) \
+                or len(eng._steppers) > 0, \
                 "hipGraph capture fell back to eager on the GPU"
     assert torch.allclose(runs[True], runs[False], atol=1e-6, rtol=1e-5)
 
@@ -80,7 +82,9 @@ def test_graphed_adam_matches_eager():
         eng.run(5)
         runs[graphs] = eng.global_flat.clone()
         if graphs:
-            assert len(eng._steppers) > 0, \
+            assert any(g is not None
+                       for g in eng._train_graphs.values()) \
+                or len(eng._steppers) > 0, \
                 "graphed Adam fell back to eager on the GPU"
     # identical math modulo device-vs-host powf for the bias corrections
     assert torch.allclose(runs[True], runs[False], atol=1e-4, rtol=1e-4)
