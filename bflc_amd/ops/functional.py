@@ -239,7 +239,8 @@ class _MaxPool2dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, kernel: int, stride: int):
         if x.is_cuda:
-            y, idx = hip_ops().maxpool2d_fwd(x, kernel, stride)
+            y, idx = hip_ops().maxpool2d_fwd(x, kernel, stride,
+                                             torch.is_grad_enabled())
         else:
             y, idx = F.max_pool2d(x.permute(0, 3, 1, 2), kernel, stride,
                                   return_indices=True)

@@ -66,7 +66,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     long pad, c10::optional<torch::Tensor> col_cache, bool want_db);
 std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
                                                        long kernel,
-                                                       long stride);
+                                                       long stride,
+                                                       bool want_idx);
 torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx,
                             std::vector<long> in_shape, long kernel,
                             long stride);
@@ -137,7 +138,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w"), py::arg("dy"), py::arg("stride"),
         py::arg("pad"), py::arg("col_cache") = py::none(),
         py::arg("want_db") = true);
-  m.def("maxpool2d_fwd", &bflc::maxpool2d_fwd);
+  m.def("maxpool2d_fwd", &bflc::maxpool2d_fwd, py::arg("x"),
+        py::arg("kernel"), py::arg("stride"),
+        py::arg("want_idx") = true);
   m.def("maxpool2d_bwd", &bflc::maxpool2d_bwd);
   m.def("batchnorm_fwd", &bflc::batchnorm_fwd,
         py::arg("x"), py::arg("gamma"), py::arg("beta"), py::arg("eps"),

@@ -272,7 +272,8 @@ __global__ void maxpool_nhwc_vec_fwd(const bf16* __restrict__ x,
       oidx[j] = (uint8_t)bi[j];
     }
     *reinterpret_cast<bf16x8_t*>(&y[m * sh.C + c8]) = out;
-    *reinterpret_cast<u8x8_t*>(&idx[m * sh.C + c8]) = oidx;
+    if (idx)  // grad-free forwards skip the mask write entirely
+      *reinterpret_cast<u8x8_t*>(&idx[m * sh.C + c8]) = oidx;
   }
 }
 
@@ -629,7 +630,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
 
 std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
                                                        long kernel,
-                                                       long stride) {
+                                                       long stride,
+                                                       bool want_idx) {
   CHECK_GPU(x); CHECK_CONTIG(x);
   ConvShape sh;
   sh.N = (int)x.size(0); sh.H = (int)x.size(1);
@@ -641,13 +643,20 @@ std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
   auto y = torch::empty({(long)sh.N, (long)sh.OH, (long)sh.OW, (long)sh.C},
                         x.options());
   TORCH_CHECK(sh.R * sh.S <= 255, "maxpool window too large for u8 idx");
-  auto idx = torch::empty_like(y, y.options().dtype(at::kByte));
+  // grad-free forwards (committee scoring) never read the argmax mask:
+  // skip both its allocation and its write (vec path only; the C%8!=0
+  // scalar path is not on any model's hot shape)
+  const bool skip_idx = !want_idx && sh.C % 8 == 0;
+  auto idx = skip_idx
+                 ? torch::empty({0}, y.options().dtype(at::kByte))
+                 : torch::empty_like(y, y.options().dtype(at::kByte));
   if (sh.C % 8 == 0) {
     const long total_g = sh.M() * (sh.C / 8);
     hipLaunchKernelGGL(maxpool_nhwc_vec_fwd, dim3(ew_grid(total_g)),
                        dim3(256), 0, cur_stream(),
                        (const bf16*)x.data_ptr(), (bf16*)y.data_ptr(),
-                       idx.data_ptr<uint8_t>(), sh, total_g);
+                       skip_idx ? nullptr : idx.data_ptr<uint8_t>(), sh,
+                       total_g);
   } else {
     const long total = y.numel();
     hipLaunchKernelGGL(maxpool_nhwc_fwd, dim3(ew_grid(total)), dim3(256), 0,
