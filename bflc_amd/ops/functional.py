@@ -162,16 +162,16 @@ class _Conv2dFn(torch.autograd.Function):
     permute to NCHW, torch fp32 conv, permute back."""
 
     @staticmethod
-    def forward(ctx, x, w, b, stride: int, padding: int, relu: bool):
+    def forward(ctx, x, w, b, stride: int, padding: int, relu: bool,
+                want_col: bool):
+        # NOTE: want_col is computed by the conv2d() wrapper OUTSIDE
+        # this method — autograd.Function.forward always runs with grad
+        # mode disabled internally, so torch.is_grad_enabled() in here
+        # is unconditionally False
         ctx.stride, ctx.padding = stride, padding
         ctx.has_bias = b is not None
         ctx.relu = relu
         if x.is_cuda:
-            # grad-free forwards (committee scoring / sponsor eval) skip
-            # the col materialization entirely — thin shapes gather the
-            # window inside the GEMM (gemm_thin_conv_kernel), identical
-            # output, no 2x O(M*RSC) col traffic
-            want_col = torch.is_grad_enabled()
             y, col = hip_ops().conv2d_fwd_col(
                 x, w, b if b is not None else
                 torch.zeros(w.shape[0], device=x.device, dtype=x.dtype),
@@ -227,20 +227,29 @@ class _Conv2dFn(torch.autograd.Function):
             dx = dx.permute(0, 2, 3, 1).contiguous()
             dw = dw.permute(0, 2, 3, 1).contiguous()
             db = dy.sum(dim=(0, 1, 2))
-        return dx, dw, (db if ctx.has_bias else None), None, None, None
+        return dx, dw, (db if ctx.has_bias else None), None, None, None, \
+            None
 
 
 def conv2d(x, w, b=None, stride: int = 1, padding: int = 0,
            relu: bool = False) -> torch.Tensor:
-    return _Conv2dFn.apply(x, w, b, stride, padding, relu)
+    # grad-free forwards (committee scoring / sponsor eval) skip the
+    # col materialization entirely — thin shapes gather the window
+    # inside the GEMM (gemm_thin_conv_kernel), identical output, no
+    # 2x O(M*RSC) col traffic. Evaluated HERE (outside the Function:
+    # forward() always executes with grad mode off).
+    want_col = torch.is_grad_enabled() and \
+        (x.requires_grad or w.requires_grad)
+    return _Conv2dFn.apply(x, w, b, stride, padding, relu, want_col)
 
 
 class _MaxPool2dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, kernel: int, stride: int):
+    def forward(ctx, x, kernel: int, stride: int, want_idx: bool):
+        # want_idx comes from the maxpool2d() wrapper (grad mode is
+        # always off inside Function.forward)
         if x.is_cuda:
-            y, idx = hip_ops().maxpool2d_fwd(x, kernel, stride,
-                                             torch.is_grad_enabled())
+            y, idx = hip_ops().maxpool2d_fwd(x, kernel, stride, want_idx)
         else:
             y, idx = F.max_pool2d(x.permute(0, 3, 1, 2), kernel, stride,
                                   return_indices=True)
@@ -262,11 +271,12 @@ class _MaxPool2dFn(torch.autograd.Function):
                                 ctx.stride,
                                 output_size=ctx.in_shape[1:3])
             dx = dx.permute(0, 2, 3, 1).contiguous()
-        return dx, None, None
+        return dx, None, None, None
 
 
 def maxpool2d(x, kernel: int = 2, stride: Optional[int] = None) -> torch.Tensor:
-    return _MaxPool2dFn.apply(x, kernel, stride or kernel)
+    want_idx = torch.is_grad_enabled() and x.requires_grad
+    return _MaxPool2dFn.apply(x, kernel, stride or kernel, want_idx)
 
 
 class _BatchNormFn(torch.autograd.Function):

@@ -672,6 +672,9 @@ torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx,
                             std::vector<long> in_shape, long kernel,
                             long stride) {
   CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(idx);
+  TORCH_CHECK(idx.numel() == dy.numel(),
+              "maxpool2d_bwd: argmax mask missing/size-mismatched (the "
+              "forward ran with want_idx=false?)");
   ConvShape sh;
   sh.N = (int)in_shape[0]; sh.H = (int)in_shape[1];
   sh.W = (int)in_shape[2]; sh.C = (int)in_shape[3];
