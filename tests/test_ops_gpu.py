@@ -340,3 +340,37 @@ class TestEndToEndGPU:
         acc = eng.evaluate_global()
         assert acc > 0.85
         assert stats[-1].global_loss < stats[0].global_loss
+
+
+class TestEvalFastPaths:
+    def test_maxpool_eval_skips_mask_and_bwd_raises_loudly(self):
+        """want_idx=False must produce identical pooled values with an
+        empty mask, and feeding that mask to the backward must raise a
+        diagnosable error instead of a device fault (regression: grad
+        mode is always off inside Function.forward, so the wrapper
+        computes want_idx — a bug here once faulted at dispatch)."""
+        import pytest as _pytest
+        x = torch.randn(64, 28, 28, 32, device=DEV).bfloat16()
+        y1, idx1 = hip().maxpool2d_fwd(x, 2, 2, True)
+        y2, idx2 = hip().maxpool2d_fwd(x, 2, 2, False)
+        assert torch.equal(y1, y2)
+        assert idx1.numel() == y1.numel() and idx2.numel() == 0
+        dy = torch.randn_like(y1)
+        with _pytest.raises(RuntimeError, match="mask"):
+            hip().maxpool2d_bwd(dy, idx2, list(x.shape), 2, 2)
+
+    def test_train_conv_backward_works_after_eval_forwards(self):
+        """The exact crash scenario: grad-free forwards (scoring) then
+        a training forward+backward through the same model — the train
+        pass must save real masks/cols even though eval passes do not."""
+        from bflc_amd.config import FLConfig
+        from bflc_amd.models import build_model
+        cfg = FLConfig.for_world(1, model="femnist_cnn", n_class=62)
+        m = build_model(cfg, DEV)
+        xe = torch.randn(256, 28, 28, 1, device=DEV)
+        ye = torch.randint(0, 62, (256,), device=DEV)
+        m.accuracy_t(xe, ye)          # eval (no_grad): no masks saved
+        loss = m.loss(xe, ye)         # train forward
+        loss.backward()               # must not fault / raise
+        torch.cuda.synchronize()
+        assert m.grad_flat().abs().sum() > 0
