@@ -1158,9 +1158,24 @@ __global__ void splitk_reduce_kernel(const float* __restrict__ Cpart, int S,
   const long stride = (long)gridDim.x * 32;
   __shared__ float red[8][33];
   for (; i - elane < total; i += stride) {
-    float acc = 0.f;
-    if (i < total)
-      for (int s = slane; s < S; s += 8) acc += Cpart[(long)s * total + i];
+    // 4 independent accumulators per s-lane: the serial s-walk left
+    // one outstanding load per iteration on the small-output reduces
+    // (dw[16][144], S~128: 10.8 us/call, 6% of a ResNet-20 round);
+    // fixed combine order keeps the reduction deterministic.
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    if (i < total) {
+      int s = slane;
+      for (; s + 24 < S; s += 32) {
+        a0 += Cpart[(long)s * total + i];
+        a1 += Cpart[(long)(s + 8) * total + i];
+        a2 += Cpart[(long)(s + 16) * total + i];
+        a3 += Cpart[(long)(s + 24) * total + i];
+      }
+      if (s < S) { a0 += Cpart[(long)s * total + i]; s += 8; }
+      if (s < S) { a1 += Cpart[(long)s * total + i]; s += 8; }
+      if (s < S) { a2 += Cpart[(long)s * total + i]; }
+    }
+    const float acc = (a0 + a1) + (a2 + a3);
     red[slane][elane] = acc;
     __syncthreads();
     if (slane == 0 && i < total) {
