@@ -238,7 +238,11 @@ def conv2d(x, w, b=None, stride: int = 1, padding: int = 0,
     # inside the GEMM (gemm_thin_conv_kernel), identical output, no
     # 2x O(M*RSC) col traffic. Evaluated HERE (outside the Function:
     # forward() always executes with grad mode off).
-    want_col = torch.is_grad_enabled() and \
+    # BFLC_CONV_NO_COL=1: A/B knob — skip the col cache in TRAINING too
+    # (backward then uses the implicit wgrad route).
+    import os
+    no_col = os.environ.get("BFLC_CONV_NO_COL", "0") == "1"
+    want_col = (not no_col) and torch.is_grad_enabled() and \
         (x.requires_grad or w.requires_grad)
     return _Conv2dFn.apply(x, w, b, stride, padding, relu, want_col)
 
