@@ -238,10 +238,14 @@ def conv2d(x, w, b=None, stride: int = 1, padding: int = 0,
     # inside the GEMM (gemm_thin_conv_kernel), identical output, no
     # 2x O(M*RSC) col traffic. Evaluated HERE (outside the Function:
     # forward() always executes with grad mode off).
-    # BFLC_CONV_NO_COL=1: A/B knob — skip the col cache in TRAINING too
-    # (backward then uses the implicit wgrad route).
+    # Training also skips the col cache by default since the round-2
+    # A/B: the backward's implicit wgrad route (Kout <= 64) beat the
+    # materialize-in-forward-and-reuse design on MI355X (FEMNIST c1
+    # 2.40 -> 2.36 ms/round) — 288 GB of HBM made caching free, but
+    #8 TB/s makes NOT writing it faster. BFLC_CONV_NO_COL=0 restores
+    # the cache for A/B.
     import os
-    no_col = os.environ.get("BFLC_CONV_NO_COL", "0") == "1"
+    no_col = os.environ.get("BFLC_CONV_NO_COL", "1") == "1"
     want_col = (not no_col) and torch.is_grad_enabled() and \
         (x.requires_grad or w.requires_grad)
     return _Conv2dFn.apply(x, w, b, stride, padding, relu, want_col)
