@@ -156,3 +156,22 @@ def test_byzantine_defense_on_gpu():
     assert set(control_sel) & attackers  # control really is attacked
     assert defense_acc > 0.9
     assert defense_acc > control_acc + 0.04, (defense_acc, control_acc)
+
+
+def test_whole_phase_graphs_active_in_protocol_round():
+    """The committee protocol round must actually run on the whole-phase
+    graphs (one replay per client training pass, one per scorer) — not
+    silently fall back to eager/per-batch paths."""
+    from bflc_amd.config import FLConfig
+    cfg = FLConfig.for_world(8, model="femnist_cnn", n_class=62,
+                             samples_per_client=256, batch_size=128,
+                             eval_samples=256, partition="dirichlet")
+    eng = _engine(cfg)
+    eng.run(2)
+    assert any(g is not None for g in eng._train_graphs.values()), \
+        "no whole-train graph captured"
+    assert any(g is not None for g in eng._score_graphs.values()), \
+        "no whole-scoring graph captured"
+    # the packed candidate stack exists and matches the quota
+    assert eng._cand_stack is not None
+    assert eng._cand_stack.shape[0] == cfg.needed_update_count
