@@ -138,7 +138,11 @@ def main():
         shards, test = load_occupancy(clients=cfg.client_num)
     else:
         shards, test = make_federated(cfg)
-    chain = LocalChain(cfg, log_path="bflc_chain_log.jsonl")
+    # per-client HMAC credentials (reference get_batch_accounts.sh):
+    # the chain verifies every update/score transaction's tag
+    from bflc_amd.chain.identity import KeyTable
+    keys = KeyTable([f"node_{i}" for i in range(cfg.client_num)], cfg.seed)
+    chain = LocalChain(cfg, log_path="bflc_chain_log.jsonl", keys=keys)
 
     stop = threading.Event()
     threads = []
