@@ -135,6 +135,21 @@ class FLEngine:
         self._score_graphs: Dict[int, object] = {}
         self._cand_stack: Optional[torch.Tensor] = None
         self._graph_pool = None  # shared across all captures (one pool)
+        # concurrent client streams: the FL-model kernels are
+        # issue/launch-bound, not HBM-bound (profiles/r02_pmc_femnist.md)
+        # — so a rank hosting several clients overlaps their whole-phase
+        # graph replays on one HIP stream PER CLIENT. That requires one
+        # model replica and one private graph pool per local client
+        # (concurrently-replaying graphs must not share weight buffers
+        # or pool memory); the cost is #local_clients x (model +
+        # activation footprint), sized for 288 GB HBM3E (ResNet-50 at 8
+        # clients/rank ~ tens of GB). Per-client compute is unchanged
+        # and each client's kernels stay serialized on its own stream,
+        # so results are identical to the sequential path.
+        self._client_models: Dict[int, object] = {}
+        self._client_streams: Dict[int, object] = {}
+        self._client_pools: Dict[int, object] = {}
+        self._client_delta: Dict[int, torch.Tensor] = {}
         # round-persistent scratch (avoid per-round alloc/fill of
         # O(model)-sized tensors: the gather stack is ~400 MB/round on
         # ResNet-50)
@@ -143,6 +158,8 @@ class FLEngine:
         self._fedavg_buf: Optional[torch.Tensor] = None
         self._use_graphs = (cfg.use_graphs and self.device.type == "cuda"
                             and os.environ.get("BFLC_GRAPHS", "1") != "0")
+        self._concurrent = (self._use_graphs
+                            and os.environ.get("BFLC_STREAMS", "1") != "0")
 
     # ------------------------------------------------------------------
     def _planned_submitters(self) -> List[int]:
@@ -185,21 +202,54 @@ class FLEngine:
         return self._steppers[key]
 
     # ------------------------------------------------------------------
+    def _client_model(self, client: int):
+        """Per-client model replica (concurrent mode): each local
+        client's graphs capture against their OWN flat/cflat buffers so
+        their replays can overlap on per-client streams."""
+        if client not in self._client_models:
+            self._client_models[client] = build_model(self.cfg, self.device)
+        return self._client_models[client]
+
+    def _client_stream(self, client: int):
+        if client not in self._client_streams:
+            self._client_streams[client] = torch.cuda.Stream()
+        return self._client_streams[client]
+
+    def _capture_pool(self, client: int):
+        """Graph memory pool for this client's captures. Concurrent
+        mode: one PRIVATE pool per client (graphs replaying in parallel
+        must not share pool memory); sequential mode: the one shared
+        pool (round-1 ADVICE — serialized replays can share)."""
+        if self._concurrent:
+            return self._client_pools.get(client)
+        return self._graph_pool
+
+    def _note_pool(self, client: int, g) -> None:
+        if self._concurrent:
+            if self._client_pools.get(client) is None:
+                self._client_pools[client] = g.pool()
+        elif self._graph_pool is None:
+            self._graph_pool = g.pool()
+
+    # ------------------------------------------------------------------
     def _whole_train_graph(self, client: int):
         """Per-client whole-local-training graph (shard slices baked in,
-        zero per-batch copies); None if capture is unavailable."""
+        zero per-batch copies); None if capture is unavailable. In
+        concurrent mode the graph is captured against the client's own
+        model replica (g.model) so replays can overlap."""
         if client in self._train_graphs:
             return self._train_graphs[client]
         try:
             from bflc_amd.fl.graphs import GraphedLocalTrain
             shard = self.shards[client]
-            g = GraphedLocalTrain(self.model, self.cfg.learning_rate,
+            model = (self._client_model(client) if self._concurrent
+                     else self.model)
+            g = GraphedLocalTrain(model, self.cfg.learning_rate,
                                   shard.x, shard.y, self.cfg.batch_size,
                                   self.cfg.local_epochs,
                                   optimizer=self.cfg.optimizer,
-                                  pool=self._graph_pool)
-            if self._graph_pool is None:
-                self._graph_pool = g.pool()
+                                  pool=self._capture_pool(client))
+            self._note_pool(client, g)
         except Exception as e:
             warnings.warn(f"whole-train graph capture failed for client "
                           f"{client}, using per-batch path: {e}")
@@ -218,12 +268,12 @@ class FLEngine:
         # (capture executes real steps and mutates flat/cflat)
         wg = self._whole_train_graph(client) if self._use_graphs else None
         if wg is not None:
-            self.model.set_flat(self.global_flat)
+            wg.model.set_flat(self.global_flat)
             cost = wg.run()
             avg_cost = float(cost) / wg.n_steps
-            delta = self.global_flat.clone()
-            O.axpy_(delta, -1.0, self.model.flat.data)
-            delta.div_(cfg.learning_rate)
+            delta = torch.empty_like(self.global_flat)
+            O.delta_extract_(delta, self.global_flat, wg.model.flat.data,
+                             cfg.learning_rate)
             return delta, n, avg_cost
         bs = min(cfg.batch_size, n)
         total_batches = max(n // bs, 1)
@@ -263,9 +313,9 @@ class FLEngine:
                     cost_accum += loss.detach()
         avg_cost = float(cost_accum) / (total_batches * cfg.local_epochs)
         # delta = (W0 - W)/lr  (reference main.py:153-154)
-        delta = self.global_flat.clone()
-        O.axpy_(delta, -1.0, self.model.flat.data)
-        delta.div_(cfg.learning_rate)
+        delta = torch.empty_like(self.global_flat)
+        O.delta_extract_(delta, self.global_flat, self.model.flat.data,
+                         cfg.learning_rate)
         return delta, n, avg_cost
 
     # ------------------------------------------------------------------
@@ -293,12 +343,13 @@ class FLEngine:
         try:
             from bflc_amd.fl.graphs import GraphedScorePhase
             shard = self.shards[scorer]
-            g = GraphedScorePhase(self.model, shard.x, shard.y,
+            model = (self._client_model(scorer) if self._concurrent
+                     else self.model)
+            g = GraphedScorePhase(model, shard.x, shard.y,
                                   self._cand_stack, self.global_flat,
                                   self.cfg.learning_rate,
-                                  pool=self._graph_pool)
-            if self._graph_pool is None:
-                self._graph_pool = g.pool()
+                                  pool=self._capture_pool(scorer))
+            self._note_pool(scorer, g)
         except Exception as e:
             warnings.warn(f"whole-scoring graph capture failed for "
                           f"scorer {scorer}, per-candidate path: {e}")
@@ -399,14 +450,54 @@ class FLEngine:
         t0 = time.perf_counter()
 
         # ---- phase T: local training on designated submitters ----------
+        # Concurrent path: each local trainer's whole-training graph
+        # replays on its own HIP stream (its own model replica and
+        # private pool) — set_flat, the replay, and the fused delta
+        # extraction are all enqueued per stream, joined once at the
+        # phase end. Falls back to the sequential loop if any capture is
+        # unavailable.
         submitters = self._planned_submitters()
         local_subs = [i for i in submitters if i in self.shards]
         local_updates: List[Tuple[str, torch.Tensor, int, float]] = []
         samples_trained = 0
-        for i in local_subs:
-            delta, n, cost = self._local_train(i)
-            local_updates.append((self.origins[i], delta, n, cost))
-            samples_trained += n * cfg.local_epochs
+        conc = self._concurrent and len(local_subs) > 1
+        tgraphs = {}
+        if conc:
+            for i in local_subs:
+                g = self._whole_train_graph(i)
+                if g is None:
+                    conc = False
+                    break
+                tgraphs[i] = g
+        if conc:
+            for i in local_subs:  # allocate on the main stream
+                if i not in self._client_delta:
+                    self._client_delta[i] = torch.empty_like(
+                        self.global_flat)
+            cur = torch.cuda.current_stream()
+            costs = {}
+            for i in local_subs:
+                g, s = tgraphs[i], self._client_stream(i)
+                s.wait_stream(cur)
+                with torch.cuda.stream(s):
+                    g.model.set_flat(self.global_flat)
+                    costs[i] = g.run()
+                    O.delta_extract_(self._client_delta[i],
+                                     self.global_flat, g.model.flat.data,
+                                     cfg.learning_rate)
+            for i in local_subs:
+                cur.wait_stream(self._client_streams[i])
+            for i in local_subs:
+                n = self.shards[i].n
+                avg_cost = float(costs[i]) / tgraphs[i].n_steps
+                local_updates.append((self.origins[i],
+                                      self._client_delta[i], n, avg_cost))
+                samples_trained += n * cfg.local_epochs
+        else:
+            for i in local_subs:
+                delta, n, cost = self._local_train(i)
+                local_updates.append((self.origins[i], delta, n, cost))
+                samples_trained += n * cfg.local_epochs
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
         t1 = time.perf_counter()
@@ -456,11 +547,41 @@ class FLEngine:
                              if roles.get(self.origins[i]) == "comm"]
         packed = bool(local_scorers) and self._pack_candidates(updates)
         my_scores = []
-        for i in local_scorers:
-            o = self.origins[i]
-            smap = self._score_candidates(i, updates, packed=packed)
-            my_scores.append((o, smap, self.keys.sign(
-                "scores", o, epoch, scores_payload(smap))))
+        # Concurrent path: every local scorer's whole-scoring graph (all
+        # K candidates in one replay) runs on that client's stream;
+        # scores are read back once after the join.
+        conc_s = self._concurrent and packed and len(local_scorers) > 1
+        sgraphs = {}
+        if conc_s:
+            for i in local_scorers:
+                g = self._phase_scorer(i)
+                if g is None:
+                    conc_s = False
+                    break
+                sgraphs[i] = g
+        if conc_s:
+            cur = torch.cuda.current_stream()
+            res = {}
+            for i in local_scorers:
+                s = self._client_stream(i)
+                s.wait_stream(cur)
+                with torch.cuda.stream(s):
+                    res[i] = sgraphs[i].run_inplace()
+            for i in local_scorers:
+                cur.wait_stream(self._client_streams[i])
+            for i in local_scorers:
+                accs = res[i].tolist()
+                smap = {origin: accs[k]
+                        for k, (origin, _) in enumerate(updates)}
+                o = self.origins[i]
+                my_scores.append((o, smap, self.keys.sign(
+                    "scores", o, epoch, scores_payload(smap))))
+        else:
+            for i in local_scorers:
+                o = self.origins[i]
+                smap = self._score_candidates(i, updates, packed=packed)
+                my_scores.append((o, smap, self.keys.sign(
+                    "scores", o, epoch, scores_payload(smap))))
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
         t3 = time.perf_counter()

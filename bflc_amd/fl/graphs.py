@@ -226,24 +226,25 @@ class GraphedScorePhase:
         from bflc_amd.ops import functional as O
         self.model = model
         self.K = cand_stack.shape[0]
-        self.cand = torch.empty_like(model.flat)
         self.accs = torch.zeros(self.K, device=model.device,
                                 dtype=torch.float32)
         lr = float(lr)
 
+        # candidate load is ONE fused kernel writing the compute-dtype
+        # shadow directly (the eval forward's only weight consumer);
+        # the fp32 master stays stale during scoring — every later
+        # phase reloads it via set_flat (ops/functional.py score_load_)
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
-            self.cand.copy_(global_flat)
-            model.set_flat(self.cand)
+            O.score_load_(model.cflat.data, global_flat, cand_stack[0], lr)
             model.accuracy_t(shard_x, shard_y)
         torch.cuda.current_stream().wait_stream(side)
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph, pool=pool):
             for k in range(self.K):
-                self.cand.copy_(global_flat)
-                O.axpy_(self.cand, -lr, cand_stack[k])
-                model.set_flat(self.cand)
+                O.score_load_(model.cflat.data, global_flat,
+                              cand_stack[k], lr)
                 self.accs[k] = model.accuracy_t(shard_x, shard_y)
 
     def pool(self):
@@ -253,6 +254,13 @@ class GraphedScorePhase:
         """Replay; returns a snapshot of the K accuracies (device)."""
         self.graph.replay()
         return self.accs.clone()
+
+    def run_inplace(self) -> torch.Tensor:
+        """Replay; returns the live accs buffer (no allocation) — for
+        concurrent per-scorer stream replays, where the caller reads
+        the buffer only after joining all scorer streams."""
+        self.graph.replay()
+        return self.accs
 
 
 class GraphedScore:

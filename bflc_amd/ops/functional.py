@@ -489,6 +489,33 @@ def axpy_(y: torch.Tensor, alpha: float, x: torch.Tensor) -> torch.Tensor:
     return y.add_(x, alpha=alpha)
 
 
+def score_load_(shadow: torch.Tensor, global_flat: torch.Tensor,
+                delta: torch.Tensor, lr: float) -> None:
+    """shadow = compute_dtype(global - lr*delta) in ONE pass: the
+    committee-scoring candidate load (reference candidate
+    reconstruction, main.py:215-216). The scorer's forward reads only
+    the compute-dtype shadow, so the fp32 master copy of set_flat is
+    skipped entirely; fp32 math with one rounding — bitwise identical
+    to the copy+axpy+set_flat chain it replaces."""
+    if shadow.is_cuda:
+        hip_ops().score_load_(shadow, global_flat, delta, float(lr))
+    else:
+        with torch.no_grad():
+            shadow.copy_((global_flat - lr * delta).to(shadow.dtype))
+
+
+def delta_extract_(out: torch.Tensor, global_flat: torch.Tensor,
+                   w: torch.Tensor, lr: float) -> None:
+    """out = (global - w)/lr in ONE pass: pseudo-gradient extraction
+    (reference main.py:153-154), replacing clone + axpy + scalar-div."""
+    if out.is_cuda:
+        hip_ops().delta_extract_(out, global_flat, w, float(lr))
+    else:
+        with torch.no_grad():
+            torch.sub(global_flat, w, out=out)
+            out.div_(lr)
+
+
 def sgd_step_(flat_param: torch.Tensor, flat_grad: torch.Tensor,
               lr: float) -> None:
     """Fused flat SGD update (reference main.py:127-130)."""

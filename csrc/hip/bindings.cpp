@@ -23,6 +23,10 @@ void adam_master_graph_(torch::Tensor p, torch::Tensor shadow,
                         torch::Tensor step, torch::Tensor bc, double lr,
                         double beta1, double beta2, double eps);
 void refresh_shadow_(torch::Tensor p, torch::Tensor shadow);
+void score_load_(torch::Tensor shadow, torch::Tensor global_flat,
+                 torch::Tensor delta, double lr);
+void delta_extract_(torch::Tensor out, torch::Tensor global_flat,
+                    torch::Tensor w, double lr);
 torch::Tensor relu_fwd(torch::Tensor x);
 torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy);
 std::tuple<torch::Tensor, torch::Tensor> relu_bwd_colsum(
@@ -101,6 +105,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hipGraph-capturable Adam: device step counter + on-device bias "
         "corrections (tick kernel), fp32 master + bf16 shadow");
   m.def("refresh_shadow_", &bflc::refresh_shadow_, "shadow = bf16(master)");
+  m.def("score_load_", &bflc::score_load_,
+        "shadow = compute_dtype(global - lr*delta): one-pass scoring "
+        "candidate load (replaces copy+axpy+set_flat)");
+  m.def("delta_extract_", &bflc::delta_extract_,
+        "out = (global - w)/lr: one-pass pseudo-gradient extraction");
   m.def("relu_fwd", &bflc::relu_fwd);
   m.def("relu_bwd", &bflc::relu_bwd);
   m.def("relu_bwd_colsum", &bflc::relu_bwd_colsum,

@@ -191,6 +191,50 @@ __global__ void cast_f32_bf16_kernel(const float* __restrict__ p,
   for (long k = i; k < n; k += stride) shadow[k] = f2b(p[k]);
 }
 
+// Committee-scoring candidate load in ONE pass (reference candidate
+// reconstruction W0 - lr*dW, main.py:215-216): the scorer's forward
+// reads only the bf16 compute shadow, so write bf16(global - lr*delta)
+// straight into it. fp32 math with a single bf16 round — bitwise
+// identical to the copy + axpy + master-copy + shadow-cast chain it
+// replaces (that chain rounded the same fp32 value once too), at 10P
+// bytes / 1 launch instead of 34P / 4. The scoring phase runs
+// committee x quota of these per round and its kernels are
+// launch/issue-bound at FL-model sizes (profiles/r02_pmc_femnist.md).
+__global__ void score_load_bf16_kernel(bf16* __restrict__ shadow,
+                                       const float* __restrict__ gl,
+                                       const float* __restrict__ d,
+                                       float lr, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride)
+    shadow[k] = f2b(fmaf(-lr, d[k], gl[k]));
+}
+
+// fp32-compute variant (cflat IS flat): flat = global - lr*delta.
+__global__ void score_load_f32_kernel(float* __restrict__ shadow,
+                                      const float* __restrict__ gl,
+                                      const float* __restrict__ d,
+                                      float lr, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride)
+    shadow[k] = fmaf(-lr, d[k], gl[k]);
+}
+
+// Pseudo-gradient extraction in ONE pass (reference main.py:153-154):
+// delta = (W0 - W) / lr. Replaces clone + axpy + scalar-div (3 kernels,
+// 28P bytes) with one kernel and 12P. fmaf(-1,w,g) and the fp32 IEEE
+// divide match the old chain bitwise.
+__global__ void delta_extract_kernel(float* __restrict__ out,
+                                     const float* __restrict__ gl,
+                                     const float* __restrict__ w,
+                                     float lr, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long k = i; k < n; k += stride)
+    out[k] = (gl[k] - w[k]) / lr;
+}
+
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
 
 // 16-B granule main loop + scalar tail (scalar 2-B accesses measured
@@ -355,6 +399,49 @@ void adam_master_graph_(torch::Tensor p, torch::Tensor shadow,
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      bc.data_ptr<float>(), (float)lr, (float)beta1,
                      (float)beta2, (float)eps, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void score_load_(torch::Tensor shadow, torch::Tensor global_flat,
+                 torch::Tensor delta, double lr) {
+  CHECK_GPU(shadow); CHECK_CONTIG(shadow);
+  CHECK_GPU(global_flat); CHECK_CONTIG(global_flat);
+  CHECK_GPU(delta); CHECK_CONTIG(delta);
+  TORCH_CHECK(global_flat.scalar_type() == at::kFloat &&
+              delta.scalar_type() == at::kFloat);
+  long n = shadow.numel();
+  TORCH_CHECK(global_flat.numel() == n && delta.numel() == n);
+  if (shadow.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(score_load_bf16_kernel, dim3(grid_for(n)),
+                       dim3(kBlock), 0, cur_stream(),
+                       (bf16*)shadow.data_ptr(),
+                       global_flat.data_ptr<float>(),
+                       delta.data_ptr<float>(), (float)lr, n);
+  } else {
+    TORCH_CHECK(shadow.scalar_type() == at::kFloat);
+    hipLaunchKernelGGL(score_load_f32_kernel, dim3(grid_for(n)),
+                       dim3(kBlock), 0, cur_stream(),
+                       shadow.data_ptr<float>(),
+                       global_flat.data_ptr<float>(),
+                       delta.data_ptr<float>(), (float)lr, n);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+void delta_extract_(torch::Tensor out, torch::Tensor global_flat,
+                    torch::Tensor w, double lr) {
+  CHECK_GPU(out); CHECK_CONTIG(out);
+  CHECK_GPU(global_flat); CHECK_CONTIG(global_flat);
+  CHECK_GPU(w); CHECK_CONTIG(w);
+  TORCH_CHECK(out.scalar_type() == at::kFloat &&
+              global_flat.scalar_type() == at::kFloat &&
+              w.scalar_type() == at::kFloat);
+  long n = out.numel();
+  TORCH_CHECK(global_flat.numel() == n && w.numel() == n);
+  hipLaunchKernelGGL(delta_extract_kernel, dim3(grid_for(n)), dim3(kBlock),
+                     0, cur_stream(), out.data_ptr<float>(),
+                     global_flat.data_ptr<float>(), w.data_ptr<float>(),
+                     (float)lr, n);
   HIP_CHECK(hipGetLastError());
 }
 

@@ -158,6 +158,35 @@ def test_byzantine_defense_on_gpu():
     assert defense_acc > control_acc + 0.04, (defense_acc, control_acc)
 
 
+def test_concurrent_streams_match_sequential(monkeypatch):
+    """The per-client stream overlap (one model replica + one HIP
+    stream per local client, concurrent whole-phase graph replays) must
+    be a pure scheduling change: identical global model bits and
+    identical committee decisions vs the sequential single-stream
+    path."""
+    from bflc_amd.config import FLConfig
+
+    def run(streams: str):
+        monkeypatch.setenv("BFLC_STREAMS", streams)
+        cfg = FLConfig.for_world(8, model="femnist_cnn", n_class=62,
+                                 samples_per_client=256, batch_size=128,
+                                 eval_samples=256, partition="dirichlet")
+        eng = _engine(cfg)
+        sel = []
+        for _ in range(6):
+            eng.run_round()
+            sel.append(tuple(o for o, _ in eng.last_decision.selected))
+        return eng.global_flat.clone(), sel, eng
+
+    flat_c, sel_c, eng_c = run("1")
+    # the concurrent path really engaged: replicas + streams exist
+    assert len(eng_c._client_streams) > 1, "no per-client streams made"
+    assert len(eng_c._client_models) > 1, "no per-client replicas made"
+    flat_s, sel_s, _ = run("0")
+    assert sel_c == sel_s
+    assert torch.equal(flat_c, flat_s)
+
+
 def test_whole_phase_graphs_active_in_protocol_round():
     """The committee protocol round must actually run on the whole-phase
     graphs (one replay per client training pass, one per scorer) — not

@@ -88,6 +88,32 @@ def test_sgd_and_axpy():
     assert torch.allclose(y, torch.full((10,), 4.0))
 
 
+def test_score_load_and_delta_extract():
+    """CPU fallbacks of the fused one-pass ops (the GPU kernels are
+    oracle-tested in test_ops_gpu.py): score_load_ writes
+    compute_dtype(global - lr*delta); delta_extract_ writes
+    (global - w)/lr."""
+    torch.manual_seed(3)
+    g = torch.randn(1000)
+    d = torch.randn(1000)
+    lr = 0.01
+    out32 = torch.empty(1000)
+    O.score_load_(out32, g, d, lr)
+    assert torch.allclose(out32, g - lr * d, atol=1e-7)
+    out16 = torch.empty(1000, dtype=torch.bfloat16)
+    O.score_load_(out16, g, d, lr)
+    assert torch.equal(out16, (g - lr * d).bfloat16())
+
+    w = torch.randn(1000)
+    delta = torch.empty(1000)
+    O.delta_extract_(delta, g, w, lr)
+    # bitwise match with the clone+axpy+div chain it replaces
+    ref = g.clone()
+    O.axpy_(ref, -1.0, w)
+    ref.div_(lr)
+    assert torch.equal(delta, ref)
+
+
 def test_adam_matches_torch_adam():
     torch.manual_seed(1)
     p = torch.randn(64)

@@ -219,6 +219,41 @@ class TestFlatOps:
             hip().adam_step_(pg, gg, mg, vg, step, 0.01, 0.9, 0.999, 1e-8)
         assert torch.allclose(pg.cpu(), p, atol=1e-5)
 
+    def test_score_load_bitwise_vs_old_chain(self):
+        """score_load_ (one pass: shadow = bf16(global - lr*delta))
+        must be BITWISE equal to the copy + axpy + cast chain it
+        replaced — both round the same fmaf(-lr, d, g) fp32 value
+        once."""
+        torch.manual_seed(21)
+        g = torch.randn(100003, device=DEV)
+        d = torch.randn(100003, device=DEV)
+        lr = 0.003
+        shadow = torch.empty(100003, device=DEV, dtype=torch.bfloat16)
+        hip().score_load_(shadow, g, d, lr)
+        cand = g.clone()
+        hip().axpy_(cand, d, -lr)
+        ref = torch.empty_like(shadow)
+        hip().refresh_shadow_(cand, ref)
+        assert torch.equal(shadow, ref)
+        # fp32 variant (fp32-compute models: cflat IS flat)
+        s32 = torch.empty(100003, device=DEV)
+        hip().score_load_(s32, g, d, lr)
+        assert torch.equal(s32, cand)
+
+    def test_delta_extract_bitwise_vs_old_chain(self):
+        """delta_extract_ (one pass: (global - w)/lr) is bitwise equal
+        to the clone + axpy(-1) + scalar-div chain it replaced."""
+        torch.manual_seed(22)
+        g = torch.randn(54321, device=DEV)
+        w = torch.randn(54321, device=DEV)
+        lr = 0.01
+        out = torch.empty_like(g)
+        hip().delta_extract_(out, g, w, lr)
+        ref = g.clone()
+        hip().axpy_(ref, w, -1.0)
+        ref.div_(lr)
+        assert torch.equal(out, ref)
+
     def test_thin_conv_implicit_matches_col_path(self):
         """The grad-free conv forward (gemm_thin_conv_kernel: window
         gathered inside the GEMM, no im2col materialization) must be
