@@ -462,6 +462,13 @@ class FLEngine:
         samples_trained = 0
         conc = self._concurrent and len(local_subs) > 1
         tgraphs = {}
+        if self._concurrent and self._round == 0:
+            # Pre-capture EVERY local client's training graph now:
+            # committee rotation means each client eventually trains,
+            # and a lazy capture during a later (timed/steady-state)
+            # round would put one-time capture cost inside it.
+            for i in self.local_clients:
+                self._whole_train_graph(i)
         if conc:
             for i in local_subs:
                 g = self._whole_train_graph(i)
@@ -547,6 +554,12 @@ class FLEngine:
                              if roles.get(self.origins[i]) == "comm"]
         packed = bool(local_scorers) and self._pack_candidates(updates)
         my_scores = []
+        if self._concurrent and packed and self._round == 0:
+            # pre-capture every local client's scoring graph (rotation:
+            # each client eventually sits on the committee) so later
+            # rounds never pay capture cost
+            for i in self.local_clients:
+                self._phase_scorer(i)
         # Concurrent path: every local scorer's whole-scoring graph (all
         # K candidates in one replay) runs on that client's stream;
         # scores are read back once after the join.

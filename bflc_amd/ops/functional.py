@@ -507,7 +507,10 @@ def score_load_(shadow: torch.Tensor, global_flat: torch.Tensor,
 def delta_extract_(out: torch.Tensor, global_flat: torch.Tensor,
                    w: torch.Tensor, lr: float) -> None:
     """out = (global - w)/lr in ONE pass: pseudo-gradient extraction
-    (reference main.py:153-154), replacing clone + axpy + scalar-div."""
+    (reference main.py:153-154), replacing clone + axpy + scalar-div.
+    The quotient may differ from torch's reciprocal-multiply div_ by
+    1 ulp; every rank runs the same kernel, so replicas stay
+    bitwise-identical."""
     if out.is_cuda:
         hip_ops().delta_extract_(out, global_flat, w, float(lr))
     else:

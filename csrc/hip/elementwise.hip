@@ -223,8 +223,10 @@ __global__ void score_load_f32_kernel(float* __restrict__ shadow,
 
 // Pseudo-gradient extraction in ONE pass (reference main.py:153-154):
 // delta = (W0 - W) / lr. Replaces clone + axpy + scalar-div (3 kernels,
-// 28P bytes) with one kernel and 12P. fmaf(-1,w,g) and the fp32 IEEE
-// divide match the old chain bitwise.
+// 28P bytes) with one kernel and 12P. The subtraction matches the old
+// chain bitwise; the IEEE fp32 divide may differ from torch's div_
+// (a reciprocal multiply) by 1 ulp — every rank runs this same kernel,
+// so cross-replica determinism is unaffected.
 __global__ void delta_extract_kernel(float* __restrict__ out,
                                      const float* __restrict__ gl,
                                      const float* __restrict__ w,
