@@ -160,6 +160,7 @@ class FLEngine:
                             and os.environ.get("BFLC_GRAPHS", "1") != "0")
         self._concurrent = (self._use_graphs
                             and os.environ.get("BFLC_STREAMS", "1") != "0")
+        self._phase_debug = os.environ.get("BFLC_PHASE_DEBUG", "0") == "1"
 
     # ------------------------------------------------------------------
     def _planned_submitters(self) -> List[int]:
@@ -483,17 +484,30 @@ class FLEngine:
                         self.global_flat)
             cur = torch.cuda.current_stream()
             costs = {}
+            evs = {}
             for i in local_subs:
                 g, s = tgraphs[i], self._client_stream(i)
                 s.wait_stream(cur)
                 with torch.cuda.stream(s):
+                    if self._phase_debug:
+                        evs[i] = (torch.cuda.Event(enable_timing=True),
+                                  torch.cuda.Event(enable_timing=True))
+                        evs[i][0].record(s)
                     g.model.set_flat(self.global_flat)
                     costs[i] = g.run()
                     O.delta_extract_(self._client_delta[i],
                                      self.global_flat, g.model.flat.data,
                                      cfg.learning_rate)
+                    if self._phase_debug:
+                        evs[i][1].record(s)
             for i in local_subs:
                 cur.wait_stream(self._client_streams[i])
+            if self._phase_debug:
+                torch.cuda.synchronize(self.device)
+                tt = " ".join(f"c{i}:{evs[i][0].elapsed_time(evs[i][1]):.1f}"
+                              for i in local_subs)
+                print(f"# [debug] epoch {epoch} train per-client ms: {tt}",
+                      flush=True)
             for i in local_subs:
                 n = self.shards[i].n
                 avg_cost = float(costs[i]) / tgraphs[i].n_steps
@@ -575,13 +589,26 @@ class FLEngine:
         if conc_s:
             cur = torch.cuda.current_stream()
             res = {}
+            evs = {}
             for i in local_scorers:
                 s = self._client_stream(i)
                 s.wait_stream(cur)
                 with torch.cuda.stream(s):
+                    if self._phase_debug:
+                        evs[i] = (torch.cuda.Event(enable_timing=True),
+                                  torch.cuda.Event(enable_timing=True))
+                        evs[i][0].record(s)
                     res[i] = sgraphs[i].run_inplace()
+                    if self._phase_debug:
+                        evs[i][1].record(s)
             for i in local_scorers:
                 cur.wait_stream(self._client_streams[i])
+            if self._phase_debug:
+                torch.cuda.synchronize(self.device)
+                tt = " ".join(f"c{i}:{evs[i][0].elapsed_time(evs[i][1]):.1f}"
+                              for i in local_scorers)
+                print(f"# [debug] epoch {epoch} score per-client ms: {tt}",
+                      flush=True)
             for i in local_scorers:
                 accs = res[i].tolist()
                 smap = {origin: accs[k]
