@@ -26,6 +26,11 @@ import json
 import os
 import time
 
+# 8 hardware queues (ROCm defaults to 4): the engine replays up to 8
+# clients' whole-phase hipGraphs on 8 concurrent streams per rank;
+# streams sharing a HW queue serialize. Must be set before HIP init.
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
+
 import torch
 
 
