@@ -508,9 +508,11 @@ class FLEngine:
                               for i in local_subs)
                 print(f"# [debug] epoch {epoch} train per-client ms: {tt}",
                       flush=True)
-            for i in local_subs:
+            # ONE host sync for every trainer's accumulated loss
+            costl = torch.stack([costs[i] for i in local_subs]).tolist()
+            for j, i in enumerate(local_subs):
                 n = self.shards[i].n
-                avg_cost = float(costs[i]) / tgraphs[i].n_steps
+                avg_cost = costl[j] / tgraphs[i].n_steps
                 local_updates.append((self.origins[i],
                                       self._client_delta[i], n, avg_cost))
                 samples_trained += n * cfg.local_epochs
@@ -609,8 +611,10 @@ class FLEngine:
                               for i in local_scorers)
                 print(f"# [debug] epoch {epoch} score per-client ms: {tt}",
                       flush=True)
-            for i in local_scorers:
-                accs = res[i].tolist()
+            # ONE host sync for all scorers' [K] accuracy vectors
+            alla = torch.stack([res[i] for i in local_scorers]).tolist()
+            for j, i in enumerate(local_scorers):
+                accs = alla[j]
                 smap = {origin: accs[k]
                         for k, (origin, _) in enumerate(updates)}
                 o = self.origins[i]
