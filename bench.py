@@ -26,10 +26,12 @@ import json
 import os
 import time
 
-# 8 hardware queues (ROCm defaults to 4): the engine replays up to 8
-# clients' whole-phase hipGraphs on 8 concurrent streams per rank;
-# streams sharing a HW queue serialize. Must be set before HIP init.
-os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
+# GPU_MAX_HW_QUEUES stays at the ROCm default (4): the engine replays
+# up to 8 clients' whole-phase hipGraphs on concurrent streams, but a
+# same-box A/B (gpurun_out/fe_q4*.json vs fe_q8*.json) measured 8 HW
+# queues ~8% SLOWER on the FEMNIST protocol round and neutral on
+# ResNet-50 — queue-scheduling overhead outweighs the extra
+# parallelism at these kernel sizes.
 
 import torch
 
