@@ -485,6 +485,7 @@ class FLEngine:
             cur = torch.cuda.current_stream()
             costs = {}
             evs = {}
+            th0 = time.perf_counter()
             for i in local_subs:
                 g, s = tgraphs[i], self._client_stream(i)
                 s.wait_stream(cur)
@@ -500,14 +501,17 @@ class FLEngine:
                                      cfg.learning_rate)
                     if self._phase_debug:
                         evs[i][1].record(s)
+            th1 = time.perf_counter()
             for i in local_subs:
                 cur.wait_stream(self._client_streams[i])
             if self._phase_debug:
                 torch.cuda.synchronize(self.device)
+                th2 = time.perf_counter()
                 tt = " ".join(f"c{i}:{evs[i][0].elapsed_time(evs[i][1]):.1f}"
                               for i in local_subs)
-                print(f"# [debug] epoch {epoch} train per-client ms: {tt}",
-                      flush=True)
+                print(f"# [debug] epoch {epoch} train per-client ms: {tt} "
+                      f"| host enqueue {1e3*(th1-th0):.2f} "
+                      f"join {1e3*(th2-th1):.2f}", flush=True)
             # ONE host sync for every trainer's accumulated loss
             costl = torch.stack([costs[i] for i in local_subs]).tolist()
             for j, i in enumerate(local_subs):
@@ -592,6 +596,7 @@ class FLEngine:
             cur = torch.cuda.current_stream()
             res = {}
             evs = {}
+            th0 = time.perf_counter()
             for i in local_scorers:
                 s = self._client_stream(i)
                 s.wait_stream(cur)
@@ -603,14 +608,17 @@ class FLEngine:
                     res[i] = sgraphs[i].run_inplace()
                     if self._phase_debug:
                         evs[i][1].record(s)
+            th1 = time.perf_counter()
             for i in local_scorers:
                 cur.wait_stream(self._client_streams[i])
             if self._phase_debug:
                 torch.cuda.synchronize(self.device)
+                th2 = time.perf_counter()
                 tt = " ".join(f"c{i}:{evs[i][0].elapsed_time(evs[i][1]):.1f}"
                               for i in local_scorers)
-                print(f"# [debug] epoch {epoch} score per-client ms: {tt}",
-                      flush=True)
+                print(f"# [debug] epoch {epoch} score per-client ms: {tt} "
+                      f"| host enqueue {1e3*(th1-th0):.2f} "
+                      f"join {1e3*(th2-th1):.2f}", flush=True)
             # ONE host sync for all scorers' [K] accuracy vectors
             alla = torch.stack([res[i] for i in local_scorers]).tolist()
             for j, i in enumerate(local_scorers):
