@@ -1953,7 +1953,69 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
   if (N % 64 != 0 || N < 64) {
     // narrow-N convs (ResNet-20's Kout 16/32): implicit gather in the
     // synchronous small-tile kernel — the col read it replaces is the
-    // traffic bound there (col is R*S times the activation)
+    // traffic bound there (col is R*S times the activation).
+    //
+    // Double-buffered narrow-BN route: the sync kernel is
+    // stage-latency bound at ~10x the traffic bound on the ResNet-20
+    // stage-1/2 convs (28.8% of a protocol round,
+    // profiles/r02_streams_kernel_stats.md) — every K-step serializes
+    // gather-stage -> barrier -> MFMA -> barrier. gemm256_kernel's
+    // BN==32 wave grid (4Mx2N) register-stages K-step t+2 while the
+    // MFMAs of t run, hiding the gather latency. Large-M shapes only
+    // (512-thread blocks need tiles >> CUs); BFLC_NARROW_DBUF=0
+    // restores the sync route for A/B.
+    static const bool narrow_dbuf = [] {
+      const char* e = getenv("BFLC_NARROW_DBUF");
+      return !(e && e[0] == '0');
+    }();
+    if (narrow_dbuf && N <= 32 && M >= 8192) {
+      const int bm2 = M >= 32768 ? 128 : 64;
+      const long tiles = ((M + bm2 - 1) / bm2) * ((N + 31) / 32);
+      const long ksteps64 = (K + BK2 - 1) / BK2;
+      const double tf = bm2 == 128 ? 450.0e12 : 250.0e12;
+      long S = pick_splitk(M, N, K, tiles, ksteps64, 256, tf);
+      const long kslice = ((ksteps64 + S - 1) / S) * BK2;
+      S = (K + kslice - 1) / kslice;
+      torch::Tensor part;
+      float* part_ptr = nullptr;
+      if (S > 1) {
+        part = torch::empty({S, M, N}, x.options().dtype(at::kFloat));
+        part_ptr = part.data_ptr<float>();
+      }
+      const bf16* a = (const bf16*)x.data_ptr();
+      const bf16* b = (const bf16*)w2.data_ptr();
+      bf16* c = (bf16*)y.data_ptr();
+      const bf16* bs = bias ? (const bf16*)bias->data_ptr() : nullptr;
+      dim3 grid((unsigned)tiles, (unsigned)S);
+      dim3 block(512);
+      float* stp = nullptr;
+      float* stq = nullptr;
+      if (bn_stats && CMODE == 1 && S == 1) {
+        const long chunks = (M + bm2 - 1) / bm2;
+        bn_stats->first = torch::empty({chunks, N},
+                                       x.options().dtype(at::kFloat));
+        bn_stats->second = torch::empty({chunks, N},
+                                        x.options().dtype(at::kFloat));
+        stp = bn_stats->first.data_ptr<float>();
+        stq = bn_stats->second.data_ptr<float>();
+      }
+      if (bm2 == 128) {
+        hipLaunchKernelGGL((gemm256_kernel<128, 32, CMODE>), grid, block,
+                           0, cur_stream(), a, b, c, part_ptr, bs, M, N,
+                           K, kslice, relu ? 1 : 0, (int)EpStore::kPlain,
+                           0, sh, stp, stq);
+      } else {
+        hipLaunchKernelGGL((gemm256_kernel<64, 32, CMODE>), grid, block,
+                           0, cur_stream(), a, b, c, part_ptr, bs, M, N,
+                           K, kslice, relu ? 1 : 0, (int)EpStore::kPlain,
+                           0, sh, stp, stq);
+      }
+      HIP_CHECK(hipGetLastError());
+      if (S > 1)
+        launch_splitk_reduce(part_ptr, S, M, N, c, bs, relu ? 1 : 0,
+                             EpStore::kPlain, 0);
+      return true;
+    }
     const bf16* a = (const bf16*)x.data_ptr();
     const bf16* b = (const bf16*)w2.data_ptr();
     bf16* c = (bf16*)y.data_ptr();
