@@ -588,6 +588,15 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
   long cv_rowbase[CMODE ? GA : 1];
   int cv_ih0[CMODE ? GA : 1], cv_iw0[CMODE ? GA : 1];
   bool cv_mok[CMODE ? GA : 1];
+  // incremental k-decode state (carry-based, no divides in the staging
+  // loop — the per-load gk/C and rs/S divides were two ~24-instruction
+  // VALU sequences per granule per K-step; these kernels are
+  // issue-bound, not memory-bound: MfmaUtil 7.6%, MemUnitStalled 0.1%
+  // on the FEMNIST conv2 fwd, profiles/r02_pmc_femnist.md). kd_in =
+  // c8 (CMODE 1: channel offset) / k8 (CMODE 2: Kout offset); load
+  // calls advance by BK2 per K-step, monotone, so carries suffice —
+  // same design as the sync kernel's staging (commit b3bec0e).
+  int kd_in[CMODE ? GA : 1], kd_rr[CMODE ? GA : 1], kd_ss[CMODE ? GA : 1];
   if (CMODE == 1) {
 #pragma unroll
     for (int i = 0; i < GA; ++i) {
@@ -601,6 +610,11 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
       cv_rowbase[i] = (long)n * csh.H * csh.W * csh.C;
       cv_ih0[i] = oh * csh.stride - csh.pad;
       cv_iw0[i] = ow * csh.stride - csh.pad;
+      const long gk0 = k_begin + (g & 7) * 8;
+      const int rs0 = (int)(gk0 / csh.C);  // one divide at init only
+      kd_in[i] = (int)(gk0 - (long)rs0 * csh.C);
+      kd_rr[i] = rs0 / csh.S;
+      kd_ss[i] = rs0 - kd_rr[i] * csh.S;
     }
   } else if (CMODE == 2) {
 #pragma unroll
@@ -615,6 +629,11 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
       cv_rowbase[i] = (long)n * csh.OH * csh.OW * csh.Kout;
       cv_ih0[i] = ih + csh.pad;  // oh_num = ih + pad - r
       cv_iw0[i] = iw + csh.pad;
+      const long gk0 = k_begin + (g & 7) * 8;
+      const int rs0 = (int)(gk0 / csh.Kout);  // one divide at init only
+      kd_in[i] = (int)(gk0 - (long)rs0 * csh.Kout);
+      kd_rr[i] = rs0 / csh.S;
+      kd_ss[i] = rs0 - kd_rr[i] * csh.S;
     }
   }
   auto load_tiles = [&](long k0) {
@@ -624,46 +643,50 @@ __global__ void gemm256_kernel(const bf16* __restrict__ A,
       const long gk = k0 + (g & 7) * 8;
       if (CMODE == 1) {
         bool ok = cv_mok[i] && gk < k_end;
-        int ih = 0, iw = 0, c8 = 0;
-        if (ok) {
-          const int rs = (int)(gk / csh.C);
-          c8 = (int)(gk - (long)rs * csh.C);
-          const int rr = rs / csh.S, ss = rs - rr * csh.S;
-          ih = cv_ih0[i] + rr;
-          iw = cv_iw0[i] + ss;
-          ok = ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W;
-        }
+        const int ih = cv_ih0[i] + kd_rr[i];
+        const int iw = cv_iw0[i] + kd_ss[i];
+        ok = ok && ih >= 0 && ih < csh.H && iw >= 0 && iw < csh.W;
         if (ok) {
           ra[i] = *reinterpret_cast<const bf16x8_t*>(
-              &A[cv_rowbase[i] + ((long)ih * csh.W + iw) * csh.C + c8]);
+              &A[cv_rowbase[i] + ((long)ih * csh.W + iw) * csh.C
+                 + kd_in[i]]);
         } else {
           u16x8_t z = {};
           ra[i] = *reinterpret_cast<const bf16x8_t*>(&z);
         }
+        // advance the decode to this granule's next K-step (carries
+        // only: load calls are monotone in k by exactly BK2)
+        int c8n = kd_in[i] + BK2;
+        int dss = 0;
+        while (c8n >= csh.C) { c8n -= csh.C; ++dss; }
+        kd_in[i] = c8n;
+        kd_ss[i] += dss;
+        while (kd_ss[i] >= csh.S) { kd_ss[i] -= csh.S; ++kd_rr[i]; }
         continue;
       }
       if (CMODE == 2) {
         bool ok = cv_mok[i] && gk < k_end;
-        int oh = 0, ow = 0, k8 = 0;
-        if (ok) {
-          const int rs = (int)(gk / csh.Kout);
-          k8 = (int)(gk - (long)rs * csh.Kout);
-          const int rr = rs / csh.S, ss = rs - rr * csh.S;
-          const int oh_num = cv_ih0[i] - rr;
-          const int ow_num = cv_iw0[i] - ss;
-          oh = oh_num / csh.stride;
-          ow = ow_num / csh.stride;
-          ok = oh_num >= 0 && ow_num >= 0 &&
-               oh_num % csh.stride == 0 && ow_num % csh.stride == 0 &&
-               oh < csh.OH && ow < csh.OW;
-        }
+        const int oh_num = cv_ih0[i] - kd_rr[i];
+        const int ow_num = cv_iw0[i] - kd_ss[i];
+        const int oh = oh_num / csh.stride;
+        const int ow = ow_num / csh.stride;
+        ok = ok && oh_num >= 0 && ow_num >= 0 &&
+             oh_num % csh.stride == 0 && ow_num % csh.stride == 0 &&
+             oh < csh.OH && ow < csh.OW;
         if (ok) {
           ra[i] = *reinterpret_cast<const bf16x8_t*>(
-              &A[cv_rowbase[i] + ((long)oh * csh.OW + ow) * csh.Kout + k8]);
+              &A[cv_rowbase[i] + ((long)oh * csh.OW + ow) * csh.Kout
+                 + kd_in[i]]);
         } else {
           u16x8_t z = {};
           ra[i] = *reinterpret_cast<const bf16x8_t*>(&z);
         }
+        int k8n = kd_in[i] + BK2;
+        int dss = 0;
+        while (k8n >= csh.Kout) { k8n -= csh.Kout; ++dss; }
+        kd_in[i] = k8n;
+        kd_ss[i] += dss;
+        while (kd_ss[i] >= csh.S) { kd_ss[i] -= csh.S; ++kd_rr[i]; }
         continue;
       }
       const long gm = tile_m + (g >> 3);
