@@ -1978,18 +1978,21 @@ bool conv_implicit_gemm(const torch::Tensor& x, const torch::Tensor& w2,
     // synchronous small-tile kernel — the col read it replaces is the
     // traffic bound there (col is R*S times the activation).
     //
-    // Double-buffered narrow-BN route: the sync kernel is
-    // stage-latency bound at ~10x the traffic bound on the ResNet-20
-    // stage-1/2 convs (28.8% of a protocol round,
-    // profiles/r02_streams_kernel_stats.md) — every K-step serializes
-    // gather-stage -> barrier -> MFMA -> barrier. gemm256_kernel's
-    // BN==32 wave grid (4Mx2N) register-stages K-step t+2 while the
-    // MFMAs of t run, hiding the gather latency. Large-M shapes only
-    // (512-thread blocks need tiles >> CUs); BFLC_NARROW_DBUF=0
-    // restores the sync route for A/B.
+    // Double-buffered narrow-BN route (default OFF — measured
+    // negative): gemm256_kernel's BN==32 wave grid register-stages
+    // K-step t+2 while the MFMAs of t run, the theory being that the
+    // sync kernel serializes gather-stage -> barrier -> MFMA at ~10x
+    // the traffic bound. A/B on hardware (gpurun_out/r20_kd*.json):
+    // ResNet-20 protocol round 67.5 ms dbuf vs 64.0 sync, FEMNIST 9.05
+    // vs 9.07 — the sync kernel's 256-thread blocks oversubscribe each
+    // CU with multiple blocks, which already hides the stage latency;
+    // the 512-thread dbuf blocks halve that block-level parallelism
+    // and its staging VALU count is no lower (the real bound —
+    // MemUnitStalled ~0, profiles/r02_pmc_femnist.md). Kept behind
+    // BFLC_NARROW_DBUF=1 as the documented experiment.
     static const bool narrow_dbuf = [] {
       const char* e = getenv("BFLC_NARROW_DBUF");
-      return !(e && e[0] == '0');
+      return e && e[0] == '1';
     }();
     if (narrow_dbuf && N <= 32 && M >= 8192) {
       const int bm2 = M >= 32768 ? 128 : 64;
