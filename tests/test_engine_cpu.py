@@ -121,6 +121,31 @@ class TestNonIID:
         assert stats[-1].epoch == 1
 
 
+class TestDirichletBalance:
+    def test_remainders_not_dumped_on_last_client(self):
+        """Regression: per-class counts were floored with the summed
+        remainder dumped on the LAST client — at many classes and few
+        samples per class (ResNet-50 protocol shape: 1000 classes,
+        ~2 samples each) client k-1 collected ~one sample per class
+        (1375 of 2048) and its serial minibatch chain dominated every
+        FL round. Cumulative rounding keeps every shard within the
+        Dirichlet proportions."""
+        import torch
+        from bflc_amd.data.synthetic import partition_dirichlet
+        gen = torch.Generator().manual_seed(0)
+        n, k, n_class = 2048, 8, 1000
+        x = torch.randn(n, 4)
+        y = torch.randint(0, n_class, (n,), generator=gen)
+        parts = partition_dirichlet(x, y, k, 0.3, n_class, gen)
+        sizes = [p[0].shape[0] for p in parts]
+        assert sum(sizes) == n
+        # the buggy flooring gave the last client >half of everything;
+        # proportional rounding keeps every shard near n/k at alpha-0.3
+        # noise levels (the last client is NOT special)
+        assert max(sizes) < n // 2, sizes
+        assert sizes[-1] < 2 * n // k, sizes
+
+
 class TestClientAssignment:
     def test_uneven_split_covers_all_clients(self):
         from bflc_amd.fl.engine import client_rank
