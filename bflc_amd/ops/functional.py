@@ -205,13 +205,19 @@ class _Conv2dFn(torch.autograd.Function):
                 dy = hip_ops().relu_bwd(y, dy)
             else:
                 dy = dy * (y > 0).to(dy.dtype)
+        # first-layer convs (x = input data, no grad needed) skip the
+        # entire dgrad GEMM + col2im — the ResNet-50 stem's dcol alone
+        # is M x 147 (~236 MB) per backward
+        want_dx = ctx.needs_input_grad[0]
         if x.is_cuda:
             col = ctx.saved_tensors[2]
             if col.numel() == 0:
                 col = None
             dx, dw, db = hip_ops().conv2d_bwd(
                 x, w, dy, ctx.stride, ctx.padding, col,
-                ctx.has_bias and db_fused is None)
+                ctx.has_bias and db_fused is None, want_dx)
+            if not want_dx:
+                dx = None
             if db_fused is not None:
                 db = db_fused
             elif not ctx.has_bias:
@@ -220,11 +226,14 @@ class _Conv2dFn(torch.autograd.Function):
             xn = x.permute(0, 3, 1, 2)
             wn = w.permute(0, 3, 1, 2)
             dyn = dy.permute(0, 3, 1, 2)
-            dx = torch.nn.grad.conv2d_input(
-                xn.shape, wn, dyn, stride=ctx.stride, padding=ctx.padding)
+            dx = None
+            if want_dx:
+                dx = torch.nn.grad.conv2d_input(
+                    xn.shape, wn, dyn, stride=ctx.stride,
+                    padding=ctx.padding)
+                dx = dx.permute(0, 2, 3, 1).contiguous()
             dw = torch.nn.grad.conv2d_weight(
                 xn, wn.shape, dyn, stride=ctx.stride, padding=ctx.padding)
-            dx = dx.permute(0, 2, 3, 1).contiguous()
             dw = dw.permute(0, 2, 3, 1).contiguous()
             db = dy.sum(dim=(0, 1, 2))
         return dx, dw, (db if ctx.has_bias else None), None, None, None, \
@@ -386,10 +395,15 @@ class _ConvBNFn(torch.autograd.Function):
         dyc, dgamma, dbeta = h.batchnorm_bwd(yc, dy, mean, invstd, gamma,
                                              yr)
         # want_db=False: the BN absorbs any bias, so no colsum pass over
-        # dyc (it was ~3.3% of a ResNet-20 round for a discarded value)
+        # dyc (it was ~3.3% of a ResNet-20 round for a discarded value).
+        # want_dx=False on the stem (x = input images): skips the dgrad
+        # GEMM + col2im entirely.
+        want_dx = ctx.needs_input_grad[0]
         dx, dw, _db = h.conv2d_bwd(x, w, dyc, ctx.stride, ctx.padding,
                                    col if col.numel() > 0 else None,
-                                   False)
+                                   False, want_dx)
+        if not want_dx:
+            dx = None
         return (dx, dw, dgamma, dbeta, None, None, None, None, dres)
 
 

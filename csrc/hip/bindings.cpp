@@ -67,7 +67,8 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          long stride, long pad, bool relu);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
-    long pad, c10::optional<torch::Tensor> col_cache, bool want_db);
+    long pad, c10::optional<torch::Tensor> col_cache, bool want_db,
+    bool want_dx);
 std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
                                                        long kernel,
                                                        long stride,
@@ -143,10 +144,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("residual") = py::none(),
         "normalization pass with known stats");
   m.def("conv2d_bwd", &bflc::conv2d_bwd,
-        "(dx, dw, db) - db empty unless want_db",
+        "(dx, dw, db) - db empty unless want_db, dx empty unless "
+        "want_dx (first-layer convs need no input gradient)",
         py::arg("x"), py::arg("w"), py::arg("dy"), py::arg("stride"),
         py::arg("pad"), py::arg("col_cache") = py::none(),
-        py::arg("want_db") = true);
+        py::arg("want_db") = true, py::arg("want_dx") = true);
   m.def("maxpool2d_fwd", &bflc::maxpool2d_fwd, py::arg("x"),
         py::arg("kernel"), py::arg("stride"),
         py::arg("want_idx") = true);

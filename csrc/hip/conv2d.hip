@@ -537,16 +537,22 @@ conv2d_fwd_bn(torch::Tensor x, torch::Tensor w, long stride, long pad) {
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, long stride,
-    long pad, c10::optional<torch::Tensor> col_cache, bool want_db) {
+    long pad, c10::optional<torch::Tensor> col_cache, bool want_db,
+    bool want_dx) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w); CHECK_CONTIG(dy);
   auto sh = make_shape(x, w, stride, pad);
   auto w2 = w.view({(long)sh.Kout, sh.RSC()});
   auto dy2 = dy.view({sh.M(), (long)sh.Kout});  // NHWC: free view
 
-  // dgrad
+  // dgrad. want_dx=false (first-layer convs: the input is data, not an
+  // activation) skips the whole dgrad GEMM + col2im — on the ResNet-50
+  // stem that is a dcol of M x 147 (~236 MB) plus its col2im pass,
+  // computed for a gradient every model threw away.
   torch::Tensor dx;
-  bool dgrad_done = false;
-  if (is_1x1_s1(sh)) {
+  bool dgrad_done = !want_dx;
+  if (!want_dx) {
+    dx = torch::empty({0}, x.options());
+  } else if (is_1x1_s1(sh)) {
     auto wT = transpose_bf16(w2);  // [C, Kout]
     dx = torch::empty_like(x);
     auto dxv = dx.view({sh.M(), (long)sh.C});

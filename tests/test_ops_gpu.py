@@ -219,6 +219,19 @@ class TestFlatOps:
             hip().adam_step_(pg, gg, mg, vg, step, 0.01, 0.9, 0.999, 1e-8)
         assert torch.allclose(pg.cpu(), p, atol=1e-5)
 
+    def test_conv_bwd_skips_dx_for_first_layer(self):
+        """want_dx=False (first-layer convs: the input is data) must
+        skip the dgrad work, return an empty dx, and leave dw/db
+        bitwise identical to the full backward."""
+        torch.manual_seed(17)
+        x = torch.randn(8, 14, 14, 16, device=DEV).bfloat16()
+        w = torch.randn(32, 3, 3, 16, device=DEV).bfloat16() * 0.1
+        dy = torch.randn(8, 14, 14, 32, device=DEV).bfloat16()
+        dx1, dw1, db1 = hip().conv2d_bwd(x, w, dy, 1, 1)
+        dx0, dw0, db0 = hip().conv2d_bwd(x, w, dy, 1, 1, want_dx=False)
+        assert dx0.numel() == 0 and dx1.numel() == x.numel()
+        assert torch.equal(dw0, dw1) and torch.equal(db0, db1)
+
     def test_score_load_bitwise_vs_old_chain(self):
         """score_load_ (one pass: shadow = bf16(global - lr*delta))
         must be BITWISE equal to the copy + axpy + cast chain it
