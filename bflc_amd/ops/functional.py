@@ -81,10 +81,15 @@ class _LinearFn(torch.autograd.Function):
         if ctx.relu:  # fused-relu epilogue: mask dy by the saved output
             dy = hip_ops().relu_bwd(y, dy) if x.is_cuda \
                 else dy * (y > 0).to(dy.dtype)
+        # first-layer linears (logreg/MLP on raw features) need no
+        # input gradient — skip the dx GEMM
+        want_dx = ctx.needs_input_grad[0]
         if x.is_cuda:
-            dx, dw, db = hip_ops().linear_bwd(x, w, dy)
+            dx, dw, db = hip_ops().linear_bwd(x, w, dy, want_dx)
+            if not want_dx:
+                dx = None
         else:
-            dx = dy @ w.t()
+            dx = dy @ w.t() if want_dx else None
             dw = x.t() @ dy
             db = dy.sum(0)
         return dx, dw, db, None

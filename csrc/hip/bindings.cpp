@@ -47,7 +47,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          bool relu);
 torch::Tensor gemm_raw(torch::Tensor A, torch::Tensor B, bool ta, bool tb);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
-    torch::Tensor x, torch::Tensor w, torch::Tensor dy);
+    torch::Tensor x, torch::Tensor w, torch::Tensor dy, bool want_dx);
 
 // conv_im2col.hip
 std::tuple<torch::Tensor, torch::Tensor> conv2d_fwd_col(
@@ -124,7 +124,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_fwd", &bflc::linear_fwd, "MFMA bf16 GEMM + bias (+relu)",
         py::arg("x"), py::arg("w"), py::arg("b"), py::arg("relu") = false);
   m.def("gemm_raw", &bflc::gemm_raw, "raw GEMM (bench/ablation)");
-  m.def("linear_bwd", &bflc::linear_bwd, "(dx, dw, db)");
+  m.def("linear_bwd", &bflc::linear_bwd,
+        "(dx, dw, db) - dx empty unless want_dx", py::arg("x"),
+        py::arg("w"), py::arg("dy"), py::arg("want_dx") = true);
   m.def("conv2d_fwd", &bflc::conv2d_fwd,
         "NHWC implicit-GEMM convolution (MFMA); inference entry, no col",
         py::arg("x"), py::arg("w"), py::arg("b"), py::arg("stride"),

@@ -2290,12 +2290,19 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
-    torch::Tensor x, torch::Tensor w, torch::Tensor dy) {
+    torch::Tensor x, torch::Tensor w, torch::Tensor dy, bool want_dx) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w); CHECK_CONTIG(dy);
   long M = x.size(0), K = x.size(1), N = w.size(1);
-  auto dx = torch::empty({M, K}, x.options());
-  gemm_bf16_raw(dy, w, dx, M, K, N, false, true, nullptr, false,
-                EpStore::kPlain, 0);
+  // want_dx=false: first-layer linears (logreg/MLP on raw features)
+  // need no input gradient — skip the dx GEMM
+  torch::Tensor dx;
+  if (want_dx) {
+    dx = torch::empty({M, K}, x.options());
+    gemm_bf16_raw(dy, w, dx, M, K, N, false, true, nullptr, false,
+                  EpStore::kPlain, 0);
+  } else {
+    dx = torch::empty({0}, x.options());
+  }
   auto dw = torch::empty({K, N}, x.options());
   gemm_bf16_raw(x, dy, dw, K, N, M, true, false, nullptr, false,
                 EpStore::kPlain, 0);
