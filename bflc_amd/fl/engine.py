@@ -211,10 +211,20 @@ class FLEngine:
             self._client_models[client] = build_model(self.cfg, self.device)
         return self._client_models[client]
 
-    def _client_stream(self, client: int):
-        if client not in self._client_streams:
-            self._client_streams[client] = torch.cuda.Stream()
-        return self._client_streams[client]
+    def _client_stream(self, slot: int):
+        """Stream for the slot-th CONCURRENT participant of the current
+        phase (NOT per client): ROCm multiplexes streams onto
+        GPU_MAX_HW_QUEUES (default 4) hardware queues and streams
+        sharing a queue serialize — with one stream per CLIENT (8
+        streams, roles rotating), a phase's 4 participants could land
+        2-per-queue and run pairwise-serial (per-client stream timings
+        showed exactly that). At most max-participants-per-phase
+        streams ever exist, so each phase's replays map to distinct
+        queues. Safe because phases are joined before the next launch
+        and a client appears at most once per phase."""
+        if slot not in self._client_streams:
+            self._client_streams[slot] = torch.cuda.Stream()
+        return self._client_streams[slot]
 
     def _capture_pool(self, client: int):
         """Graph memory pool for this client's captures. Concurrent
@@ -486,8 +496,8 @@ class FLEngine:
             costs = {}
             evs = {}
             th0 = time.perf_counter()
-            for i in local_subs:
-                g, s = tgraphs[i], self._client_stream(i)
+            for j, i in enumerate(local_subs):
+                g, s = tgraphs[i], self._client_stream(j)
                 s.wait_stream(cur)
                 with torch.cuda.stream(s):
                     if self._phase_debug:
@@ -502,8 +512,8 @@ class FLEngine:
                     if self._phase_debug:
                         evs[i][1].record(s)
             th1 = time.perf_counter()
-            for i in local_subs:
-                cur.wait_stream(self._client_streams[i])
+            for j in range(len(local_subs)):
+                cur.wait_stream(self._client_streams[j])
             if self._phase_debug:
                 torch.cuda.synchronize(self.device)
                 th2 = time.perf_counter()
@@ -597,8 +607,8 @@ class FLEngine:
             res = {}
             evs = {}
             th0 = time.perf_counter()
-            for i in local_scorers:
-                s = self._client_stream(i)
+            for j, i in enumerate(local_scorers):
+                s = self._client_stream(j)
                 s.wait_stream(cur)
                 with torch.cuda.stream(s):
                     if self._phase_debug:
@@ -609,8 +619,8 @@ class FLEngine:
                     if self._phase_debug:
                         evs[i][1].record(s)
             th1 = time.perf_counter()
-            for i in local_scorers:
-                cur.wait_stream(self._client_streams[i])
+            for j in range(len(local_scorers)):
+                cur.wait_stream(self._client_streams[j])
             if self._phase_debug:
                 torch.cuda.synchronize(self.device)
                 th2 = time.perf_counter()
