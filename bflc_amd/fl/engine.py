@@ -138,7 +138,8 @@ class FLEngine:
         # concurrent client streams: the FL-model kernels are
         # issue/launch-bound, not HBM-bound (profiles/r02_pmc_femnist.md)
         # — so a rank hosting several clients overlaps their whole-phase
-        # graph replays on one HIP stream PER CLIENT. That requires one
+        # graph replays on concurrent HIP streams (slot-indexed per
+        # phase participant, see _client_stream). That requires one
         # model replica and one private graph pool per local client
         # (concurrently-replaying graphs must not share weight buffers
         # or pool memory); the cost is #local_clients x (model +
