@@ -232,21 +232,18 @@ __global__ void col2im_nhwc_kernel(const bf16* __restrict__ dcol,
 // themselves — uint8 makes the fwd+bwd index traffic 1/4.
 typedef uint8_t u8x8_t __attribute__((ext_vector_type(8)));
 
-// Grid: (ceil(OH*OW/256), C/8, N) — the channel group and image index
-// ride blockIdx.y/z, so the per-item decode is ONE divide (ow/oh from
-// the in-image offset) instead of four; these kernels are issue-bound
-// (MemUnitStalled 0.3%, profiles/r02_pmc_femnist_streams.md).
 __global__ void maxpool_nhwc_vec_fwd(const bf16* __restrict__ x,
                                      bf16* __restrict__ y,
-                                     uint8_t* __restrict__ idx,
-                                     ConvShape sh) {
-  const int c8 = blockIdx.y * 8;
-  const int n = blockIdx.z;
-  const long ohw = (long)sh.OH * sh.OW;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < ohw;
-       i += (long)gridDim.x * blockDim.x) {
-    const int ow = (int)(i % sh.OW), oh = (int)(i / sh.OW);
-    const long m = (long)n * ohw + i;
+                                     uint8_t* __restrict__ idx, ConvShape sh,
+                                     long total_g) {
+  const int c8g = sh.C / 8;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total_g; g += stride) {
+    const int c8 = (int)(g % c8g) * 8;
+    const long m = g / c8g;
+    const int ow = (int)(m % sh.OW), oh = (int)((m / sh.OW) % sh.OH);
+    const int n = (int)(m / ((long)sh.OW * sh.OH));
     float best[8];
     int bi[8];
 #pragma unroll
@@ -282,14 +279,16 @@ __global__ void maxpool_nhwc_vec_fwd(const bf16* __restrict__ x,
 
 __global__ void maxpool_nhwc_vec_bwd(const bf16* __restrict__ dy,
                                      const uint8_t* __restrict__ idx,
-                                     bf16* __restrict__ dx, ConvShape sh) {
-  const int c8 = blockIdx.y * 8;
-  const int n = blockIdx.z;
-  const long hw = (long)sh.H * sh.W;
-  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < hw;
-       p += (long)gridDim.x * blockDim.x) {
-    const int iw = (int)(p % sh.W), ih = (int)(p / sh.W);
-    const long i = (long)n * hw + p;  // dx row (n, ih, iw)
+                                     bf16* __restrict__ dx, ConvShape sh,
+                                     long total_g) {
+  const int c8g = sh.C / 8;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; g < total_g; g += stride) {
+    const int c8 = (int)(g % c8g) * 8;
+    const long i = g / c8g;  // (n, ih, iw)
+    const int iw = (int)(i % sh.W), ih = (int)((i / sh.W) % sh.H);
+    const int n = (int)(i / ((long)sh.W * sh.H));
     const int oh_lo = max(0, (ih - sh.R + sh.stride) / sh.stride);
     const int oh_hi = min(sh.OH - 1, ih / sh.stride);
     const int ow_lo = max(0, (iw - sh.S + sh.stride) / sh.stride);
@@ -658,14 +657,12 @@ std::tuple<torch::Tensor, torch::Tensor> maxpool2d_fwd(torch::Tensor x,
                  ? torch::empty({0}, y.options().dtype(at::kByte))
                  : torch::empty_like(y, y.options().dtype(at::kByte));
   if (sh.C % 8 == 0) {
-    TORCH_CHECK(sh.N <= 65535, "maxpool batch exceeds gridDim.z");
-    const long ohw = (long)sh.OH * sh.OW;
-    dim3 grid((unsigned)std::min<long>((ohw + 255) / 256, 1024),
-              (unsigned)(sh.C / 8), (unsigned)sh.N);
-    hipLaunchKernelGGL(maxpool_nhwc_vec_fwd, grid, dim3(256), 0,
-                       cur_stream(), (const bf16*)x.data_ptr(),
-                       (bf16*)y.data_ptr(),
-                       skip_idx ? nullptr : idx.data_ptr<uint8_t>(), sh);
+    const long total_g = sh.M() * (sh.C / 8);
+    hipLaunchKernelGGL(maxpool_nhwc_vec_fwd, dim3(ew_grid(total_g)),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)x.data_ptr(), (bf16*)y.data_ptr(),
+                       skip_idx ? nullptr : idx.data_ptr<uint8_t>(), sh,
+                       total_g);
   } else {
     const long total = y.numel();
     hipLaunchKernelGGL(maxpool_nhwc_fwd, dim3(ew_grid(total)), dim3(256), 0,
@@ -693,13 +690,11 @@ torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx,
   auto dx = torch::empty({(long)sh.N, (long)sh.H, (long)sh.W, (long)sh.C},
                          dy.options());
   if (sh.C % 8 == 0) {
-    TORCH_CHECK(sh.N <= 65535, "maxpool batch exceeds gridDim.z");
-    const long hw = (long)sh.H * sh.W;
-    dim3 grid((unsigned)std::min<long>((hw + 255) / 256, 1024),
-              (unsigned)(sh.C / 8), (unsigned)sh.N);
-    hipLaunchKernelGGL(maxpool_nhwc_vec_bwd, grid, dim3(256), 0,
-                       cur_stream(), (const bf16*)dy.data_ptr(),
-                       idx.data_ptr<uint8_t>(), (bf16*)dx.data_ptr(), sh);
+    const long total_g = (long)sh.N * sh.H * sh.W * (sh.C / 8);
+    hipLaunchKernelGGL(maxpool_nhwc_vec_bwd, dim3(ew_grid(total_g)),
+                       dim3(256), 0, cur_stream(),
+                       (const bf16*)dy.data_ptr(), idx.data_ptr<uint8_t>(),
+                       (bf16*)dx.data_ptr(), sh, total_g);
   } else {
     const long total = dx.numel();
     hipLaunchKernelGGL(maxpool_nhwc_bwd, dim3(ew_grid(total)), dim3(256), 0,
