@@ -592,7 +592,7 @@ class FLEngine:
             for i in self.local_clients:
                 self._phase_scorer(i)
         # Concurrent path: every local scorer's whole-scoring graph (all
-        # K candidates in one replay) runs on that client's stream;
+        # K candidates in one replay) runs on its own slot stream;
         # scores are read back once after the join.
         conc_s = self._concurrent and packed and len(local_scorers) > 1
         sgraphs = {}
