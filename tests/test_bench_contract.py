@@ -36,6 +36,9 @@ def test_bench_json_contract():
     assert d["dtype"] in ("bf16", "fp32")
     assert d["data"] == "synthetic"
     assert 0.0 <= d["test_acc"] <= 1.0
+    # untimed convergence-soak fields are always present; on CPU the
+    # soak is skipped (GPU femnist runs default to 1500 rounds)
+    assert d["soak_rounds"] == 0 and d["soak_test_acc"] is None
     cfg = d["config"]
     for key in ("model", "global_batch", "parallelism", "clients",
                 "committee", "update_quota", "aggregate_top_k",
