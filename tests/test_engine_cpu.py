@@ -253,6 +253,47 @@ def test_random_config_protocol_soak():
     check()
 
 
+class TestDegradedRound:
+    def test_all_forged_updates_degrade_not_wedge(self):
+        """If EVERY update submission fails HMAC verification (hostile
+        transport / mis-bound keys), zero candidates are admitted —
+        the round must still complete: empty decision, unchanged
+        model, epoch advances, committee rotates, and the next clean
+        round recovers. (The reference's chain would simply not reach
+        its quota and poll forever; the barrier engine degrades
+        instead.)"""
+        cfg = FLConfig.for_world(4, model="logreg", n_features=8,
+                                 n_class=2, samples_per_client=64,
+                                 batch_size=32, eval_samples=64)
+        shards, test = make_federated(cfg)
+        eng = FLEngine(cfg, Transport(), shards, test)
+        flat0 = eng.global_flat.clone()
+        from bflc_amd.chain.identity import KeyTable
+        orig = eng.keys
+        wrong = KeyTable(eng.origins, seed=999)  # attacker's keys
+
+        class SplitKeys:
+            """Sign updates with the WRONG key table (an attacker
+            without the bootstrap seed); verify with the true one."""
+            def sign(self, kind, o, e, p):
+                return (wrong if kind == "update" else orig).sign(
+                    kind, o, e, p)
+
+            def verify(self, *a):
+                return orig.verify(*a)
+
+        eng.keys = SplitKeys()
+        st = eng.run_round()
+        assert st.n_updates == 0 and st.n_selected == 0
+        assert torch.equal(eng.global_flat, flat0)  # model unchanged
+        assert eng.ledger.epoch == 1  # epoch still advanced
+
+        eng.keys = orig  # clean round recovers
+        st2 = eng.run_round()
+        assert st2.n_updates > 0 and st2.n_selected > 0
+        assert not torch.equal(eng.global_flat, flat0)
+
+
 class TestMaxEpoch:
     """max_epoch is enforced end-to-end (reference main.py:65,251-252:
     clients exit once epoch > MAX_EPOCH) — VERDICT round-1 missing #6."""
