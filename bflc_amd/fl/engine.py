@@ -22,6 +22,17 @@ first-come PBFT ordering; here admission order is the deterministic
 admitted and skip training the surplus trainers entirely — same
 semantics (rejected updates never influence state), less wasted compute.
 
+Execution model (GPU): each local client's ENTIRE training pass and
+ENTIRE candidate-scoring sweep are captured once as hipGraphs
+(fl/graphs.py) against that client's own model replica and private
+graph pool; within a phase, the participants' replays run CONCURRENTLY
+on slot-indexed HIP streams and join at the phase barrier. Per-client
+kernel order is unchanged, so results are bitwise-identical to the
+sequential path (BFLC_STREAMS=0). The FL-model kernels are
+issue/launch-bound rather than HBM-bound (profiles/r02_pmc_*.md), which
+is why overlap — not more per-kernel bandwidth — is what shortens the
+round.
+
 Data plane (flat fp32 delta tensors, the global model) stays resident on
 the GPU; the ledger's blobs hold b"" markers in engine mode (the
 ABI-compatible JSON blob path lives in bflc_amd.chain.client).
