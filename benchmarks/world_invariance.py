@@ -27,11 +27,13 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--rounds", type=int, default=100)
     ap.add_argument("--model", default="mlp")
+    ap.add_argument("--byzantine", type=int, default=0)
     args = ap.parse_args()
     cfg = FLConfig.for_world(8, model=args.model, n_features=32,
                              n_class=8, samples_per_client=128,
                              batch_size=64, eval_samples=128,
                              partition="dirichlet",
+                             byzantine_clients=args.byzantine,
                              max_epoch=args.rounds + 10)
     t = Transport()
     shards, test = make_federated(cfg)
@@ -40,8 +42,10 @@ def main():
     h = hashlib.sha256(
         eng.global_flat.cpu().numpy().tobytes()).hexdigest()
     if t.rank == 0:
+        sel = sorted({o for o, _ in eng.last_decision.selected})
         print(f"world={t.world_size} rounds={args.rounds} "
-              f"model_sha256={h} acc={eng.evaluate_global():.4f}",
+              f"byz={args.byzantine} model_sha256={h} "
+              f"acc={eng.evaluate_global():.4f} last_selected={sel}",
               flush=True)
     t.close()
 
