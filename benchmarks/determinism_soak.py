@@ -36,6 +36,7 @@ def main():
     ap.add_argument("--rounds", type=int, default=2000)
     ap.add_argument("--model", default="mlp",
                     choices=["logreg", "mlp", "femnist_cnn"])
+    ap.add_argument("--optimizer", default="sgd", choices=["sgd", "adam"])
     ap.add_argument("--restore-at", type=int, default=None,
                     help="checkpoint/restore engine B at this round "
                          "(default rounds//2)")
@@ -46,6 +47,7 @@ def main():
                              n_class=8, samples_per_client=128,
                              batch_size=64, eval_samples=128,
                              partition="dirichlet",
+                             optimizer=args.optimizer,
                              max_epoch=args.rounds + 10)
     a, b = build(cfg), build(cfg)
     ck = os.path.join(tempfile.gettempdir(), "det_soak_ck.pt")
